@@ -1,0 +1,87 @@
+// Python bindings for the deeplearning_amd HIP kernel library (gfx950-only).
+#include <torch/extension.h>
+
+// layernorm.hip
+std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor w,
+                                         torch::Tensor b, double eps);
+std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
+                                         torch::Tensor w, torch::Tensor mean,
+                                         torch::Tensor rstd);
+// elementwise.hip
+torch::Tensor gelu_fwd(torch::Tensor x);
+torch::Tensor gelu_bwd(torch::Tensor dy, torch::Tensor x);
+torch::Tensor silu_fwd(torch::Tensor x);
+torch::Tensor silu_bwd(torch::Tensor dy, torch::Tensor x);
+torch::Tensor add_relu_fwd(torch::Tensor a, torch::Tensor b);
+torch::Tensor relu_mask_bwd(torch::Tensor dy, torch::Tensor y);
+// batchnorm.hip
+std::vector<torch::Tensor> batchnorm_fwd(torch::Tensor x, torch::Tensor weight,
+                                         torch::Tensor bias,
+                                         c10::optional<torch::Tensor> running_mean,
+                                         c10::optional<torch::Tensor> running_var,
+                                         double momentum, double eps, bool relu);
+torch::Tensor bn_apply(torch::Tensor x, torch::Tensor scale, torch::Tensor shift,
+                       bool relu);
+std::vector<torch::Tensor> batchnorm_bwd(torch::Tensor dy, torch::Tensor x,
+                                         c10::optional<torch::Tensor> y,
+                                         torch::Tensor weight, torch::Tensor mean,
+                                         torch::Tensor rstd, bool relu);
+// softmax_ce.hip
+std::vector<torch::Tensor> softmax_ce_fwd(torch::Tensor logits,
+                                          c10::optional<torch::Tensor> target,
+                                          c10::optional<torch::Tensor> soft_target,
+                                          double smoothing, int64_t ignore_index);
+torch::Tensor softmax_ce_bwd(torch::Tensor logits,
+                             c10::optional<torch::Tensor> target,
+                             c10::optional<torch::Tensor> soft_target,
+                             torch::Tensor lse, double smoothing,
+                             int64_t ignore_index, double grad_scale);
+// focal.hip
+torch::Tensor focal_loss_fwd(torch::Tensor logits, torch::Tensor targets,
+                             double alpha, double gamma);
+torch::Tensor focal_loss_bwd(torch::Tensor dloss, torch::Tensor logits,
+                             torch::Tensor targets, double alpha, double gamma);
+// boxes.hip
+torch::Tensor box_iou_gpu(torch::Tensor a, torch::Tensor b, bool giou);
+torch::Tensor nms_gpu(torch::Tensor boxes_sorted, double iou_threshold);
+// roialign.hip
+torch::Tensor roialign_fwd(torch::Tensor input, torch::Tensor rois, int64_t PH,
+                           int64_t PW, double spatial_scale,
+                           int64_t sampling_ratio, bool aligned);
+torch::Tensor roialign_bwd(torch::Tensor grad_out, torch::Tensor rois, int64_t N,
+                           int64_t C, int64_t H, int64_t W, double spatial_scale,
+                           int64_t sampling_ratio, bool aligned);
+// window.hip
+torch::Tensor window_partition_fwd(torch::Tensor x, int64_t ws, int64_t shift);
+torch::Tensor window_partition_bwd(torch::Tensor grad, int64_t B, int64_t H,
+                                   int64_t W, int64_t ws, int64_t shift);
+torch::Tensor window_merge_fwd(torch::Tensor windows, int64_t B, int64_t H,
+                               int64_t W, int64_t ws, int64_t shift);
+torch::Tensor window_merge_bwd(torch::Tensor grad, int64_t ws, int64_t shift);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "deeplearning_amd gfx950 HIP kernels";
+  m.def("layernorm_fwd", &layernorm_fwd);
+  m.def("layernorm_bwd", &layernorm_bwd);
+  m.def("gelu_fwd", &gelu_fwd);
+  m.def("gelu_bwd", &gelu_bwd);
+  m.def("silu_fwd", &silu_fwd);
+  m.def("silu_bwd", &silu_bwd);
+  m.def("add_relu_fwd", &add_relu_fwd);
+  m.def("relu_mask_bwd", &relu_mask_bwd);
+  m.def("batchnorm_fwd", &batchnorm_fwd);
+  m.def("bn_apply", &bn_apply);
+  m.def("batchnorm_bwd", &batchnorm_bwd);
+  m.def("softmax_ce_fwd", &softmax_ce_fwd);
+  m.def("softmax_ce_bwd", &softmax_ce_bwd);
+  m.def("focal_loss_fwd", &focal_loss_fwd);
+  m.def("focal_loss_bwd", &focal_loss_bwd);
+  m.def("box_iou", &box_iou_gpu);
+  m.def("nms", &nms_gpu);
+  m.def("roialign_fwd", &roialign_fwd);
+  m.def("roialign_bwd", &roialign_bwd);
+  m.def("window_partition_fwd", &window_partition_fwd);
+  m.def("window_partition_bwd", &window_partition_bwd);
+  m.def("window_merge_fwd", &window_merge_fwd);
+  m.def("window_merge_bwd", &window_merge_bwd);
+}

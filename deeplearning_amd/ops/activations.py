@@ -1,0 +1,80 @@
+"""Activation ops backed by HIP kernels (csrc/elementwise.hip) on GPU.
+
+gelu (erf form, matches nn.GELU default), silu (yolov5 Conv blocks), fused
+residual add+ReLU (ResNet block join, classification/resnet/models/networks.py).
+"""
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+
+from ._ext import ext, use_hip
+
+
+class _GeluFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        x = x.contiguous()
+        ctx.save_for_backward(x)
+        return ext().gelu_fwd(x)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (x,) = ctx.saved_tensors
+        return ext().gelu_bwd(dy.contiguous(), x)
+
+
+class _SiluFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        x = x.contiguous()
+        ctx.save_for_backward(x)
+        return ext().silu_fwd(x)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (x,) = ctx.saved_tensors
+        return ext().silu_bwd(dy.contiguous(), x)
+
+
+class _AddReluFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, a, b):
+        y = ext().add_relu_fwd(a.contiguous(), b.contiguous())
+        ctx.save_for_backward(y)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (y,) = ctx.saved_tensors
+        dx = ext().relu_mask_bwd(dy.contiguous(), y)
+        return dx, dx
+
+
+def gelu(x: torch.Tensor) -> torch.Tensor:
+    if use_hip(x):
+        return _GeluFn.apply(x)
+    return torch.nn.functional.gelu(x)
+
+
+def silu(x: torch.Tensor) -> torch.Tensor:
+    if use_hip(x):
+        return _SiluFn.apply(x)
+    return torch.nn.functional.silu(x)
+
+
+def add_relu(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    """relu(a + b) fused — the ResNet residual join."""
+    if use_hip(a, b):
+        return _AddReluFn.apply(a, b)
+    return torch.relu(a + b)
+
+
+class GELU(nn.Module):
+    def forward(self, x):
+        return gelu(x)
+
+
+class SiLU(nn.Module):
+    def forward(self, x):
+        return silu(x)
