@@ -1,0 +1,145 @@
+#!/usr/bin/env python3
+"""Flagship training-step benchmark (BASELINE.json contract).
+
+Measures whole-job images/sec for ResNet-50 (default) or ViT-B/16 at 224^2,
+bf16, synthetic data, random-init weights, full train step (forward + loss +
+backward + DDP gradient all-reduce + SGD update) on N GPUs of one node.
+
+Launch (driver contract):
+  python bench.py --gpus 1 --steps K --warmup W
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+from deeplearning_amd.core.dist import (barrier, cleanup, get_rank,
+                                        get_world_size, init_distributed)
+from deeplearning_amd.data.synthetic import DeviceBatchLoader
+from deeplearning_amd.models import build_model
+from deeplearning_amd.ops import cross_entropy
+from deeplearning_amd.parallel import wrap_data_parallel
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=30)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--model", default="resnet50", choices=["resnet50", "vit_b16"])
+    p.add_argument("--batch-size", type=int, default=0,
+                   help="per-GPU batch (0 = model default)")
+    p.add_argument("--dp", default="bucketed", choices=["bucketed", "torch"])
+    p.add_argument("--bucket-mb", type=float, default=64.0)
+    p.add_argument("--no-channels-last", action="store_true")
+    p.add_argument("--lr", type=float, default=0.1)
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+    info = init_distributed()
+    world = get_world_size()
+    rank = get_rank()
+    has_gpu = torch.cuda.is_available()
+    device = torch.device("cuda", info["local_rank"]) if has_gpu else torch.device("cpu")
+    if has_gpu:
+        torch.cuda.set_device(device)
+        torch.backends.cudnn.benchmark = True
+
+    per_gpu_batch = args.batch_size
+    if per_gpu_batch == 0:
+        per_gpu_batch = (256 if args.model == "resnet50" else 256) if has_gpu else 8
+
+    torch.manual_seed(1234)
+    model = build_model(args.model, num_classes=1000).to(device)
+    channels_last = (args.model == "resnet50") and not args.no_channels_last and has_gpu
+    if channels_last:
+        model = model.to(memory_format=torch.channels_last)
+    model = wrap_data_parallel(model, style=args.dp, bucket_cap_mb=args.bucket_mb) \
+        if world > 1 else model
+
+    params = [p for p in model.parameters() if p.requires_grad]
+    optimizer = torch.optim.SGD(params, lr=args.lr, momentum=0.9,
+                                weight_decay=1e-4)
+
+    loader = DeviceBatchLoader(per_gpu_batch, (3, 224, 224), 1000,
+                               steps=args.warmup + args.steps, device=device,
+                               distinct=2, seed=42 + rank,
+                               channels_last=channels_last)
+    batches = loader.batches
+    amp_dtype = torch.bfloat16
+    amp_enabled = has_gpu
+    finalize = getattr(model, "finalize", None)
+
+    def one_step(i):
+        x, y = batches[i % len(batches)]
+        with torch.autocast("cuda", dtype=amp_dtype, enabled=amp_enabled):
+            logits = model(x)
+            loss = cross_entropy(logits, y)
+        loss.backward()
+        if finalize is not None:
+            finalize()
+        optimizer.step()
+        optimizer.zero_grad(set_to_none=True)
+        return loss
+
+    # warmup
+    for i in range(args.warmup):
+        one_step(i)
+
+    barrier()
+    if has_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        one_step(args.warmup + i)
+    barrier()
+    if has_gpu:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+
+    # take the max elapsed over ranks
+    t = torch.tensor([elapsed], dtype=torch.float64, device=device if has_gpu else "cpu")
+    if world > 1:
+        import torch.distributed as dist
+
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    elapsed_max = float(t.item())
+
+    global_batch = per_gpu_batch * world
+    images_per_sec = global_batch * args.steps / elapsed_max
+    if rank == 0:
+        result = {
+            "metric": "images/sec",
+            "value": round(images_per_sec, 2),
+            "unit": "images/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed_max / args.steps * 1000, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if amp_enabled else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": global_batch,
+                "seq_len": None,
+                "image_size": 224,
+                "parallelism": f"dp{world}",
+            },
+        }
+        print(json.dumps(result), flush=True)
+    cleanup()
+
+
+if __name__ == "__main__":
+    main()

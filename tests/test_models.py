@@ -1,0 +1,50 @@
+"""CPU forward-shape tests for the model zoo."""
+import pytest
+import torch
+
+from deeplearning_amd.models import build_model, list_models
+
+
+@pytest.mark.parametrize("name,shape,n", [
+    ("mnist_cnn", (1, 28, 28), 10),
+    ("mnist_fcn", (1, 28, 28), 10),
+    ("resnet18", (3, 64, 64), 100),
+    ("resnet50", (3, 64, 64), 100),
+    ("resnext50_32x4d", (3, 64, 64), 10),
+])
+def test_forward_shapes(name, shape, n):
+    m = build_model(name, num_classes=n)
+    x = torch.randn(2, *shape)
+    y = m(x)
+    assert y.shape == (2, n)
+
+
+def test_vit_forward():
+    m = build_model("vit_b16", num_classes=10)
+    y = m(torch.randn(1, 3, 224, 224))
+    assert y.shape == (1, 10)
+
+
+def test_resnet50_param_count():
+    m = build_model("resnet50")
+    assert abs(sum(p.numel() for p in m.parameters()) - 25_557_032) < 10
+
+
+def test_vit_b16_param_count():
+    m = build_model("vit_b16")
+    assert abs(sum(p.numel() for p in m.parameters()) - 86_567_656) < 10
+
+
+def test_backward_smoke():
+    m = build_model("resnet18", num_classes=10)
+    x = torch.randn(2, 3, 64, 64)
+    loss = m(x).sum()
+    loss.backward()
+    assert all(p.grad is not None for p in m.parameters())
+
+
+def test_registry_lists():
+    models = list_models()
+    assert "resnet50" in models and "vit_b16" in models
+    with pytest.raises(KeyError):
+        build_model("not_a_model")
