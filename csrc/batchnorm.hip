@@ -139,6 +139,120 @@ __global__ void bn_bwd_dx_kernel(const dev_t* __restrict__ dy,
   }
 }
 
+// ======================= NHWC (channels_last) variants =======================
+// NHWC is the fast layout on MI355X (MIOpen conv prefers it; BN reductions are
+// column sums over a [N*H*W, C] row-major matrix -> fully coalesced).
+
+// pass 1: column sums. threads = channels (coalesced), grid.y splits rows.
+template <typename dev_t>
+__global__ void bn_stats_nhwc_kernel(const dev_t* __restrict__ x,
+                                     float* __restrict__ sums, int C,
+                                     int64_t rows) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float sum = 0.f, sumsq = 0.f;
+  for (int64_t r = blockIdx.y; r < rows; r += gridDim.y) {
+    const float f = to_f32(x[r * C + c]);
+    sum += f;
+    sumsq += f * f;
+  }
+  if (gridDim.y == 1) {
+    sums[c] = sum;
+    sums[C + c] = sumsq;
+  } else {
+    atomicAdd(&sums[c], sum);
+    atomicAdd(&sums[C + c], sumsq);
+  }
+}
+
+// pass 2: y = x*scale[c] + shift[c] (+ReLU); V consecutive CHANNELS per lane.
+template <typename dev_t, int V, bool RELU>
+__global__ void bn_apply_nhwc_kernel(const dev_t* __restrict__ x,
+                                     const float* __restrict__ scale,
+                                     const float* __restrict__ shift,
+                                     dev_t* __restrict__ y, int C,
+                                     int64_t n_total) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x * V;
+  for (int64_t i = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * V;
+       i < n_total; i += stride) {
+    const int c0 = (int)(i % C);
+    Vec<dev_t, V> xv = vload<dev_t, V>(x + i);
+    Vec<dev_t, V> yv;
+#pragma unroll
+    for (int j = 0; j < V; ++j) {
+      float f = to_f32(xv.v[j]) * scale[c0 + j] + shift[c0 + j];
+      yv.v[j] = from_f32<dev_t>(RELU ? fmaxf(f, 0.f) : f);
+    }
+    vstore<dev_t, V>(y + i, yv);
+  }
+}
+
+template <typename dev_t, bool RELU>
+__global__ void bn_bwd_stats_nhwc_kernel(const dev_t* __restrict__ dy,
+                                         const dev_t* __restrict__ x,
+                                         const dev_t* __restrict__ y,
+                                         const float* __restrict__ mean,
+                                         const float* __restrict__ rstd,
+                                         float* __restrict__ sums, int C,
+                                         int64_t rows) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  const float mu = mean[c], rs = rstd[c];
+  float s_dy = 0.f, s_dyxh = 0.f;
+  for (int64_t r = blockIdx.y; r < rows; r += gridDim.y) {
+    const int64_t idx = r * C + c;
+    float g = to_f32(dy[idx]);
+    if (RELU && to_f32(y[idx]) <= 0.f) g = 0.f;
+    const float xh = (to_f32(x[idx]) - mu) * rs;
+    s_dy += g;
+    s_dyxh += g * xh;
+  }
+  if (gridDim.y == 1) {
+    sums[c] = s_dy;
+    sums[C + c] = s_dyxh;
+  } else {
+    atomicAdd(&sums[c], s_dy);
+    atomicAdd(&sums[C + c], s_dyxh);
+  }
+}
+
+template <typename dev_t, int V, bool RELU>
+__global__ void bn_bwd_dx_nhwc_kernel(const dev_t* __restrict__ dy,
+                                      const dev_t* __restrict__ x,
+                                      const dev_t* __restrict__ y,
+                                      const float* __restrict__ mean,
+                                      const float* __restrict__ rstd,
+                                      const float* __restrict__ weight,
+                                      const float* __restrict__ sums,
+                                      dev_t* __restrict__ dx, int C,
+                                      int64_t n_total, float inv_count) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x * V;
+  for (int64_t i = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * V;
+       i < n_total; i += stride) {
+    const int c0 = (int)(i % C);
+    Vec<dev_t, V> dyv = vload<dev_t, V>(dy + i);
+    Vec<dev_t, V> xv = vload<dev_t, V>(x + i);
+    Vec<dev_t, V> yv;
+    if (RELU) yv = vload<dev_t, V>(y + i);
+    Vec<dev_t, V> dxv;
+#pragma unroll
+    for (int j = 0; j < V; ++j) {
+      const int c = c0 + j;
+      float g = to_f32(dyv.v[j]);
+      if (RELU && to_f32(yv.v[j]) <= 0.f) g = 0.f;
+      const float xh = (to_f32(xv.v[j]) - mean[c]) * rstd[c];
+      dxv.v[j] = from_f32<dev_t>(weight[c] * rstd[c] *
+                                 (g - sums[c] * inv_count - xh * sums[C + c] * inv_count));
+    }
+    vstore<dev_t, V>(dx + i, dxv);
+  }
+}
+
+inline bool is_nhwc(const torch::Tensor& t) {
+  return t.dim() == 4 && t.is_contiguous(at::MemoryFormat::ChannelsLast) &&
+         !t.is_contiguous();
+}
+
 }  // namespace dla
 
 // Returns {y, save_mean, save_rstd}. Updates running stats in-place when given.
@@ -147,8 +261,11 @@ std::vector<torch::Tensor> batchnorm_fwd(torch::Tensor x, torch::Tensor weight,
                                          c10::optional<torch::Tensor> running_mean,
                                          c10::optional<torch::Tensor> running_var,
                                          double momentum, double eps, bool relu) {
-  DLA_CHECK_INPUT(x);
-  TORCH_CHECK(x.dim() == 4, "batchnorm_fwd expects NCHW");
+  DLA_CHECK_CUDA(x);
+  TORCH_CHECK(x.dim() == 4, "batchnorm_fwd expects a 4D tensor");
+  const bool nhwc = dla::is_nhwc(x);
+  TORCH_CHECK(nhwc || x.is_contiguous(), "batchnorm_fwd: x must be contiguous "
+              "(NCHW or channels_last)");
   const int N = (int)x.size(0), C = (int)x.size(1);
   const int64_t HW = x.size(2) * x.size(3);
   auto opts_f = x.options().dtype(torch::kFloat);
@@ -161,12 +278,20 @@ std::vector<torch::Tensor> batchnorm_fwd(torch::Tensor x, torch::Tensor weight,
   auto w32 = weight.to(torch::kFloat);
   auto b32 = bias.to(torch::kFloat);
 
-  const int ysplit = (int)std::min<int64_t>(((int64_t)N * HW + 65535) / 65536 + 1,
+  const int64_t rows = (int64_t)N * HW;
+  const int ysplit = (int)std::min<int64_t>((rows + 65535) / 65536 + 1,
                                             std::max(1, 2048 / C));
   DLA_DISPATCH_FLOAT_TYPES(x.scalar_type(), "batchnorm_fwd", [&] {
-    hipLaunchKernelGGL((dla::bn_stats_kernel<dev_t>), dim3(C, ysplit), dim3(256), 0,
-                       dla::stream(), (const dev_t*)x.data_ptr(),
-                       sums.data_ptr<float>(), N, C, HW);
+    if (nhwc) {
+      dim3 g((C + 255) / 256, 64);
+      hipLaunchKernelGGL((dla::bn_stats_nhwc_kernel<dev_t>), g, dim3(256), 0,
+                         dla::stream(), (const dev_t*)x.data_ptr(),
+                         sums.data_ptr<float>(), C, rows);
+    } else {
+      hipLaunchKernelGGL((dla::bn_stats_kernel<dev_t>), dim3(C, ysplit), dim3(256), 0,
+                         dla::stream(), (const dev_t*)x.data_ptr(),
+                         sums.data_ptr<float>(), N, C, HW);
+    }
     hipLaunchKernelGGL(dla::bn_finalize_kernel, dim3((C + 255) / 256), dim3(256), 0,
                        dla::stream(), sums.data_ptr<float>(), w32.data_ptr<float>(),
                        b32.data_ptr<float>(), mean.data_ptr<float>(),
@@ -174,19 +299,26 @@ std::vector<torch::Tensor> batchnorm_fwd(torch::Tensor x, torch::Tensor weight,
                        running_mean.has_value() ? running_mean->data_ptr<float>() : nullptr,
                        running_var.has_value() ? running_var->data_ptr<float>() : nullptr,
                        scale.data_ptr<float>(), shift.data_ptr<float>(), C,
-                       (float)((int64_t)N * HW), (float)eps, (float)momentum);
+                       (float)rows, (float)eps, (float)momentum);
     const int64_t n_total = x.numel();
     constexpr int VMAX = 16 / (int)sizeof(dev_t);
     auto launch = [&](auto vtag, auto rtag) {
       constexpr int V = decltype(vtag)::value;
       constexpr bool R = decltype(rtag)::value;
       const int grid = dla::grid_1d((n_total + V - 1) / V, 256);
-      hipLaunchKernelGGL((dla::bn_apply_kernel<dev_t, V, R>), dim3(grid), dim3(256),
-                         0, dla::stream(), (const dev_t*)x.data_ptr(),
-                         scale.data_ptr<float>(), shift.data_ptr<float>(),
-                         (dev_t*)y.data_ptr(), C, HW, n_total);
+      if (nhwc)
+        hipLaunchKernelGGL((dla::bn_apply_nhwc_kernel<dev_t, V, R>), dim3(grid),
+                           dim3(256), 0, dla::stream(), (const dev_t*)x.data_ptr(),
+                           scale.data_ptr<float>(), shift.data_ptr<float>(),
+                           (dev_t*)y.data_ptr(), C, n_total);
+      else
+        hipLaunchKernelGGL((dla::bn_apply_kernel<dev_t, V, R>), dim3(grid), dim3(256),
+                           0, dla::stream(), (const dev_t*)x.data_ptr(),
+                           scale.data_ptr<float>(), shift.data_ptr<float>(),
+                           (dev_t*)y.data_ptr(), C, HW, n_total);
     };
-    if (HW % VMAX == 0) {
+    const bool vec_ok = nhwc ? (C % VMAX == 0) : (HW % VMAX == 0);
+    if (vec_ok) {
       if (relu) launch(std::integral_constant<int, VMAX>{}, std::true_type{});
       else launch(std::integral_constant<int, VMAX>{}, std::false_type{});
     } else {
@@ -201,8 +333,10 @@ std::vector<torch::Tensor> batchnorm_fwd(torch::Tensor x, torch::Tensor weight,
 // Frozen/eval BN apply (+optional ReLU): per-channel scale/shift precomputed host-side.
 torch::Tensor bn_apply(torch::Tensor x, torch::Tensor scale, torch::Tensor shift,
                        bool relu) {
-  DLA_CHECK_INPUT(x);
-  TORCH_CHECK(x.dim() == 4, "bn_apply expects NCHW");
+  DLA_CHECK_CUDA(x);
+  TORCH_CHECK(x.dim() == 4, "bn_apply expects a 4D tensor");
+  const bool nhwc = dla::is_nhwc(x);
+  TORCH_CHECK(nhwc || x.is_contiguous(), "bn_apply: x must be contiguous");
   const int C = (int)x.size(1);
   const int64_t HW = x.size(2) * x.size(3);
   auto y = torch::empty_like(x);
@@ -215,12 +349,19 @@ torch::Tensor bn_apply(torch::Tensor x, torch::Tensor scale, torch::Tensor shift
       constexpr int V = decltype(vtag)::value;
       constexpr bool R = decltype(rtag)::value;
       const int grid = dla::grid_1d((n_total + V - 1) / V, 256);
-      hipLaunchKernelGGL((dla::bn_apply_kernel<dev_t, V, R>), dim3(grid), dim3(256),
-                         0, dla::stream(), (const dev_t*)x.data_ptr(),
-                         sc.data_ptr<float>(), sh.data_ptr<float>(),
-                         (dev_t*)y.data_ptr(), C, HW, n_total);
+      if (nhwc)
+        hipLaunchKernelGGL((dla::bn_apply_nhwc_kernel<dev_t, V, R>), dim3(grid),
+                           dim3(256), 0, dla::stream(), (const dev_t*)x.data_ptr(),
+                           sc.data_ptr<float>(), sh.data_ptr<float>(),
+                           (dev_t*)y.data_ptr(), C, n_total);
+      else
+        hipLaunchKernelGGL((dla::bn_apply_kernel<dev_t, V, R>), dim3(grid), dim3(256),
+                           0, dla::stream(), (const dev_t*)x.data_ptr(),
+                           sc.data_ptr<float>(), sh.data_ptr<float>(),
+                           (dev_t*)y.data_ptr(), C, HW, n_total);
     };
-    if (HW % VMAX == 0) {
+    const bool vec_ok = nhwc ? (C % VMAX == 0) : (HW % VMAX == 0);
+    if (vec_ok) {
       if (relu) launch(std::integral_constant<int, VMAX>{}, std::true_type{});
       else launch(std::integral_constant<int, VMAX>{}, std::false_type{});
     } else {
@@ -237,46 +378,71 @@ std::vector<torch::Tensor> batchnorm_bwd(torch::Tensor dy, torch::Tensor x,
                                          c10::optional<torch::Tensor> y,
                                          torch::Tensor weight, torch::Tensor mean,
                                          torch::Tensor rstd, bool relu) {
-  DLA_CHECK_INPUT(dy); DLA_CHECK_INPUT(x);
+  DLA_CHECK_CUDA(dy); DLA_CHECK_CUDA(x);
+  const bool nhwc = dla::is_nhwc(x);
+  TORCH_CHECK(nhwc || x.is_contiguous(), "batchnorm_bwd: x must be contiguous");
+  if (nhwc && !dla::is_nhwc(dy)) dy = dy.contiguous(at::MemoryFormat::ChannelsLast);
+  if (!nhwc) dy = dy.contiguous();
   const int N = (int)x.size(0), C = (int)x.size(1);
   const int64_t HW = x.size(2) * x.size(3);
+  const int64_t rows = (int64_t)N * HW;
   auto opts_f = x.options().dtype(torch::kFloat);
   auto sums = torch::zeros({2 * C}, opts_f);
   auto dx = torch::empty_like(x);
   auto w32 = weight.to(torch::kFloat);
-  const int ysplit = (int)std::min<int64_t>(((int64_t)N * HW + 65535) / 65536 + 1,
+  const int ysplit = (int)std::min<int64_t>((rows + 65535) / 65536 + 1,
                                             std::max(1, 2048 / C));
   TORCH_CHECK(!relu || y.has_value(), "batchnorm_bwd: relu mask needs y");
   DLA_DISPATCH_FLOAT_TYPES(x.scalar_type(), "batchnorm_bwd", [&] {
     const dev_t* yp = y.has_value() ? (const dev_t*)y->data_ptr() : nullptr;
-    if (relu) {
-      hipLaunchKernelGGL((dla::bn_bwd_stats_kernel<dev_t, true>), dim3(C, ysplit),
-                         dim3(256), 0, dla::stream(), (const dev_t*)dy.data_ptr(),
-                         (const dev_t*)x.data_ptr(), yp, mean.data_ptr<float>(),
-                         rstd.data_ptr<float>(), sums.data_ptr<float>(), N, C, HW);
-    } else {
-      hipLaunchKernelGGL((dla::bn_bwd_stats_kernel<dev_t, false>), dim3(C, ysplit),
-                         dim3(256), 0, dla::stream(), (const dev_t*)dy.data_ptr(),
-                         (const dev_t*)x.data_ptr(), yp, mean.data_ptr<float>(),
-                         rstd.data_ptr<float>(), sums.data_ptr<float>(), N, C, HW);
-    }
+    auto launch_stats = [&](auto rtag) {
+      constexpr bool R = decltype(rtag)::value;
+      if (nhwc) {
+        dim3 g((C + 255) / 256, 64);
+        hipLaunchKernelGGL((dla::bn_bwd_stats_nhwc_kernel<dev_t, R>), g, dim3(256),
+                           0, dla::stream(), (const dev_t*)dy.data_ptr(),
+                           (const dev_t*)x.data_ptr(), yp, mean.data_ptr<float>(),
+                           rstd.data_ptr<float>(), sums.data_ptr<float>(), C, rows);
+      } else {
+        hipLaunchKernelGGL((dla::bn_bwd_stats_kernel<dev_t, R>), dim3(C, ysplit),
+                           dim3(256), 0, dla::stream(), (const dev_t*)dy.data_ptr(),
+                           (const dev_t*)x.data_ptr(), yp, mean.data_ptr<float>(),
+                           rstd.data_ptr<float>(), sums.data_ptr<float>(), N, C, HW);
+      }
+    };
+    if (relu) launch_stats(std::true_type{});
+    else launch_stats(std::false_type{});
+
     const int64_t n_total = x.numel();
-    const int grid = dla::grid_1d(n_total, 256);
-    const float inv_count = 1.f / (float)((int64_t)N * HW);
-    if (relu) {
-      hipLaunchKernelGGL((dla::bn_bwd_dx_kernel<dev_t, true>), dim3(grid), dim3(256),
-                         0, dla::stream(), (const dev_t*)dy.data_ptr(),
-                         (const dev_t*)x.data_ptr(), yp, mean.data_ptr<float>(),
-                         rstd.data_ptr<float>(), w32.data_ptr<float>(),
-                         sums.data_ptr<float>(), (dev_t*)dx.data_ptr(), C, HW,
-                         n_total, inv_count);
+    const float inv_count = 1.f / (float)rows;
+    constexpr int VMAX = 16 / (int)sizeof(dev_t);
+    auto launch_dx = [&](auto vtag, auto rtag) {
+      constexpr int V = decltype(vtag)::value;
+      constexpr bool R = decltype(rtag)::value;
+      const int grid = dla::grid_1d((n_total + V - 1) / V, 256);
+      if (nhwc) {
+        hipLaunchKernelGGL((dla::bn_bwd_dx_nhwc_kernel<dev_t, V, R>), dim3(grid),
+                           dim3(256), 0, dla::stream(), (const dev_t*)dy.data_ptr(),
+                           (const dev_t*)x.data_ptr(), yp, mean.data_ptr<float>(),
+                           rstd.data_ptr<float>(), w32.data_ptr<float>(),
+                           sums.data_ptr<float>(), (dev_t*)dx.data_ptr(), C,
+                           n_total, inv_count);
+      } else {
+        hipLaunchKernelGGL((dla::bn_bwd_dx_kernel<dev_t, R>), dim3(dla::grid_1d(n_total, 256)),
+                           dim3(256), 0, dla::stream(), (const dev_t*)dy.data_ptr(),
+                           (const dev_t*)x.data_ptr(), yp, mean.data_ptr<float>(),
+                           rstd.data_ptr<float>(), w32.data_ptr<float>(),
+                           sums.data_ptr<float>(), (dev_t*)dx.data_ptr(), C, HW,
+                           n_total, inv_count);
+      }
+    };
+    const bool vec_ok = nhwc && (C % VMAX == 0);
+    if (vec_ok) {
+      if (relu) launch_dx(std::integral_constant<int, VMAX>{}, std::true_type{});
+      else launch_dx(std::integral_constant<int, VMAX>{}, std::false_type{});
     } else {
-      hipLaunchKernelGGL((dla::bn_bwd_dx_kernel<dev_t, false>), dim3(grid), dim3(256),
-                         0, dla::stream(), (const dev_t*)dy.data_ptr(),
-                         (const dev_t*)x.data_ptr(), yp, mean.data_ptr<float>(),
-                         rstd.data_ptr<float>(), w32.data_ptr<float>(),
-                         sums.data_ptr<float>(), (dev_t*)dx.data_ptr(), C, HW,
-                         n_total, inv_count);
+      if (relu) launch_dx(std::integral_constant<int, 1>{}, std::true_type{});
+      else launch_dx(std::integral_constant<int, 1>{}, std::false_type{});
     }
   });
   HIP_CHECK_ERR();

@@ -17,6 +17,13 @@
 #define DLA_CHECK_CUDA(x) TORCH_CHECK((x).is_cuda(), #x " must be a GPU tensor")
 #define DLA_CHECK_CONTIG(x) TORCH_CHECK((x).is_contiguous(), #x " must be contiguous")
 #define DLA_CHECK_INPUT(x) DLA_CHECK_CUDA(x); DLA_CHECK_CONTIG(x)
+// dense = standard-contiguous OR channels_last-contiguous (flat elementwise ok)
+#define DLA_CHECK_DENSE(x)                                                     \
+  DLA_CHECK_CUDA(x);                                                           \
+  TORCH_CHECK((x).is_contiguous() ||                                           \
+                  ((x).dim() == 4 &&                                           \
+                   (x).is_contiguous(at::MemoryFormat::ChannelsLast)),         \
+              #x " must be dense (contiguous or channels_last)")
 
 #define HIP_CHECK_ERR()                                                        \
   do {                                                                         \

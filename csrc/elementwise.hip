@@ -92,7 +92,7 @@ __global__ void relu_mask_grad_kernel(const dev_t* __restrict__ dy,
 
 template <EwOp OP>
 torch::Tensor ew_unary(torch::Tensor x, c10::optional<torch::Tensor> g) {
-  DLA_CHECK_INPUT(x);
+  DLA_CHECK_DENSE(x);
   auto y = torch::empty_like(x);
   const int64_t n = x.numel();
   DLA_DISPATCH_FLOAT_TYPES(x.scalar_type(), "ew_unary", [&] {
@@ -128,7 +128,7 @@ torch::Tensor silu_bwd(torch::Tensor dy, torch::Tensor x) {
 }
 
 torch::Tensor add_relu_fwd(torch::Tensor a, torch::Tensor b) {
-  DLA_CHECK_INPUT(a); DLA_CHECK_INPUT(b);
+  DLA_CHECK_DENSE(a); DLA_CHECK_DENSE(b);
   TORCH_CHECK(a.sizes() == b.sizes(), "add_relu: shape mismatch");
   auto y = torch::empty_like(a);
   const int64_t n = a.numel();
@@ -149,7 +149,7 @@ torch::Tensor add_relu_fwd(torch::Tensor a, torch::Tensor b) {
 }
 
 torch::Tensor relu_mask_bwd(torch::Tensor dy, torch::Tensor y) {
-  DLA_CHECK_INPUT(dy); DLA_CHECK_INPUT(y);
+  DLA_CHECK_DENSE(dy); DLA_CHECK_DENSE(y);
   auto dx = torch::empty_like(dy);
   const int64_t n = dy.numel();
   DLA_DISPATCH_FLOAT_TYPES(dy.scalar_type(), "relu_mask_bwd", [&] {

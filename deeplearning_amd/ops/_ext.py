@@ -47,6 +47,24 @@ def ext():
     return m
 
 
+def dense(t: torch.Tensor) -> torch.Tensor:
+    """Return a densely-laid-out tensor WITHOUT destroying channels_last:
+    .contiguous() on a channels_last tensor would copy to NCHW."""
+    if t.is_contiguous():
+        return t
+    if t.dim() == 4 and t.is_contiguous(memory_format=torch.channels_last):
+        return t
+    return t.contiguous()
+
+
+def same_layout(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    """Make b match a's memory format (for binary elementwise kernels)."""
+    if a.dim() == 4 and a.is_contiguous(memory_format=torch.channels_last) \
+            and not a.is_contiguous():
+        return b.contiguous(memory_format=torch.channels_last)
+    return b.contiguous()
+
+
 def use_hip(*tensors) -> bool:
     """True iff all tensors are on GPU. On GPU the extension is REQUIRED:
     if it is missing this raises instead of falling back."""

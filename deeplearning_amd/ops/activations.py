@@ -8,46 +8,47 @@ from __future__ import annotations
 import torch
 import torch.nn as nn
 
-from ._ext import ext, use_hip
+from ._ext import dense, ext, same_layout, use_hip
 
 
 class _GeluFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x):
-        x = x.contiguous()
+        x = dense(x)
         ctx.save_for_backward(x)
         return ext().gelu_fwd(x)
 
     @staticmethod
     def backward(ctx, dy):
         (x,) = ctx.saved_tensors
-        return ext().gelu_bwd(dy.contiguous(), x)
+        return ext().gelu_bwd(same_layout(x, dy), x)
 
 
 class _SiluFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x):
-        x = x.contiguous()
+        x = dense(x)
         ctx.save_for_backward(x)
         return ext().silu_fwd(x)
 
     @staticmethod
     def backward(ctx, dy):
         (x,) = ctx.saved_tensors
-        return ext().silu_bwd(dy.contiguous(), x)
+        return ext().silu_bwd(same_layout(x, dy), x)
 
 
 class _AddReluFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, a, b):
-        y = ext().add_relu_fwd(a.contiguous(), b.contiguous())
+        a = dense(a)
+        y = ext().add_relu_fwd(a, same_layout(a, b))
         ctx.save_for_backward(y)
         return y
 
     @staticmethod
     def backward(ctx, dy):
         (y,) = ctx.saved_tensors
-        dx = ext().relu_mask_bwd(dy.contiguous(), y)
+        dx = ext().relu_mask_bwd(same_layout(y, dy), y)
         return dx, dx
 
 

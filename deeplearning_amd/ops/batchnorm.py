@@ -10,13 +10,13 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from ._ext import ext, use_hip
+from ._ext import dense, ext, use_hip
 
 
 class _BNFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, running_mean, running_var, momentum, eps, relu):
-        x = x.contiguous()
+        x = dense(x)
         y, mean, rstd = ext().batchnorm_fwd(
             x, weight, bias, running_mean, running_var, momentum, eps, relu)
         ctx.save_for_backward(x, y, weight, mean, rstd)
@@ -27,7 +27,7 @@ class _BNFn(torch.autograd.Function):
     def backward(ctx, dy):
         x, y, weight, mean, rstd = ctx.saved_tensors
         dx, dw, db = ext().batchnorm_bwd(
-            dy.contiguous(), x, y if ctx.relu else None, weight, mean, rstd, ctx.relu)
+            dy, x, y if ctx.relu else None, weight, mean, rstd, ctx.relu)
         return dx, dw, db, None, None, None, None, None
 
 
@@ -60,7 +60,7 @@ class BatchNorm2d(nn.BatchNorm2d):
         if torch.is_grad_enabled() and (x.requires_grad or self.weight.requires_grad):
             y = x * scale.reshape(1, -1, 1, 1).to(x.dtype) + shift.reshape(1, -1, 1, 1).to(x.dtype)
             return torch.relu(y) if self.relu else y
-        return ext().bn_apply(x.contiguous(), scale, shift, self.relu)
+        return ext().bn_apply(dense(x), scale, shift, self.relu)
 
 
 class FrozenBatchNorm2d(nn.Module):
@@ -83,7 +83,7 @@ class FrozenBatchNorm2d(nn.Module):
         scale = self.weight.float() * rstd
         shift = self.bias.float() - self.running_mean.float() * scale
         if use_hip(x) and not (torch.is_grad_enabled() and x.requires_grad):
-            return ext().bn_apply(x.contiguous(), scale, shift, False)
+            return ext().bn_apply(dense(x), scale, shift, False)
         sc = scale.reshape(1, -1, 1, 1).to(x.dtype)
         sh = shift.reshape(1, -1, 1, 1).to(x.dtype)
         return x * sc + sh
