@@ -65,9 +65,16 @@ def same_layout(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
     return b.contiguous()
 
 
+import os
+
+_FORCE_EAGER = os.environ.get("DLA_FORCE_EAGER", "0") == "1"  # A/B perf testing only
+
+
 def use_hip(*tensors) -> bool:
     """True iff all tensors are on GPU. On GPU the extension is REQUIRED:
     if it is missing this raises instead of falling back."""
+    if _FORCE_EAGER:
+        return False
     on_gpu = all(t.is_cuda for t in tensors if isinstance(t, torch.Tensor))
     if not on_gpu:
         return False
