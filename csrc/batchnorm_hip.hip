@@ -144,26 +144,50 @@ __global__ void bn_bwd_dx_kernel(const dev_t* __restrict__ dy,
 // NHWC is the fast layout on MI355X (MIOpen conv prefers it; BN reductions are
 // column sums over a [N*H*W, C] row-major matrix -> fully coalesced).
 
-// pass 1: column sums. threads = channels (coalesced), grid.y splits rows.
+// pass 1 (stage A): column partial sums of a [rows, C] row-major matrix.
+// Flat thread space = C channels x K row-chunks: t -> (c = t % C, chunk = t/C).
+// Lane-adjacent threads read adjacent channels -> fully coalesced; each thread
+// accumulates rows/K rows locally and writes ONE partial pair (no atomics).
+// part layout: [0, K*C) sums, [K*C, 2*K*C) sumsq.
+inline int nhwc_chunks(int C, int64_t rows) {
+  int64_t k = 524288 / C;            // ~full-chip thread count / C
+  if (k > rows) k = rows;
+  if (k > 2048) k = 2048;
+  if (k < 1) k = 1;
+  return (int)k;
+}
+
 template <typename dev_t>
-__global__ void bn_stats_nhwc_kernel(const dev_t* __restrict__ x,
-                                     float* __restrict__ sums, int C,
-                                     int64_t rows) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+__global__ void bn_stats_nhwc_partial_kernel(const dev_t* __restrict__ x,
+                                             float* __restrict__ part, int C,
+                                             int K, int64_t rows) {
+  const int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (t >= (int64_t)C * K) return;
+  const int c = (int)(t % C);
+  const int chunk = (int)(t / C);
   float sum = 0.f, sumsq = 0.f;
-  for (int64_t r = blockIdx.y; r < rows; r += gridDim.y) {
+  for (int64_t r = chunk; r < rows; r += K) {
     const float f = to_f32(x[r * C + c]);
     sum += f;
     sumsq += f * f;
   }
-  if (gridDim.y == 1) {
-    sums[c] = sum;
-    sums[C + c] = sumsq;
-  } else {
-    atomicAdd(&sums[c], sum);
-    atomicAdd(&sums[C + c], sumsq);
+  part[t] = sum;
+  part[(int64_t)K * C + t] = sumsq;
+}
+
+// stage B: fold K partials per channel into sums[0..2C)
+__global__ void bn_stats_nhwc_fold_kernel(const float* __restrict__ part,
+                                          float* __restrict__ sums, int C, int K) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float s = 0.f, sq = 0.f;
+  const int64_t off2 = (int64_t)K * C;
+  for (int j = 0; j < K; ++j) {
+    s += part[(int64_t)j * C + c];
+    sq += part[off2 + (int64_t)j * C + c];
   }
+  sums[c] = s;
+  sums[C + c] = sq;
 }
 
 // pass 2: y = x*scale[c] + shift[c] (+ReLU); V consecutive CHANNELS per lane.
@@ -189,18 +213,20 @@ __global__ void bn_apply_nhwc_kernel(const dev_t* __restrict__ x,
 }
 
 template <typename dev_t, bool RELU>
-__global__ void bn_bwd_stats_nhwc_kernel(const dev_t* __restrict__ dy,
-                                         const dev_t* __restrict__ x,
-                                         const dev_t* __restrict__ y,
-                                         const float* __restrict__ mean,
-                                         const float* __restrict__ rstd,
-                                         float* __restrict__ sums, int C,
-                                         int64_t rows) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+__global__ void bn_bwd_stats_nhwc_partial_kernel(const dev_t* __restrict__ dy,
+                                                 const dev_t* __restrict__ x,
+                                                 const dev_t* __restrict__ y,
+                                                 const float* __restrict__ mean,
+                                                 const float* __restrict__ rstd,
+                                                 float* __restrict__ part, int C,
+                                                 int K, int64_t rows) {
+  const int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (t >= (int64_t)C * K) return;
+  const int c = (int)(t % C);
+  const int chunk = (int)(t / C);
   const float mu = mean[c], rs = rstd[c];
   float s_dy = 0.f, s_dyxh = 0.f;
-  for (int64_t r = blockIdx.y; r < rows; r += gridDim.y) {
+  for (int64_t r = chunk; r < rows; r += K) {
     const int64_t idx = r * C + c;
     float g = to_f32(dy[idx]);
     if (RELU && to_f32(y[idx]) <= 0.f) g = 0.f;
@@ -208,13 +234,8 @@ __global__ void bn_bwd_stats_nhwc_kernel(const dev_t* __restrict__ dy,
     s_dy += g;
     s_dyxh += g * xh;
   }
-  if (gridDim.y == 1) {
-    sums[c] = s_dy;
-    sums[C + c] = s_dyxh;
-  } else {
-    atomicAdd(&sums[c], s_dy);
-    atomicAdd(&sums[C + c], s_dyxh);
-  }
+  part[t] = s_dy;
+  part[(int64_t)K * C + t] = s_dyxh;
 }
 
 template <typename dev_t, int V, bool RELU>
@@ -284,10 +305,16 @@ std::vector<torch::Tensor> batchnorm_fwd(torch::Tensor x, torch::Tensor weight,
                                             std::max(1, 2048 / C));
   DLA_DISPATCH_FLOAT_TYPES(x.scalar_type(), "batchnorm_fwd", [&] {
     if (nhwc) {
-      dim3 g((C + 255) / 256, 64);
-      hipLaunchKernelGGL((dla::bn_stats_nhwc_kernel<dev_t>), g, dim3(256), 0,
+      const int K = dla::nhwc_chunks(C, rows);
+      auto part = torch::empty({2LL * K * C}, opts_f);
+      const int64_t nthreads = (int64_t)K * C;
+      hipLaunchKernelGGL((dla::bn_stats_nhwc_partial_kernel<dev_t>),
+                         dim3((nthreads + 255) / 256), dim3(256), 0,
                          dla::stream(), (const dev_t*)x.data_ptr(),
-                         sums.data_ptr<float>(), C, rows);
+                         part.data_ptr<float>(), C, K, rows);
+      hipLaunchKernelGGL(dla::bn_stats_nhwc_fold_kernel,
+                         dim3((C + 255) / 256), dim3(256), 0, dla::stream(),
+                         part.data_ptr<float>(), sums.data_ptr<float>(), C, K);
     } else {
       hipLaunchKernelGGL((dla::bn_stats_kernel<dev_t>), dim3(C, ysplit), dim3(256), 0,
                          dla::stream(), (const dev_t*)x.data_ptr(),
@@ -399,11 +426,17 @@ std::vector<torch::Tensor> batchnorm_bwd(torch::Tensor dy, torch::Tensor x,
     auto launch_stats = [&](auto rtag) {
       constexpr bool R = decltype(rtag)::value;
       if (nhwc) {
-        dim3 g((C + 255) / 256, 64);
-        hipLaunchKernelGGL((dla::bn_bwd_stats_nhwc_kernel<dev_t, R>), g, dim3(256),
-                           0, dla::stream(), (const dev_t*)dy.data_ptr(),
+        const int K = dla::nhwc_chunks(C, rows);
+        auto part = torch::empty({2LL * K * C}, opts_f);
+        const int64_t nthreads = (int64_t)K * C;
+        hipLaunchKernelGGL((dla::bn_bwd_stats_nhwc_partial_kernel<dev_t, R>),
+                           dim3((nthreads + 255) / 256), dim3(256), 0,
+                           dla::stream(), (const dev_t*)dy.data_ptr(),
                            (const dev_t*)x.data_ptr(), yp, mean.data_ptr<float>(),
-                           rstd.data_ptr<float>(), sums.data_ptr<float>(), C, rows);
+                           rstd.data_ptr<float>(), part.data_ptr<float>(), C, K, rows);
+        hipLaunchKernelGGL(dla::bn_stats_nhwc_fold_kernel,
+                           dim3((C + 255) / 256), dim3(256), 0, dla::stream(),
+                           part.data_ptr<float>(), sums.data_ptr<float>(), C, K);
       } else {
         hipLaunchKernelGGL((dla::bn_bwd_stats_kernel<dev_t, R>), dim3(C, ysplit),
                            dim3(256), 0, dla::stream(), (const dev_t*)dy.data_ptr(),

@@ -49,6 +49,11 @@ class BatchNorm2d(nn.BatchNorm2d):
             y = F.batch_norm(x, self.running_mean, self.running_var, self.weight,
                              self.bias, self.training, self.momentum, self.eps)
             return torch.relu(y) if self.relu else y
+        if self.running_mean.dtype != torch.float32:
+            # .to(bf16) on the module converts buffers; the HIP kernel keeps
+            # running statistics in fp32 — restore them once.
+            self.running_mean.data = self.running_mean.data.float()
+            self.running_var.data = self.running_var.data.float()
         if self.training:
             if self.num_batches_tracked is not None:
                 self.num_batches_tracked.add_(1)

@@ -204,7 +204,7 @@ def test_soft_target_ce(dtype):
 @pytest.mark.parametrize("alpha", [0.25, -1.0])
 def test_focal_loss(dtype, alpha):
     torch.manual_seed(0)
-    logits = torch.randn(5000, device=DEV, dtype=dtype, requires_grad=True) * 3
+    logits = (torch.randn(5000, device=DEV, dtype=dtype) * 3).requires_grad_(True)
     targets = (torch.rand(5000, device=DEV) > 0.9).to(dtype)
     loss = ops.sigmoid_focal_loss(logits, targets, alpha=alpha, reduction="sum")
     lr_ = logits.detach().float().requires_grad_(True)
