@@ -144,98 +144,130 @@ __global__ void bn_bwd_dx_kernel(const dev_t* __restrict__ dy,
 // NHWC is the fast layout on MI355X (MIOpen conv prefers it; BN reductions are
 // column sums over a [N*H*W, C] row-major matrix -> fully coalesced).
 
-// pass 1 (stage A): column partial sums of a [rows, C] row-major matrix.
-// Flat thread space = C channels x K row-chunks: t -> (c = t % C, chunk = t/C).
-// Lane-adjacent threads read adjacent channels -> fully coalesced; each thread
-// accumulates rows/K rows locally and writes ONE partial pair (no atomics).
-// part layout: [0, K*C) sums, [K*C, 2*K*C) sumsq.
-inline int nhwc_chunks(int C, int64_t rows) {
-  int64_t k = 524288 / C;            // ~full-chip thread count / C
+// NHWC kernels share one thread mapping: flat thread t -> (c0 = (t %
+// lanes_per_row)*V, row = t / lanes_per_row), rows strided by
+// used_threads/lanes_per_row. Adjacent lanes read adjacent 16B channel chunks
+// (fully coalesced); per-channel parameters load ONCE per thread (c0 is
+// loop-invariant). Stats kernels block-reduce through LDS (ds_add_f32) and
+// issue 2*C global atomics per block.
+
+inline int64_t nhwc_used_threads(int lanes_per_row, int64_t rows,
+                                 int64_t target_lanes) {
+  int64_t k = target_lanes / lanes_per_row;
   if (k > rows) k = rows;
-  if (k > 2048) k = 2048;
   if (k < 1) k = 1;
-  return (int)k;
+  return (int64_t)lanes_per_row * k;
 }
 
-template <typename dev_t>
-__global__ void bn_stats_nhwc_partial_kernel(const dev_t* __restrict__ x,
-                                             float* __restrict__ part, int C,
-                                             int K, int64_t rows) {
+template <typename dev_t, int V>
+__global__ void bn_stats_nhwc_kernel(const dev_t* __restrict__ x,
+                                     float* __restrict__ sums, int C,
+                                     int64_t rows, int64_t used) {
+  extern __shared__ float ls[];  // 2*C floats
+  for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) ls[i] = 0.f;
+  __syncthreads();
   const int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (t >= (int64_t)C * K) return;
-  const int c = (int)(t % C);
-  const int chunk = (int)(t / C);
-  float sum = 0.f, sumsq = 0.f;
-  for (int64_t r = chunk; r < rows; r += K) {
-    const float f = to_f32(x[r * C + c]);
-    sum += f;
-    sumsq += f * f;
+  const int lanes_per_row = C / V;
+  if (t < used) {
+    const int c0 = (int)(t % lanes_per_row) * V;
+    const int64_t rstride = used / lanes_per_row;
+    float s[V], q[V];
+#pragma unroll
+    for (int j = 0; j < V; ++j) { s[j] = 0.f; q[j] = 0.f; }
+    for (int64_t r = t / lanes_per_row; r < rows; r += rstride) {
+      Vec<dev_t, V> xv = vload<dev_t, V>(x + r * C + c0);
+#pragma unroll
+      for (int j = 0; j < V; ++j) {
+        const float f = to_f32(xv.v[j]);
+        s[j] += f;
+        q[j] += f * f;
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < V; ++j) {
+      atomicAdd(&ls[c0 + j], s[j]);
+      atomicAdd(&ls[C + c0 + j], q[j]);
+    }
   }
-  part[t] = sum;
-  part[(int64_t)K * C + t] = sumsq;
+  __syncthreads();
+  for (int i = threadIdx.x; i < 2 * C; i += blockDim.x)
+    atomicAdd(&sums[i], ls[i]);
 }
 
-// stage B: fold K partials per channel into sums[0..2C)
-__global__ void bn_stats_nhwc_fold_kernel(const float* __restrict__ part,
-                                          float* __restrict__ sums, int C, int K) {
-  const int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
-  float s = 0.f, sq = 0.f;
-  const int64_t off2 = (int64_t)K * C;
-  for (int j = 0; j < K; ++j) {
-    s += part[(int64_t)j * C + c];
-    sq += part[off2 + (int64_t)j * C + c];
-  }
-  sums[c] = s;
-  sums[C + c] = sq;
-}
-
-// pass 2: y = x*scale[c] + shift[c] (+ReLU); V consecutive CHANNELS per lane.
 template <typename dev_t, int V, bool RELU>
 __global__ void bn_apply_nhwc_kernel(const dev_t* __restrict__ x,
                                      const float* __restrict__ scale,
                                      const float* __restrict__ shift,
-                                     dev_t* __restrict__ y, int C,
-                                     int64_t n_total) {
-  const int64_t stride = (int64_t)gridDim.x * blockDim.x * V;
-  for (int64_t i = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * V;
-       i < n_total; i += stride) {
-    const int c0 = (int)(i % C);
-    Vec<dev_t, V> xv = vload<dev_t, V>(x + i);
+                                     dev_t* __restrict__ y, int C, int64_t rows,
+                                     int64_t used) {
+  const int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (t >= used) return;
+  const int lanes_per_row = C / V;
+  const int c0 = (int)(t % lanes_per_row) * V;
+  const int64_t rstride = used / lanes_per_row;
+  float sc[V], sh[V];
+#pragma unroll
+  for (int j = 0; j < V; ++j) { sc[j] = scale[c0 + j]; sh[j] = shift[c0 + j]; }
+  for (int64_t r = t / lanes_per_row; r < rows; r += rstride) {
+    Vec<dev_t, V> xv = vload<dev_t, V>(x + r * C + c0);
     Vec<dev_t, V> yv;
 #pragma unroll
     for (int j = 0; j < V; ++j) {
-      float f = to_f32(xv.v[j]) * scale[c0 + j] + shift[c0 + j];
+      const float f = to_f32(xv.v[j]) * sc[j] + sh[j];
       yv.v[j] = from_f32<dev_t>(RELU ? fmaxf(f, 0.f) : f);
     }
-    vstore<dev_t, V>(y + i, yv);
+    vstore<dev_t, V>(y + r * C + c0, yv);
   }
 }
 
-template <typename dev_t, bool RELU>
-__global__ void bn_bwd_stats_nhwc_partial_kernel(const dev_t* __restrict__ dy,
-                                                 const dev_t* __restrict__ x,
-                                                 const dev_t* __restrict__ y,
-                                                 const float* __restrict__ mean,
-                                                 const float* __restrict__ rstd,
-                                                 float* __restrict__ part, int C,
-                                                 int K, int64_t rows) {
+template <typename dev_t, int V, bool RELU>
+__global__ void bn_bwd_stats_nhwc_kernel(const dev_t* __restrict__ dy,
+                                         const dev_t* __restrict__ x,
+                                         const dev_t* __restrict__ y,
+                                         const float* __restrict__ mean,
+                                         const float* __restrict__ rstd,
+                                         float* __restrict__ sums, int C,
+                                         int64_t rows, int64_t used) {
+  extern __shared__ float ls[];
+  for (int i = threadIdx.x; i < 2 * C; i += blockDim.x) ls[i] = 0.f;
+  __syncthreads();
   const int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (t >= (int64_t)C * K) return;
-  const int c = (int)(t % C);
-  const int chunk = (int)(t / C);
-  const float mu = mean[c], rs = rstd[c];
-  float s_dy = 0.f, s_dyxh = 0.f;
-  for (int64_t r = chunk; r < rows; r += K) {
-    const int64_t idx = r * C + c;
-    float g = to_f32(dy[idx]);
-    if (RELU && to_f32(y[idx]) <= 0.f) g = 0.f;
-    const float xh = (to_f32(x[idx]) - mu) * rs;
-    s_dy += g;
-    s_dyxh += g * xh;
+  const int lanes_per_row = C / V;
+  if (t < used) {
+    const int c0 = (int)(t % lanes_per_row) * V;
+    const int64_t rstride = used / lanes_per_row;
+    float mu[V], rs[V], s_dy[V], s_dyxh[V];
+#pragma unroll
+    for (int j = 0; j < V; ++j) {
+      mu[j] = mean[c0 + j];
+      rs[j] = rstd[c0 + j];
+      s_dy[j] = 0.f;
+      s_dyxh[j] = 0.f;
+    }
+    for (int64_t r = t / lanes_per_row; r < rows; r += rstride) {
+      const int64_t base = r * C + c0;
+      Vec<dev_t, V> dyv = vload<dev_t, V>(dy + base);
+      Vec<dev_t, V> xv = vload<dev_t, V>(x + base);
+      Vec<dev_t, V> yv;
+      if (RELU) yv = vload<dev_t, V>(y + base);
+#pragma unroll
+      for (int j = 0; j < V; ++j) {
+        float g = to_f32(dyv.v[j]);
+        if (RELU && to_f32(yv.v[j]) <= 0.f) g = 0.f;
+        const float xh = (to_f32(xv.v[j]) - mu[j]) * rs[j];
+        s_dy[j] += g;
+        s_dyxh[j] += g * xh;
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < V; ++j) {
+      atomicAdd(&ls[c0 + j], s_dy[j]);
+      atomicAdd(&ls[C + c0 + j], s_dyxh[j]);
+    }
   }
-  part[t] = s_dy;
-  part[(int64_t)K * C + t] = s_dyxh;
+  __syncthreads();
+  for (int i = threadIdx.x; i < 2 * C; i += blockDim.x)
+    atomicAdd(&sums[i], ls[i]);
 }
 
 template <typename dev_t, int V, bool RELU>
@@ -247,26 +279,37 @@ __global__ void bn_bwd_dx_nhwc_kernel(const dev_t* __restrict__ dy,
                                       const float* __restrict__ weight,
                                       const float* __restrict__ sums,
                                       dev_t* __restrict__ dx, int C,
-                                      int64_t n_total, float inv_count) {
-  const int64_t stride = (int64_t)gridDim.x * blockDim.x * V;
-  for (int64_t i = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * V;
-       i < n_total; i += stride) {
-    const int c0 = (int)(i % C);
-    Vec<dev_t, V> dyv = vload<dev_t, V>(dy + i);
-    Vec<dev_t, V> xv = vload<dev_t, V>(x + i);
+                                      int64_t rows, int64_t used,
+                                      float inv_count) {
+  const int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (t >= used) return;
+  const int lanes_per_row = C / V;
+  const int c0 = (int)(t % lanes_per_row) * V;
+  const int64_t rstride = used / lanes_per_row;
+  float mu[V], rs[V], wr[V], m_dy[V], m_dyxh[V];
+#pragma unroll
+  for (int j = 0; j < V; ++j) {
+    mu[j] = mean[c0 + j];
+    rs[j] = rstd[c0 + j];
+    wr[j] = weight[c0 + j] * rs[j];
+    m_dy[j] = sums[c0 + j] * inv_count;
+    m_dyxh[j] = sums[C + c0 + j] * inv_count;
+  }
+  for (int64_t r = t / lanes_per_row; r < rows; r += rstride) {
+    const int64_t base = r * C + c0;
+    Vec<dev_t, V> dyv = vload<dev_t, V>(dy + base);
+    Vec<dev_t, V> xv = vload<dev_t, V>(x + base);
     Vec<dev_t, V> yv;
-    if (RELU) yv = vload<dev_t, V>(y + i);
+    if (RELU) yv = vload<dev_t, V>(y + base);
     Vec<dev_t, V> dxv;
 #pragma unroll
     for (int j = 0; j < V; ++j) {
-      const int c = c0 + j;
       float g = to_f32(dyv.v[j]);
       if (RELU && to_f32(yv.v[j]) <= 0.f) g = 0.f;
-      const float xh = (to_f32(xv.v[j]) - mean[c]) * rstd[c];
-      dxv.v[j] = from_f32<dev_t>(weight[c] * rstd[c] *
-                                 (g - sums[c] * inv_count - xh * sums[C + c] * inv_count));
+      const float xh = (to_f32(xv.v[j]) - mu[j]) * rs[j];
+      dxv.v[j] = from_f32<dev_t>(wr[j] * (g - m_dy[j] - xh * m_dyxh[j]));
     }
-    vstore<dev_t, V>(dx + i, dxv);
+    vstore<dev_t, V>(dx + r * C + c0, dxv);
   }
 }
 
@@ -305,16 +348,20 @@ std::vector<torch::Tensor> batchnorm_fwd(torch::Tensor x, torch::Tensor weight,
                                             std::max(1, 2048 / C));
   DLA_DISPATCH_FLOAT_TYPES(x.scalar_type(), "batchnorm_fwd", [&] {
     if (nhwc) {
-      const int K = dla::nhwc_chunks(C, rows);
-      auto part = torch::empty({2LL * K * C}, opts_f);
-      const int64_t nthreads = (int64_t)K * C;
-      hipLaunchKernelGGL((dla::bn_stats_nhwc_partial_kernel<dev_t>),
-                         dim3((nthreads + 255) / 256), dim3(256), 0,
-                         dla::stream(), (const dev_t*)x.data_ptr(),
-                         part.data_ptr<float>(), C, K, rows);
-      hipLaunchKernelGGL(dla::bn_stats_nhwc_fold_kernel,
-                         dim3((C + 255) / 256), dim3(256), 0, dla::stream(),
-                         part.data_ptr<float>(), sums.data_ptr<float>(), C, K);
+      constexpr int VM = 16 / (int)sizeof(dev_t);
+      auto stats = [&](auto vtag) {
+        constexpr int V = decltype(vtag)::value;
+        const int lanes = C / V;
+        const int64_t used = dla::nhwc_used_threads(lanes, rows, 262144);
+        const int lds = 2 * C * sizeof(float);
+        hipLaunchKernelGGL((dla::bn_stats_nhwc_kernel<dev_t, V>),
+                           dim3((int)((used + 255) / 256)), dim3(256), lds,
+                           dla::stream(), (const dev_t*)x.data_ptr(),
+                           sums.data_ptr<float>(), C, rows, used);
+      };
+      if (C % VM == 0) stats(std::integral_constant<int, VM>{});
+      else if (C % 2 == 0) stats(std::integral_constant<int, 2>{});
+      else stats(std::integral_constant<int, 1>{});
     } else {
       hipLaunchKernelGGL((dla::bn_stats_kernel<dev_t>), dim3(C, ysplit), dim3(256), 0,
                          dla::stream(), (const dev_t*)x.data_ptr(),
@@ -333,12 +380,18 @@ std::vector<torch::Tensor> batchnorm_fwd(torch::Tensor x, torch::Tensor weight,
     auto launch = [&](auto vtag, auto rtag) {
       constexpr int V = decltype(vtag)::value;
       constexpr bool R = decltype(rtag)::value;
-      const int grid = dla::grid_1d((n_total + V - 1) / V, 256);
-      if (nhwc)
-        hipLaunchKernelGGL((dla::bn_apply_nhwc_kernel<dev_t, V, R>), dim3(grid),
-                           dim3(256), 0, dla::stream(), (const dev_t*)x.data_ptr(),
+      if (nhwc) {
+        const int lanes = C / V;
+        const int64_t used = dla::nhwc_used_threads(lanes, rows, 524288);
+        hipLaunchKernelGGL((dla::bn_apply_nhwc_kernel<dev_t, V, R>),
+                           dim3((int)((used + 255) / 256)), dim3(256), 0,
+                           dla::stream(), (const dev_t*)x.data_ptr(),
                            scale.data_ptr<float>(), shift.data_ptr<float>(),
-                           (dev_t*)y.data_ptr(), C, n_total);
+                           (dev_t*)y.data_ptr(), C, rows, used);
+        return;
+      }
+      const int grid = dla::grid_1d((n_total + V - 1) / V, 256);
+      if (false) ;
       else
         hipLaunchKernelGGL((dla::bn_apply_kernel<dev_t, V, R>), dim3(grid), dim3(256),
                            0, dla::stream(), (const dev_t*)x.data_ptr(),
@@ -376,12 +429,19 @@ torch::Tensor bn_apply(torch::Tensor x, torch::Tensor scale, torch::Tensor shift
     auto launch = [&](auto vtag, auto rtag) {
       constexpr int V = decltype(vtag)::value;
       constexpr bool R = decltype(rtag)::value;
-      const int grid = dla::grid_1d((n_total + V - 1) / V, 256);
-      if (nhwc)
-        hipLaunchKernelGGL((dla::bn_apply_nhwc_kernel<dev_t, V, R>), dim3(grid),
-                           dim3(256), 0, dla::stream(), (const dev_t*)x.data_ptr(),
+      if (nhwc) {
+        const int64_t rows = n_total / C;
+        const int lanes = C / V;
+        const int64_t used = dla::nhwc_used_threads(lanes, rows, 524288);
+        hipLaunchKernelGGL((dla::bn_apply_nhwc_kernel<dev_t, V, R>),
+                           dim3((int)((used + 255) / 256)), dim3(256), 0,
+                           dla::stream(), (const dev_t*)x.data_ptr(),
                            sc.data_ptr<float>(), sh.data_ptr<float>(),
-                           (dev_t*)y.data_ptr(), C, n_total);
+                           (dev_t*)y.data_ptr(), C, rows, used);
+        return;
+      }
+      const int grid = dla::grid_1d((n_total + V - 1) / V, 256);
+      if (false) ;
       else
         hipLaunchKernelGGL((dla::bn_apply_kernel<dev_t, V, R>), dim3(grid), dim3(256),
                            0, dla::stream(), (const dev_t*)x.data_ptr(),
@@ -426,17 +486,22 @@ std::vector<torch::Tensor> batchnorm_bwd(torch::Tensor dy, torch::Tensor x,
     auto launch_stats = [&](auto rtag) {
       constexpr bool R = decltype(rtag)::value;
       if (nhwc) {
-        const int K = dla::nhwc_chunks(C, rows);
-        auto part = torch::empty({2LL * K * C}, opts_f);
-        const int64_t nthreads = (int64_t)K * C;
-        hipLaunchKernelGGL((dla::bn_bwd_stats_nhwc_partial_kernel<dev_t, R>),
-                           dim3((nthreads + 255) / 256), dim3(256), 0,
-                           dla::stream(), (const dev_t*)dy.data_ptr(),
-                           (const dev_t*)x.data_ptr(), yp, mean.data_ptr<float>(),
-                           rstd.data_ptr<float>(), part.data_ptr<float>(), C, K, rows);
-        hipLaunchKernelGGL(dla::bn_stats_nhwc_fold_kernel,
-                           dim3((C + 255) / 256), dim3(256), 0, dla::stream(),
-                           part.data_ptr<float>(), sums.data_ptr<float>(), C, K);
+        constexpr int VM = 16 / (int)sizeof(dev_t);
+        auto stats = [&](auto vtag) {
+          constexpr int V = decltype(vtag)::value;
+          const int lanes = C / V;
+          const int64_t used = dla::nhwc_used_threads(lanes, rows, 262144);
+          const int lds = 2 * C * sizeof(float);
+          hipLaunchKernelGGL((dla::bn_bwd_stats_nhwc_kernel<dev_t, V, R>),
+                             dim3((int)((used + 255) / 256)), dim3(256), lds,
+                             dla::stream(), (const dev_t*)dy.data_ptr(),
+                             (const dev_t*)x.data_ptr(), yp, mean.data_ptr<float>(),
+                             rstd.data_ptr<float>(), sums.data_ptr<float>(), C,
+                             rows, used);
+        };
+        if (C % VM == 0) stats(std::integral_constant<int, VM>{});
+        else if (C % 2 == 0) stats(std::integral_constant<int, 2>{});
+        else stats(std::integral_constant<int, 1>{});
       } else {
         hipLaunchKernelGGL((dla::bn_bwd_stats_kernel<dev_t, R>), dim3(C, ysplit),
                            dim3(256), 0, dla::stream(), (const dev_t*)dy.data_ptr(),
@@ -453,14 +518,16 @@ std::vector<torch::Tensor> batchnorm_bwd(torch::Tensor dy, torch::Tensor x,
     auto launch_dx = [&](auto vtag, auto rtag) {
       constexpr int V = decltype(vtag)::value;
       constexpr bool R = decltype(rtag)::value;
-      const int grid = dla::grid_1d((n_total + V - 1) / V, 256);
       if (nhwc) {
-        hipLaunchKernelGGL((dla::bn_bwd_dx_nhwc_kernel<dev_t, V, R>), dim3(grid),
-                           dim3(256), 0, dla::stream(), (const dev_t*)dy.data_ptr(),
+        const int lanes = C / V;
+        const int64_t used = dla::nhwc_used_threads(lanes, rows, 524288);
+        hipLaunchKernelGGL((dla::bn_bwd_dx_nhwc_kernel<dev_t, V, R>),
+                           dim3((int)((used + 255) / 256)), dim3(256), 0,
+                           dla::stream(), (const dev_t*)dy.data_ptr(),
                            (const dev_t*)x.data_ptr(), yp, mean.data_ptr<float>(),
                            rstd.data_ptr<float>(), w32.data_ptr<float>(),
                            sums.data_ptr<float>(), (dev_t*)dx.data_ptr(), C,
-                           n_total, inv_count);
+                           rows, used, inv_count);
       } else {
         hipLaunchKernelGGL((dla::bn_bwd_dx_kernel<dev_t, R>), dim3(dla::grid_1d(n_total, 256)),
                            dim3(256), 0, dla::stream(), (const dev_t*)dy.data_ptr(),

@@ -130,6 +130,138 @@ __global__ void ln_bwd_dwdb_kernel(const dev_t* __restrict__ dy,
   }
 }
 
+// ---- LDS-staged single-pass variants (C <= 4096) ---------------------------
+// fwd: read x once (staged in LDS as f32), one write. bwd: read dy,x once,
+// write dx, and accumulate dgamma/dbeta per block in LDS -> one atomicAdd per
+// channel per block (removes the separate dw/db tensor passes entirely).
+
+template <typename dev_t, int V>
+__global__ void ln_fwd_smem_kernel(const dev_t* __restrict__ x,
+                                   const dev_t* __restrict__ w,
+                                   const dev_t* __restrict__ b,
+                                   dev_t* __restrict__ y,
+                                   float* __restrict__ mean_out,
+                                   float* __restrict__ rstd_out,
+                                   int M, int C, float eps) {
+  extern __shared__ __attribute__((aligned(16))) float lds[];
+  float* wsh = lds;            // C
+  float* bsh = lds + C;        // C
+  float* row = lds + 2 * C;    // C
+  float* red = lds + 3 * C;    // 16
+  for (int i = threadIdx.x * V; i < C; i += blockDim.x * V) {
+    Vec<dev_t, V> wv = vload<dev_t, V>(w + i);
+    Vec<dev_t, V> bv = vload<dev_t, V>(b + i);
+#pragma unroll
+    for (int j = 0; j < V; ++j) {
+      wsh[i + j] = to_f32(wv.v[j]);
+      bsh[i + j] = to_f32(bv.v[j]);
+    }
+  }
+  __syncthreads();
+  for (int r = blockIdx.x; r < M; r += gridDim.x) {
+    const dev_t* xr = x + (int64_t)r * C;
+    dev_t* yr = y + (int64_t)r * C;
+    float sum = 0.f, sumsq = 0.f;
+    for (int i = threadIdx.x * V; i < C; i += blockDim.x * V) {
+      Vec<dev_t, V> xv = vload<dev_t, V>(xr + i);
+#pragma unroll
+      for (int j = 0; j < V; ++j) {
+        const float f = to_f32(xv.v[j]);
+        row[i + j] = f;
+        sum += f;
+        sumsq += f * f;
+      }
+    }
+    sum = block_reduce_sum(sum, red);
+    sumsq = block_reduce_sum(sumsq, red);
+    const float mu = sum / C;
+    const float rs = rsqrtf(fmaxf(sumsq / C - mu * mu, 0.f) + eps);
+    if (threadIdx.x == 0) {
+      mean_out[r] = mu;
+      rstd_out[r] = rs;
+    }
+    for (int i = threadIdx.x * V; i < C; i += blockDim.x * V) {
+      Vec<dev_t, V> yv;
+#pragma unroll
+      for (int j = 0; j < V; ++j)
+        yv.v[j] = from_f32<dev_t>((row[i + j] - mu) * rs * wsh[i + j] + bsh[i + j]);
+      vstore<dev_t, V>(yr + i, yv);
+    }
+    __syncthreads();  // row[] reused next iteration
+  }
+}
+
+template <typename dev_t, int V>
+__global__ void ln_bwd_smem_kernel(const dev_t* __restrict__ dy,
+                                   const dev_t* __restrict__ x,
+                                   const dev_t* __restrict__ w,
+                                   const float* __restrict__ mean,
+                                   const float* __restrict__ rstd,
+                                   dev_t* __restrict__ dx,
+                                   float* __restrict__ dw,
+                                   float* __restrict__ db, int M, int C) {
+  extern __shared__ __attribute__((aligned(16))) float lds[];
+  float* wsh = lds;            // C
+  float* dyrow = lds + C;      // C
+  float* xhrow = lds + 2 * C;  // C
+  float* dgs = lds + 3 * C;    // C (block dgamma)
+  float* dbs = lds + 4 * C;    // C (block dbeta)
+  float* red = lds + 5 * C;    // 16
+  for (int i = threadIdx.x * V; i < C; i += blockDim.x * V) {
+    Vec<dev_t, V> wv = vload<dev_t, V>(w + i);
+#pragma unroll
+    for (int j = 0; j < V; ++j) {
+      wsh[i + j] = to_f32(wv.v[j]);
+      dgs[i + j] = 0.f;
+      dbs[i + j] = 0.f;
+    }
+  }
+  __syncthreads();
+  for (int r = blockIdx.x; r < M; r += gridDim.x) {
+    const dev_t* dyr = dy + (int64_t)r * C;
+    const dev_t* xr = x + (int64_t)r * C;
+    dev_t* dxr = dx + (int64_t)r * C;
+    const float mu = mean[r], rs = rstd[r];
+    float s1 = 0.f, s2 = 0.f;
+    for (int i = threadIdx.x * V; i < C; i += blockDim.x * V) {
+      Vec<dev_t, V> dyv = vload<dev_t, V>(dyr + i);
+      Vec<dev_t, V> xv = vload<dev_t, V>(xr + i);
+#pragma unroll
+      for (int j = 0; j < V; ++j) {
+        const float g = to_f32(dyv.v[j]);
+        const float xh = (to_f32(xv.v[j]) - mu) * rs;
+        dyrow[i + j] = g;
+        xhrow[i + j] = xh;
+        const float gw = g * wsh[i + j];
+        s1 += gw;
+        s2 += gw * xh;
+      }
+    }
+    s1 = block_reduce_sum(s1, red) / C;
+    s2 = block_reduce_sum(s2, red) / C;
+    for (int i = threadIdx.x * V; i < C; i += blockDim.x * V) {
+      Vec<dev_t, V> dxv;
+#pragma unroll
+      for (int j = 0; j < V; ++j) {
+        const float g = dyrow[i + j];
+        const float xh = xhrow[i + j];
+        dxv.v[j] = from_f32<dev_t>(rs * (g * wsh[i + j] - s1 - xh * s2));
+        dgs[i + j] += g * xh;  // thread-exclusive slot: i+j depends on tid only
+        dbs[i + j] += g;
+      }
+      vstore<dev_t, V>(dxr + i, dxv);
+    }
+    __syncthreads();
+  }
+  for (int i = threadIdx.x * V; i < C; i += blockDim.x * V) {
+#pragma unroll
+    for (int j = 0; j < V; ++j) {
+      atomicAdd(&dw[i + j], dgs[i + j]);
+      atomicAdd(&db[i + j], dbs[i + j]);
+    }
+  }
+}
+
 }  // namespace dla
 
 std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor w,
@@ -146,11 +278,21 @@ std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor w,
     constexpr int VMAX = 16 / (int)sizeof(dev_t);
     auto launch = [&](auto vtag) {
       constexpr int V = decltype(vtag)::value;
-      hipLaunchKernelGGL((dla::ln_fwd_kernel<dev_t, V>), dim3(grid), dim3(block), 0,
-                         dla::stream(), (const dev_t*)x.data_ptr(),
-                         (const dev_t*)w.data_ptr(), (const dev_t*)b.data_ptr(),
-                         (dev_t*)y.data_ptr(), mean.data_ptr<float>(),
-                         rstd.data_ptr<float>(), (int)M, C, (float)eps);
+      if (C <= 4096) {
+        const int lds = (3 * C + 16) * sizeof(float);
+        hipLaunchKernelGGL((dla::ln_fwd_smem_kernel<dev_t, V>), dim3(grid),
+                           dim3(block), lds, dla::stream(),
+                           (const dev_t*)x.data_ptr(), (const dev_t*)w.data_ptr(),
+                           (const dev_t*)b.data_ptr(), (dev_t*)y.data_ptr(),
+                           mean.data_ptr<float>(), rstd.data_ptr<float>(), (int)M,
+                           C, (float)eps);
+      } else {
+        hipLaunchKernelGGL((dla::ln_fwd_kernel<dev_t, V>), dim3(grid), dim3(block), 0,
+                           dla::stream(), (const dev_t*)x.data_ptr(),
+                           (const dev_t*)w.data_ptr(), (const dev_t*)b.data_ptr(),
+                           (dev_t*)y.data_ptr(), mean.data_ptr<float>(),
+                           rstd.data_ptr<float>(), (int)M, C, (float)eps);
+      }
     };
     if (C % VMAX == 0) launch(std::integral_constant<int, VMAX>{});
     else if (C % 4 == 0) launch(std::integral_constant<int, 4>{});
@@ -176,21 +318,31 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
     constexpr int VMAX = 16 / (int)sizeof(dev_t);
     auto launch = [&](auto vtag) {
       constexpr int V = decltype(vtag)::value;
+      if (C <= 3072) {  // fused dx + dgamma/dbeta path (5C+16 floats of LDS)
+        const int lds = (5 * C + 16) * sizeof(float);
+        hipLaunchKernelGGL((dla::ln_bwd_smem_kernel<dev_t, V>), dim3(grid),
+                           dim3(block), lds, dla::stream(),
+                           (const dev_t*)dy.data_ptr(), (const dev_t*)x.data_ptr(),
+                           (const dev_t*)w.data_ptr(), mean.data_ptr<float>(),
+                           rstd.data_ptr<float>(), (dev_t*)dx.data_ptr(),
+                           dw.data_ptr<float>(), db.data_ptr<float>(), (int)M, C);
+        return;
+      }
       hipLaunchKernelGGL((dla::ln_bwd_dx_kernel<dev_t, V>), dim3(grid), dim3(block), 0,
                          dla::stream(), (const dev_t*)dy.data_ptr(),
                          (const dev_t*)x.data_ptr(), (const dev_t*)w.data_ptr(),
                          mean.data_ptr<float>(), rstd.data_ptr<float>(),
                          (dev_t*)dx.data_ptr(), (int)M, C);
+      dim3 g2((C + 255) / 256, ysplit);
+      hipLaunchKernelGGL((dla::ln_bwd_dwdb_kernel<dev_t>), g2, dim3(256), 0,
+                         dla::stream(), (const dev_t*)dy.data_ptr(),
+                         (const dev_t*)x.data_ptr(), mean.data_ptr<float>(),
+                         rstd.data_ptr<float>(), dw.data_ptr<float>(),
+                         db.data_ptr<float>(), (int)M, C);
     };
     if (C % VMAX == 0) launch(std::integral_constant<int, VMAX>{});
     else if (C % 4 == 0) launch(std::integral_constant<int, 4>{});
     else launch(std::integral_constant<int, 1>{});
-    dim3 g2((C + 255) / 256, ysplit);
-    hipLaunchKernelGGL((dla::ln_bwd_dwdb_kernel<dev_t>), g2, dim3(256), 0,
-                       dla::stream(), (const dev_t*)dy.data_ptr(),
-                       (const dev_t*)x.data_ptr(), mean.data_ptr<float>(),
-                       rstd.data_ptr<float>(), dw.data_ptr<float>(),
-                       db.data_ptr<float>(), (int)M, C);
   });
   HIP_CHECK_ERR();
   return {dx, dw.to(x.scalar_type()), db.to(x.scalar_type())};

@@ -88,13 +88,14 @@ def test_batchnorm_train(dtype, relu):
     if relu:
         yr = torch.relu(yr)
     torch.testing.assert_close(y.float(), yr, **_tol(dtype))
-    g = torch.randn_like(yr)
-    y.backward(g.to(dtype))
-    yr.backward(g)
+    g = torch.randn_like(yr).to(dtype)  # same (rounded) grad to both paths
+    y.backward(g)
+    yr.backward(g.float())
     torch.testing.assert_close(x.grad.float(), xr.grad, atol=2e-2 if dtype != torch.float32 else 1e-4,
                                rtol=2e-2 if dtype != torch.float32 else 1e-4)
+    rows = N * H * W
     torch.testing.assert_close(bn.weight.grad.float(), bn_ref.weight.grad,
-                               atol=5e-2, rtol=2e-2)
+                               atol=max(5e-2, 2e-3 * rows ** 0.5), rtol=2e-2)
     torch.testing.assert_close(bn.running_mean.float(), bn_ref.running_mean,
                                atol=1e-2, rtol=1e-2)
     torch.testing.assert_close(bn.running_var.float(), bn_ref.running_var,
