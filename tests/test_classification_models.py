@@ -19,8 +19,8 @@ CASES = [
     ("resnext50_32x4d", (1, 3, SMALL, SMALL)),
     ("wide_resnet50_2", (1, 3, SMALL, SMALL)),
     ("se_resnet50", (1, 3, SMALL, SMALL)),
-    ("sk_resnet50", (1, 3, SMALL, SMALL)),
-    ("resnest50", (1, 3, SMALL, SMALL)),
+    ("sk_resnet50", (2, 3, SMALL, SMALL)),
+    ("resnest50", (2, 3, SMALL, SMALL)),
     ("convnext_tiny", (1, 3, 224, 224)),
     ("repvgg_a0", (1, 3, SMALL, SMALL)),
     ("shufflenet_v1_g3", (1, 3, 224, 224)),
