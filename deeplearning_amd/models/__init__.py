@@ -1,6 +1,7 @@
 from .registry import build_model, list_models, register_model
 
 # importing submodules registers their factories
-from .classification import lenet, resnet, vit  # noqa: F401
+from . import classification  # noqa: F401,E402
+from . import segmentation  # noqa: F401,E402
 
 __all__ = ["build_model", "list_models", "register_model"]
