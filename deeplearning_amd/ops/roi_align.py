@@ -12,50 +12,99 @@ import torch.nn as nn
 from ._ext import ext, use_hip
 
 
-def _roi_align_eager(input, rois, output_size, spatial_scale, sampling_ratio, aligned):
-    """Pure-PyTorch RoIAlign (exact semantics, slow; CPU reference)."""
+def _roi_align_group(input, rois, output_size, spatial_scale, gh, gw, aligned,
+                     chunk=64):
+    """Vectorized RoIAlign for a group of rois sharing one (gh, gw) sampling
+    grid. Exact torchvision semantics (count = gh*gw, out-of-range samples
+    contribute 0, coords clamped into [0, dim-1])."""
+    PH, PW = output_size
+    N, C, H, W = input.shape
+    R = rois.shape[0]
+    off = 0.5 if aligned else 0.0
+    b_idx = rois[:, 0].long()
+    x1 = rois[:, 1] * spatial_scale - off
+    y1 = rois[:, 2] * spatial_scale - off
+    x2 = rois[:, 3] * spatial_scale - off
+    y2 = rois[:, 4] * spatial_scale - off
+    rw, rh = x2 - x1, y2 - y1
+    if not aligned:
+        rw, rh = rw.clamp(min=1.0), rh.clamp(min=1.0)
+    bh, bw = rh / PH, rw / PW
+
+    ph = torch.arange(PH, dtype=input.dtype, device=input.device)
+    pw = torch.arange(PW, dtype=input.dtype, device=input.device)
+    iy = torch.arange(gh, dtype=input.dtype, device=input.device)
+    ix = torch.arange(gw, dtype=input.dtype, device=input.device)
+    # ys: [R, PH, gh], xs: [R, PW, gw]
+    ys = y1[:, None, None] + ph[None, :, None] * bh[:, None, None] + \
+        (iy[None, None, :] + 0.5) * (bh[:, None, None] / gh)
+    xs = x1[:, None, None] + pw[None, :, None] * bw[:, None, None] + \
+        (ix[None, None, :] + 0.5) * (bw[:, None, None] / gw)
+
+    out = input.new_empty((R, C, PH, PW))
+    flat = input.reshape(N, C, H * W)
+    S = PH * gh * PW * gw
+    for s in range(0, R, chunk):
+        e = min(s + chunk, R)
+        y = ys[s:e].reshape(e - s, PH * gh, 1, 1)
+        x = xs[s:e].reshape(e - s, 1, 1, PW * gw)
+        y = y.expand(e - s, PH * gh, 1, PW * gw)
+        x = x.expand(e - s, PH * gh, 1, PW * gw)
+        valid = (y >= -1.0) & (y <= H) & (x >= -1.0) & (x <= W)
+        yy = y.clamp(min=0.0)
+        xx = x.clamp(min=0.0)
+        y0 = yy.floor().long().clamp(max=H - 1)
+        x0 = xx.floor().long().clamp(max=W - 1)
+        hi_y = (y0 >= H - 1)
+        hi_x = (x0 >= W - 1)
+        y1i = (y0 + 1).clamp(max=H - 1)
+        x1i = (x0 + 1).clamp(max=W - 1)
+        yy = torch.where(hi_y, y0.to(yy.dtype), yy)
+        xx = torch.where(hi_x, x0.to(xx.dtype), xx)
+        ly, lx = yy - y0, xx - x0
+        hy, hx = 1 - ly, 1 - lx
+        w00 = (hy * hx * valid).reshape(e - s, 1, -1)
+        w01 = (hy * lx * valid).reshape(e - s, 1, -1)
+        w10 = (ly * hx * valid).reshape(e - s, 1, -1)
+        w11 = (ly * lx * valid).reshape(e - s, 1, -1)
+        i00 = (y0 * W + x0).reshape(e - s, 1, -1).expand(-1, C, -1)
+        i01 = (y0 * W + x1i).reshape(e - s, 1, -1).expand(-1, C, -1)
+        i10 = (y1i * W + x0).reshape(e - s, 1, -1).expand(-1, C, -1)
+        i11 = (y1i * W + x1i).reshape(e - s, 1, -1).expand(-1, C, -1)
+        src = flat[b_idx[s:e]]  # [r, C, H*W]
+        acc = w00 * src.gather(2, i00) + w01 * src.gather(2, i01) + \
+            w10 * src.gather(2, i10) + w11 * src.gather(2, i11)
+        acc = acc.reshape(e - s, C, PH, gh, PW, gw).sum(dim=(3, 5))
+        out[s:e] = acc / (gh * gw)
+    return out
+
+
+def _roi_align_eager(input, rois, output_size, spatial_scale, sampling_ratio,
+                     aligned):
+    """Pure-PyTorch RoIAlign (exact semantics; CPU reference). Vectorized per
+    (gh, gw) sampling-grid group."""
     PH, PW = output_size
     R = rois.shape[0]
-    N, C, H, W = input.shape
-    out = input.new_zeros((R, C, PH, PW))
+    if R == 0:
+        return input.new_zeros((0, input.shape[1], PH, PW))
     off = 0.5 if aligned else 0.0
-    for r in range(R):
-        b = int(rois[r, 0].item())
-        x1 = rois[r, 1].item() * spatial_scale - off
-        y1 = rois[r, 2].item() * spatial_scale - off
-        x2 = rois[r, 3].item() * spatial_scale - off
-        y2 = rois[r, 4].item() * spatial_scale - off
-        rw, rh = x2 - x1, y2 - y1
-        if not aligned:
-            rw, rh = max(rw, 1.0), max(rh, 1.0)
-        bh, bw = rh / PH, rw / PW
-        gh = sampling_ratio if sampling_ratio > 0 else max(1, int(torch.tensor(rh / PH).ceil()))
-        gw = sampling_ratio if sampling_ratio > 0 else max(1, int(torch.tensor(rw / PW).ceil()))
-        for ph in range(PH):
-            for pw in range(PW):
-                acc = input.new_zeros(C)
-                cnt = 0
-                for iy in range(gh):
-                    y = y1 + ph * bh + (iy + 0.5) * bh / gh
-                    for ix in range(gw):
-                        x = x1 + pw * bw + (ix + 0.5) * bw / gw
-                        cnt += 1
-                        if y < -1.0 or y > H or x < -1.0 or x > W:
-                            continue
-                        yy, xx = max(y, 0.0), max(x, 0.0)
-                        y0, x0 = int(yy), int(xx)
-                        y1i, x1i = y0 + 1, x0 + 1
-                        if y0 >= H - 1:
-                            y0 = y1i = H - 1
-                            yy = float(y0)
-                        if x0 >= W - 1:
-                            x0 = x1i = W - 1
-                            xx = float(x0)
-                        ly, lx = yy - y0, xx - x0
-                        hy, hx = 1 - ly, 1 - lx
-                        acc += (hy * hx * input[b, :, y0, x0] + hy * lx * input[b, :, y0, x1i]
-                                + ly * hx * input[b, :, y1i, x0] + ly * lx * input[b, :, y1i, x1i])
-                out[r, :, ph, pw] = acc / max(cnt, 1)
+    rh = (rois[:, 4] - rois[:, 2]) * spatial_scale
+    rw = (rois[:, 3] - rois[:, 1]) * spatial_scale
+    if not aligned:
+        rh, rw = rh.clamp(min=1.0), rw.clamp(min=1.0)
+    if sampling_ratio > 0:
+        gh = torch.full((R,), sampling_ratio, dtype=torch.long)
+        gw = gh
+    else:
+        gh = (rh / PH).ceil().long().clamp(min=1)
+        gw = (rw / PW).ceil().long().clamp(min=1)
+    out = input.new_zeros((R, input.shape[1], PH, PW))
+    key = gh * 10000 + gw
+    for k in key.unique():
+        idx = torch.where(key == k)[0]
+        out[idx] = _roi_align_group(
+            input, rois[idx], output_size, spatial_scale,
+            int(gh[idx[0]]), int(gw[idx[0]]), aligned)
     return out
 
 
