@@ -1,3 +1,11 @@
-from .synthetic import DeviceBatchLoader, SyntheticClassification
-
-__all__ = ["SyntheticClassification", "DeviceBatchLoader"]
+from .datasets import (COCODetectionDataset, ClassificationDataset,  # noqa: F401
+                       SegmentationDataset, VOCDetectionDataset, mosaic4,
+                       read_split_data)
+from .prefetcher import DataPrefetcher  # noqa: F401
+from .samplers import (GroupedBatchSampler, InfiniteSampler,  # noqa: F401
+                       SubsetRandomSampler, YoloBatchSampler,
+                       create_aspect_ratio_groups)
+from .synthetic import DeviceBatchLoader, SyntheticClassification  # noqa: F401
+from .transforms import (Compose, Mixup, Normalize,  # noqa: F401
+                         RandomResizedCrop, classification_eval_transform,
+                         classification_train_transform)

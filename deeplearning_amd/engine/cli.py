@@ -1,0 +1,256 @@
+"""Shared CLI drivers for the per-subproject train.py / predict.py scripts
+under projects/.
+
+Each reference subproject ships its own full train loop;
+here every projects/*/train.py is a thin wrapper over these drivers, keeping
+the reference CLI surface (--data-path, --epochs, --batch-size, --lr,
+--device, --weights, --resume, --amp ...) and the reference checkpoint layout
+runs/<name>/weights/model_{e}.pth + best_model.pth
+(ref classification/mnist/train.py:141-186, others/train_with_DDP/train.py).
+"""
+from __future__ import annotations
+
+import argparse
+import time
+from pathlib import Path
+
+import torch
+from torch.utils.data import DataLoader, DistributedSampler
+
+from ..core.checkpoint import (load_checkpoint, load_pretrained,
+                               save_checkpoint, save_weights)
+from ..core.dist import (cleanup, get_rank, get_world_size, init_distributed,
+                         is_main_process)
+from ..core.env import increment_path, seed_everything, select_device
+from ..core.logging import create_logger
+from ..core.meters import AverageMeter
+from ..core.tensorboard import SummaryWriter
+from ..data import (ClassificationDataset, SyntheticClassification,
+                    classification_eval_transform,
+                    classification_train_transform, read_split_data)
+from ..engine.metrics import accuracy
+from ..engine.scheduler import WarmupScheduler
+from ..models import build_model
+from ..ops import cross_entropy
+from ..parallel import wrap_data_parallel
+
+
+def classification_argparser(default_model: str, **defaults):
+    p = argparse.ArgumentParser()
+    p.add_argument("--model", default=default_model)
+    p.add_argument("--data-path", default="", help="class-per-folder root; "
+                   "empty = synthetic data")
+    p.add_argument("--num-classes", type=int,
+                   default=defaults.get("num_classes", 1000))
+    p.add_argument("--img-size", type=int,
+                   default=defaults.get("img_size", 224))
+    p.add_argument("--in-channels", type=int,
+                   default=defaults.get("in_channels", 3))
+    p.add_argument("--epochs", type=int, default=defaults.get("epochs", 10))
+    p.add_argument("--batch-size", type=int,
+                   default=defaults.get("batch_size", 32))
+    p.add_argument("--lr", type=float, default=defaults.get("lr", 0.01))
+    p.add_argument("--weight-decay", type=float,
+                   default=defaults.get("weight_decay", 5e-4))
+    p.add_argument("--optimizer", default=defaults.get("optimizer", "sgd"),
+                   choices=["sgd", "adamw"])
+    p.add_argument("--warmup-epochs", type=int, default=1)
+    p.add_argument("--workers", type=int, default=4)
+    p.add_argument("--device", default="cuda")
+    p.add_argument("--weights", default="", help="pretrained weights")
+    p.add_argument("--resume", default="", help="checkpoint to resume, or "
+                   "'auto'")
+    p.add_argument("--amp", action="store_true", default=True)
+    p.add_argument("--no-amp", dest="amp", action="store_false")
+    p.add_argument("--seed", type=int, default=0)
+    p.add_argument("--output", default="runs")
+    p.add_argument("--name", default=defaults.get("name", "exp"))
+    p.add_argument("--syncbn", action="store_true")
+    p.add_argument("--accumulate-steps", type=int, default=1)
+    p.add_argument("--clip-grad", type=float, default=0.0)
+    p.add_argument("--synthetic-size", type=int, default=256,
+                   help="synthetic dataset length when --data-path is empty")
+    return p
+
+
+def build_classification_loaders(args):
+    if args.data_path:
+        tp, tl, vp, vl, classes = read_split_data(args.data_path)
+        train_ds = ClassificationDataset(
+            tp, tl, classification_train_transform(args.img_size))
+        val_ds = ClassificationDataset(
+            vp, vl, classification_eval_transform(args.img_size))
+    else:
+        c = getattr(args, "in_channels", 3)
+        train_ds = SyntheticClassification(
+            args.synthetic_size, (c, args.img_size, args.img_size),
+            args.num_classes)
+        val_ds = SyntheticClassification(
+            max(args.synthetic_size // 4, 8),
+            (c, args.img_size, args.img_size), args.num_classes)
+    train_sampler = DistributedSampler(train_ds) if get_world_size() > 1 \
+        else None
+    train_loader = DataLoader(
+        train_ds, batch_size=args.batch_size,
+        shuffle=train_sampler is None, sampler=train_sampler,
+        num_workers=args.workers, pin_memory=torch.cuda.is_available(),
+        drop_last=True)
+    val_loader = DataLoader(val_ds, batch_size=args.batch_size,
+                            shuffle=False, num_workers=args.workers)
+    return train_loader, val_loader, train_sampler
+
+
+def classification_train_main(args) -> dict:
+    info = init_distributed()
+    if torch.cuda.is_available() and get_world_size() > 1:
+        device = torch.device("cuda", info["local_rank"])
+        torch.cuda.set_device(device)
+    else:
+        device = select_device(args.device)
+    seed_everything(args.seed, rank=get_rank())
+
+    run_dir = Path(increment_path(Path(args.output) / args.name,
+                                  exist_ok=False)) \
+        if is_main_process() else Path(args.output) / args.name
+    weights_dir = run_dir / "weights"
+    logger = create_logger(str(run_dir) if is_main_process() else None,
+                           dist_rank=get_rank())
+    writer = SummaryWriter(str(run_dir)) if is_main_process() else None
+
+    model = build_model(args.model, num_classes=args.num_classes).to(device)
+    if args.weights:
+        load_pretrained(model, args.weights, logger=logger)
+    if args.syncbn and get_world_size() > 1:
+        from ..parallel.syncbn import convert_sync_batchnorm
+        model = convert_sync_batchnorm(model)
+    if get_world_size() > 1:
+        model = wrap_data_parallel(model)
+
+    params = [p for p in model.parameters() if p.requires_grad]
+    if args.optimizer == "adamw":
+        optimizer = torch.optim.AdamW(params, lr=args.lr,
+                                      weight_decay=args.weight_decay)
+    else:
+        optimizer = torch.optim.SGD(params, lr=args.lr, momentum=0.9,
+                                    weight_decay=args.weight_decay)
+
+    train_loader, val_loader, train_sampler = \
+        build_classification_loaders(args)
+    steps_per_epoch = len(train_loader)
+    scheduler = WarmupScheduler(
+        optimizer, total_steps=args.epochs * steps_per_epoch,
+        warmup_steps=args.warmup_epochs * steps_per_epoch)
+
+    start_epoch = 0
+    best_acc = 0.0
+    resume = args.resume
+    if resume == "auto":
+        from ..core.checkpoint import auto_resume_helper
+        resume = auto_resume_helper(weights_dir) or ""
+    if resume:
+        ckpt = load_checkpoint(resume, model, optimizer, scheduler)
+        start_epoch = ckpt.get("epoch", -1) + 1
+        best_acc = ckpt.get("max_accuracy", 0.0)
+        logger.info(f"resumed from {resume} at epoch {start_epoch}")
+
+    amp = args.amp and device.type == "cuda"
+    for epoch in range(start_epoch, args.epochs):
+        if train_sampler is not None:
+            train_sampler.set_epoch(epoch)
+        model.train()
+        loss_m, acc_m = AverageMeter(), AverageMeter()
+        t0 = time.time()
+        optimizer.zero_grad(set_to_none=True)
+        for it, (x, y) in enumerate(train_loader):
+            x = x.to(device, non_blocking=True)
+            y = y.to(device, non_blocking=True)
+            with torch.autocast(device.type, dtype=torch.bfloat16,
+                                enabled=amp):
+                out = model(x)
+                logits = out[0] if isinstance(out, tuple) else out
+                loss = cross_entropy(logits, y)
+                if isinstance(out, tuple):  # aux heads (GoogLeNet)
+                    for aux in out[1:]:
+                        if aux is not None:
+                            loss = loss + 0.3 * cross_entropy(aux, y)
+            (loss / args.accumulate_steps).backward()
+            if (it + 1) % args.accumulate_steps == 0:
+                if args.clip_grad > 0:
+                    torch.nn.utils.clip_grad_norm_(params, args.clip_grad)
+                finalize = getattr(model, "finalize", None)
+                if finalize is not None:
+                    finalize()
+                optimizer.step()
+                optimizer.zero_grad(set_to_none=True)
+                scheduler.step()
+            with torch.no_grad():
+                acc1 = accuracy(logits.float(), y)[0]
+            loss_m.update(float(loss.detach()), x.shape[0])
+            acc_m.update(float(acc1), x.shape[0])
+        logger.info(f"epoch {epoch}: loss {loss_m.avg:.4f} "
+                    f"acc {acc_m.avg:.2f} ({time.time() - t0:.1f}s)")
+
+        # eval + checkpoint (rank 0)
+        val_acc = evaluate_classification(model, val_loader, device, amp)
+        if is_main_process():
+            writer.add_scalar("train/loss", loss_m.avg, epoch)
+            writer.add_scalar("val/acc1", val_acc, epoch)
+            save_weights(model, weights_dir / f"model_{epoch}.pth")
+            save_checkpoint(weights_dir / f"ckpt_epoch_{epoch}.pth", model,
+                            optimizer, scheduler, epoch,
+                            max_accuracy=best_acc)
+            if val_acc >= best_acc:
+                best_acc = val_acc
+                save_weights(model, weights_dir / "best_model.pth")
+        logger.info(f"epoch {epoch}: val acc1 {val_acc:.2f} "
+                    f"(best {max(best_acc, val_acc):.2f})")
+    if writer:
+        writer.close()
+    cleanup()
+    return {"best_acc": best_acc, "run_dir": str(run_dir)}
+
+
+@torch.no_grad()
+def evaluate_classification(model, loader, device, amp=True) -> float:
+    model.eval()
+    correct = torch.zeros(2, device=device)
+    for x, y in loader:
+        x = x.to(device, non_blocking=True)
+        y = y.to(device, non_blocking=True)
+        with torch.autocast(device.type, dtype=torch.bfloat16, enabled=amp):
+            logits = model(x)
+        pred = logits.float().argmax(1)
+        correct[0] += (pred == y).sum()
+        correct[1] += y.numel()
+    from ..core.dist import reduce_value
+    correct = reduce_value(correct, average=False)
+    return float(correct[0] / correct[1].clamp(min=1) * 100)
+
+
+def predict_main(default_model: str, num_classes: int = 1000,
+                 img_size: int = 224):
+    """Single-image predict CLI (ref classification/*/predict.py:12-59)."""
+    p = argparse.ArgumentParser()
+    p.add_argument("image", help="path to image")
+    p.add_argument("--model", default=default_model)
+    p.add_argument("--weights", required=True)
+    p.add_argument("--num-classes", type=int, default=num_classes)
+    p.add_argument("--img-size", type=int, default=img_size)
+    p.add_argument("--device", default="cuda")
+    p.add_argument("--topk", type=int, default=5)
+    args = p.parse_args()
+
+    device = select_device(args.device)
+    model = build_model(args.model, num_classes=args.num_classes).to(device)
+    load_pretrained(model, args.weights)
+    model.eval()
+
+    from PIL import Image
+    img = Image.open(args.image).convert("RGB")
+    x = classification_eval_transform(args.img_size)(img)[None].to(device)
+    with torch.no_grad():
+        prob = model(x).softmax(1)[0]
+    topk = prob.topk(min(args.topk, prob.numel()))
+    for score, idx in zip(topk.values.tolist(), topk.indices.tolist()):
+        print(f"class {idx}: {score:.4f}")
+    return topk

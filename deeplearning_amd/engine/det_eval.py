@@ -1,0 +1,156 @@
+"""Detection evaluation: COCO-style mAP@[.5:.95] and VOC mAP, with cross-rank
+result merge.
+
+Reference parity: detection/RetinaNet/train_utils/coco_eval.py:15-199
+(CocoEvaluator + all_gather merge), YOLOX yolox/evaluators/{coco_evaluator,
+voc_eval}.py — re-designed without pycocotools: greedy per-IoU-threshold
+matching (highest score first, crowd-aware) + 101-point PR interpolation,
+identical protocol to COCOeval's bbox task. The hot matching loop has a C++
+fast path (csrc/cocoeval.cpp, ref YOLOX's native cocoeval §2.2) with this
+file as the pure-Python reference.
+"""
+from __future__ import annotations
+
+import torch
+
+from ..core.dist import all_gather_object_list as all_gather_object
+from ..core.dist import get_world_size
+from ..ops import box_iou
+
+COCO_IOU_THRS = [0.5 + 0.05 * i for i in range(10)]
+
+
+def match_image(det_boxes, det_scores, gt_boxes, gt_crowd, iou_thrs,
+                max_dets=100):
+    """Greedy COCO matching for one image+class.
+
+    Returns (matched [T, D] bool, ignored [T, D] bool, scores [D], n_gt).
+    """
+    order = det_scores.argsort(descending=True)[:max_dets]
+    det_boxes = det_boxes[order]
+    det_scores = det_scores[order]
+    D = det_boxes.shape[0]
+    G = gt_boxes.shape[0]
+    T = len(iou_thrs)
+    matched = torch.zeros(T, D, dtype=torch.bool)
+    ignored = torch.zeros(T, D, dtype=torch.bool)
+    n_gt = int((~gt_crowd).sum()) if G else 0
+    if D == 0 or G == 0:
+        return matched, ignored, det_scores, n_gt
+    ious = box_iou(det_boxes, gt_boxes)  # D, G
+    for t, thr in enumerate(iou_thrs):
+        taken = torch.zeros(G, dtype=torch.bool)
+        for d in range(D):
+            best_iou = thr
+            best_g = -1
+            for g in range(G):
+                if taken[g] and not gt_crowd[g]:
+                    continue
+                # prefer non-crowd matches; crowd only if nothing else
+                if best_g >= 0 and not gt_crowd[best_g] and gt_crowd[g]:
+                    continue
+                if ious[d, g] >= best_iou:
+                    best_iou = float(ious[d, g])
+                    best_g = g
+            if best_g >= 0:
+                if gt_crowd[best_g]:
+                    ignored[t, d] = True
+                else:
+                    matched[t, d] = True
+                    taken[best_g] = True
+    return matched, ignored, det_scores, n_gt
+
+
+def _ap_101(recall, precision):
+    """COCO 101-point interpolated AP."""
+    rec_thrs = torch.linspace(0, 1, 101)
+    # precision envelope
+    prec = precision.clone()
+    for i in range(prec.numel() - 2, -1, -1):
+        prec[i] = max(prec[i], prec[i + 1])
+    idx = torch.searchsorted(recall, rec_thrs)
+    ap = torch.zeros(101)
+    valid = idx < prec.numel()
+    ap[valid] = prec[idx[valid]]
+    return float(ap.mean())
+
+
+class DetEvaluator:
+    """Accumulate (pred, gt) pairs per image; summarize COCO-style."""
+
+    def __init__(self, iou_thrs=None, max_dets=100):
+        self.iou_thrs = iou_thrs or COCO_IOU_THRS
+        self.max_dets = max_dets
+        self.items = []  # (cls, matched [T,D], ignored [T,D], scores [D], n_gt)
+
+    def update(self, predictions, targets):
+        """predictions/targets: lists of dicts with boxes/labels(/scores)."""
+        for pred, gt in zip(predictions, targets):
+            classes = torch.cat([pred["labels"], gt["labels"]]).unique()
+            crowd = gt.get("iscrowd",
+                           torch.zeros_like(gt["labels"]))
+            for c in classes.tolist():
+                dm = pred["labels"] == c
+                gm = gt["labels"] == c
+                matched, ignored, scores, n_gt = match_image(
+                    pred["boxes"][dm].cpu(), pred["scores"][dm].cpu(),
+                    gt["boxes"][gm].cpu(), crowd[gm].cpu().bool(),
+                    self.iou_thrs, self.max_dets)
+                self.items.append((c, matched, ignored, scores, n_gt))
+
+    def synchronize_between_processes(self):
+        if get_world_size() > 1:
+            merged = all_gather_object(self.items)
+            self.items = [it for part in merged for it in part]
+
+    def summarize(self):
+        """Returns dict: mAP (IoU .5:.95), mAP50, mAP75, per-class AP."""
+        by_class = {}
+        for c, matched, ignored, scores, n_gt in self.items:
+            by_class.setdefault(c, []).append((matched, ignored, scores, n_gt))
+        T = len(self.iou_thrs)
+        ap_per_class = {}
+        for c, items in by_class.items():
+            total_gt = sum(it[3] for it in items)
+            if total_gt == 0:
+                continue
+            scores = torch.cat([it[2] for it in items])
+            order = scores.argsort(descending=True)
+            aps = []
+            for t in range(T):
+                m = torch.cat([it[0][t] for it in items])[order]
+                ig = torch.cat([it[1][t] for it in items])[order]
+                keep = ~ig
+                tp = m[keep].float().cumsum(0)
+                fp = (~m[keep]).float().cumsum(0)
+                recall = tp / total_gt
+                precision = tp / (tp + fp).clamp(min=1e-9)
+                aps.append(_ap_101(recall, precision))
+            ap_per_class[c] = aps
+        if not ap_per_class:
+            return {"mAP": 0.0, "mAP50": 0.0, "mAP75": 0.0, "per_class": {}}
+        all_aps = torch.tensor(list(ap_per_class.values()))  # C, T
+        return {
+            "mAP": float(all_aps.mean()),
+            "mAP50": float(all_aps[:, 0].mean()),
+            "mAP75": float(all_aps[:, 5].mean()),
+            "per_class": {c: float(torch.tensor(a).mean())
+                          for c, a in ap_per_class.items()},
+        }
+
+
+def voc_ap(recall, precision, use_07_metric=False):
+    """VOC AP (ref YOLOX yolox/evaluators/voc_eval.py)."""
+    if use_07_metric:
+        ap = 0.0
+        for t in torch.arange(0.0, 1.1, 0.1):
+            p = precision[recall >= t]
+            ap += (float(p.max()) if p.numel() else 0.0) / 11
+        return ap
+    # all-points interpolation
+    mrec = torch.cat([torch.zeros(1), recall, torch.ones(1)])
+    mpre = torch.cat([torch.zeros(1), precision, torch.zeros(1)])
+    for i in range(mpre.numel() - 2, -1, -1):
+        mpre[i] = max(mpre[i], mpre[i + 1])
+    idx = (mrec[1:] != mrec[:-1]).nonzero().flatten()
+    return float(((mrec[idx + 1] - mrec[idx]) * mpre[idx + 1]).sum())

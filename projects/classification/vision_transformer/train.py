@@ -1,0 +1,16 @@
+#!/usr/bin/env python3
+"""Train vision_transformer (reference: classification/vision_transformer/train.py, same CLI surface)
+on the shared MI355X engine."""
+import sys
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parents[3]))
+
+from deeplearning_amd.engine.cli import (classification_argparser,
+                                         classification_train_main)
+
+if __name__ == "__main__":
+    args = classification_argparser(
+        "vit_b16", num_classes=1000, img_size=224, name="vision_transformer", optimizer='adamw', lr=0.001
+    ).parse_args()
+    classification_train_main(args)

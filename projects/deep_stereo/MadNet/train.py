@@ -1,0 +1,48 @@
+#!/usr/bin/env python3
+"""MADNet stereo training + MAD online adaptation (reference: deep_stereo/MadNet/Stereo_Online_Adaptation.py)."""
+import sys
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parents[3]))
+
+import argparse
+import time
+
+import torch
+
+from deeplearning_amd.core.env import seed_everything, select_device
+from deeplearning_amd.models import build_model
+from deeplearning_amd.models.stereo import MADAdapter, reprojection_loss
+
+if __name__ == "__main__":
+    p = argparse.ArgumentParser()
+    p.add_argument("--steps", type=int, default=10)
+    p.add_argument("--mode", default="mad", choices=["full", "mad"],
+                   help="full backprop or modular adaptation")
+    p.add_argument("--height", type=int, default=128)
+    p.add_argument("--width", type=int, default=256)
+    p.add_argument("--lr", type=float, default=1e-4)
+    p.add_argument("--device", default="cuda")
+    args = p.parse_args()
+
+    seed_everything(0)
+    device = select_device(args.device)
+    model = build_model("madnet").to(device)
+    if args.mode == "mad":
+        adapter = MADAdapter(model, lr=args.lr)
+        for s in range(args.steps):
+            left = torch.rand(1, 3, args.height, args.width, device=device)
+            right = torch.roll(left, shifts=4, dims=3)
+            loss = adapter.step(left, right)
+            print(f"step {s}: photometric loss {loss:.4f}")
+    else:
+        opt = torch.optim.Adam(model.parameters(), lr=args.lr)
+        for s in range(args.steps):
+            left = torch.rand(1, 3, args.height, args.width, device=device)
+            right = torch.roll(left, shifts=4, dims=3)
+            disp, _ = model(left, right)
+            loss = reprojection_loss(left, right, disp)
+            opt.zero_grad(set_to_none=True)
+            loss.backward()
+            opt.step()
+            print(f"step {s}: photometric loss {float(loss):.4f}")

@@ -1,0 +1,87 @@
+#!/usr/bin/env python3
+"""Train YOLOX (reference: detection/YOLOX/tools/train.py — Trainer/Exp semantics on the shared engine; SimOTA on-GPU)."""
+import sys
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parents[3]))
+
+import argparse
+import time
+
+import torch
+from torch.utils.data import DataLoader
+
+from deeplearning_amd.core.checkpoint import save_checkpoint, save_weights
+from deeplearning_amd.core.dist import (cleanup, get_rank, get_world_size,
+                                        init_distributed, is_main_process)
+from deeplearning_amd.core.env import (increment_path, seed_everything,
+                                       select_device)
+from deeplearning_amd.core.logging import create_logger
+from deeplearning_amd.engine.cli_det import SyntheticDetection
+from deeplearning_amd.models import build_model
+from deeplearning_amd.ops import ModelEMA
+from deeplearning_amd.parallel.syncbn import all_reduce_norm
+
+if __name__ == "__main__":
+    p = argparse.ArgumentParser()
+    p.add_argument("--model", default="yolox_s",
+                   choices=["yolox_s", "yolox_m", "yolox_l", "yolox_x"])
+    p.add_argument("--num-classes", type=int, default=80)
+    p.add_argument("--img-size", type=int, default=256)
+    p.add_argument("--epochs", type=int, default=3)
+    p.add_argument("--batch-size", type=int, default=4)
+    p.add_argument("--lr", type=float, default=0.01)
+    p.add_argument("--device", default="cuda")
+    p.add_argument("--output", default="runs")
+    p.add_argument("--synthetic-size", type=int, default=16)
+    args = p.parse_args()
+
+    init_distributed()
+    seed_everything(0, rank=get_rank())
+    device = select_device(args.device)
+    run_dir = increment_path(Path(args.output) / "yolox")
+    logger = create_logger(str(run_dir) if is_main_process() else None,
+                           dist_rank=get_rank())
+
+    model = build_model(args.model, num_classes=args.num_classes).to(device)
+    if get_world_size() > 1:
+        from deeplearning_amd.parallel import wrap_data_parallel
+        model = wrap_data_parallel(model)
+    ema = ModelEMA(model)
+    ds = SyntheticDetection(args.synthetic_size,
+                            (3, args.img_size, args.img_size),
+                            args.num_classes)
+    loader = DataLoader(ds, batch_size=args.batch_size, shuffle=True,
+                        collate_fn=SyntheticDetection.collate_fn)
+    opt = torch.optim.SGD(model.parameters(), lr=args.lr, momentum=0.9,
+                          weight_decay=5e-4, nesterov=True)
+    amp = device.type == "cuda"
+    for epoch in range(args.epochs):
+        model.train()
+        t0 = time.time()
+        tot = 0.0
+        for images, targets in loader:
+            x = torch.stack(list(images)).to(device)
+            targets = [{k: v.to(device) for k, v in t.items()}
+                       for t in targets]
+            with torch.autocast(device.type, dtype=torch.bfloat16,
+                                enabled=amp):
+                losses = model(x, targets)
+                loss = sum(losses.values())
+            opt.zero_grad(set_to_none=True)
+            loss.backward()
+            finalize = getattr(model, "finalize", None)
+            if finalize is not None:
+                finalize()
+            opt.step()
+            ema.update(model)
+            tot += float(loss.detach())
+        if get_world_size() > 1:
+            all_reduce_norm(model)  # lazy BN sync before eval (YOLOX style)
+        logger.info(f"epoch {epoch}: loss {tot / len(loader):.4f} "
+                    f"({time.time() - t0:.1f}s)")
+        if is_main_process():
+            save_weights(ema.ema, run_dir / "weights" / f"model_{epoch}.pth")
+            save_checkpoint(run_dir / "weights" / "latest_ckpt.pth", model,
+                            opt, epoch=epoch)
+    cleanup()
