@@ -1,0 +1,157 @@
+"""Tests: data layer (datasets/transforms/samplers/mosaic), detection mAP
+evaluator, retrieval eval, checkpoint layout of the CLI drivers."""
+import json
+import os
+import subprocess
+import sys
+import tempfile
+from pathlib import Path
+
+import numpy as np
+import pytest
+import torch
+from PIL import Image
+
+REPO = Path(__file__).resolve().parents[1]
+
+
+def test_read_split_and_dataset():
+    from deeplearning_amd.data import (ClassificationDataset,
+                                       classification_train_transform,
+                                       read_split_data)
+    with tempfile.TemporaryDirectory() as d:
+        for c in ["a", "b"]:
+            os.makedirs(f"{d}/{c}")
+            for i in range(10):
+                Image.fromarray(np.random.randint(
+                    0, 255, (40, 50, 3), dtype=np.uint8)).save(
+                        f"{d}/{c}/{i}.jpg")
+        tp, tl, vp, vl, classes = read_split_data(d, val_rate=0.2)
+        assert classes == ["a", "b"]
+        assert len(tp) == 16 and len(vp) == 4
+        ds = ClassificationDataset(tp, tl,
+                                   classification_train_transform(32))
+        x, y = ds[0]
+        assert x.shape == (3, 32, 32)
+
+
+def test_voc_dataset_parsing():
+    from deeplearning_amd.data import VOCDetectionDataset
+    with tempfile.TemporaryDirectory() as d:
+        root = Path(d)
+        (root / "JPEGImages").mkdir()
+        (root / "Annotations").mkdir()
+        (root / "ImageSets" / "Main").mkdir(parents=True)
+        Image.fromarray(np.zeros((80, 100, 3), dtype=np.uint8)).save(
+            root / "JPEGImages" / "im0.jpg")
+        (root / "Annotations" / "im0.xml").write_text(
+            "<annotation><object><name>cat</name><bndbox>"
+            "<xmin>10</xmin><ymin>20</ymin><xmax>50</xmax><ymax>60</ymax>"
+            "</bndbox></object></annotation>")
+        (root / "ImageSets" / "Main" / "train.txt").write_text("im0\n")
+        ds = VOCDetectionDataset(root, "train")
+        img, target = ds[0]
+        assert img.shape == (3, 80, 100)
+        assert target["boxes"].tolist() == [[10.0, 20.0, 50.0, 60.0]]
+        assert target["labels"].tolist() == [8]  # 'cat' in VOC_CLASSES
+
+
+def test_coco_dataset_parsing():
+    from deeplearning_amd.data import COCODetectionDataset
+    with tempfile.TemporaryDirectory() as d:
+        Image.fromarray(np.zeros((60, 60, 3), dtype=np.uint8)).save(
+            f"{d}/x.jpg")
+        ann = {"images": [{"id": 7, "file_name": "x.jpg",
+                           "width": 60, "height": 60}],
+               "annotations": [{"id": 1, "image_id": 7, "category_id": 3,
+                                "bbox": [5, 5, 20, 30], "iscrowd": 0}],
+               "categories": [{"id": 3, "name": "dog"}]}
+        with open(f"{d}/ann.json", "w") as f:
+            json.dump(ann, f)
+        ds = COCODetectionDataset(d, f"{d}/ann.json")
+        img, target = ds[0]
+        assert target["boxes"].tolist() == [[5.0, 5.0, 25.0, 35.0]]
+        assert target["labels"].tolist() == [1]
+
+
+def test_mixup_soft_targets_sum_to_one():
+    from deeplearning_amd.data import Mixup
+    torch.manual_seed(0)
+    mix = Mixup(num_classes=7)
+    x, y = mix(torch.rand(4, 3, 16, 16), torch.tensor([0, 1, 2, 3]))
+    assert y.shape == (4, 7)
+    assert torch.allclose(y.sum(1), torch.ones(4), atol=1e-5)
+
+
+def test_mosaic_boxes_stay_inside():
+    from deeplearning_amd.data import mosaic4
+    torch.manual_seed(0)
+    imgs = [torch.rand(3, 80, 90) for _ in range(4)]
+    ts = [{"boxes": torch.tensor([[5.0, 5.0, 40.0, 50.0]]),
+           "labels": torch.tensor([1])} for _ in range(4)]
+    canvas, t = mosaic4(imgs, ts, out_size=160)
+    assert canvas.shape == (3, 160, 160)
+    if t["boxes"].numel():
+        assert t["boxes"].min() >= 0 and t["boxes"].max() <= 160
+
+
+def test_det_evaluator_localization_quality():
+    from deeplearning_amd.engine.det_eval import DetEvaluator
+    gt = {"boxes": torch.tensor([[10.0, 10.0, 50.0, 50.0]]),
+          "labels": torch.tensor([1]),
+          "iscrowd": torch.tensor([0])}
+    # slightly offset box: IoU ~0.68 -> counts at 0.5 but not at 0.75
+    pred = {"boxes": torch.tensor([[15.0, 15.0, 55.0, 55.0]]),
+            "scores": torch.tensor([0.9]),
+            "labels": torch.tensor([1])}
+    ev = DetEvaluator()
+    ev.update([pred], [gt])
+    s = ev.summarize()
+    assert s["mAP50"] == pytest.approx(1.0)
+    assert s["mAP75"] == pytest.approx(0.0)
+    assert 0.0 < s["mAP"] < 1.0
+
+
+def test_det_evaluator_crowd_ignored():
+    from deeplearning_amd.engine.det_eval import DetEvaluator
+    gt = {"boxes": torch.tensor([[0.0, 0.0, 100.0, 100.0]]),
+          "labels": torch.tensor([1]),
+          "iscrowd": torch.tensor([1])}
+    pred = {"boxes": torch.tensor([[0.0, 0.0, 100.0, 100.0]]),
+            "scores": torch.tensor([0.9]), "labels": torch.tensor([1])}
+    ev = DetEvaluator()
+    ev.update([pred], [gt])
+    # only crowd gt -> no countable gt -> empty summary, not FP explosion
+    assert ev.summarize()["mAP"] == 0.0
+
+
+def _run(script, *args, timeout=240):
+    return subprocess.run([sys.executable, str(REPO / script), *args],
+                          capture_output=True, text=True, timeout=timeout,
+                          cwd=REPO)
+
+
+@pytest.mark.slow
+def test_mnist_project_cli_checkpoint_layout(tmp_path):
+    r = _run("projects/classification/mnist/train.py", "--epochs", "2",
+             "--batch-size", "8", "--synthetic-size", "32", "--workers", "0",
+             "--device", "cpu", "--output", str(tmp_path))
+    assert r.returncode == 0, r.stderr[-2000:]
+    weights = tmp_path / "mnist" / "weights"
+    assert (weights / "model_0.pth").exists()
+    assert (weights / "model_1.pth").exists()
+    assert (weights / "best_model.pth").exists()
+    assert (weights / "ckpt_epoch_1.pth").exists()
+    ckpt = torch.load(weights / "ckpt_epoch_1.pth", map_location="cpu",
+                      weights_only=False)
+    assert {"model", "optimizer", "lr_scheduler", "epoch"} <= set(ckpt)
+
+
+@pytest.mark.slow
+def test_unet_project_cli(tmp_path):
+    r = _run("projects/Image_segmentation/U-Net/train.py", "--epochs", "1",
+             "--img-size", "64", "--batch-size", "2", "--synthetic-size",
+             "4", "--num-classes", "3", "--workers", "0", "--device", "cpu",
+             "--output", str(tmp_path))
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert (tmp_path / "U-Net" / "weights" / "model_0.pth").exists()
