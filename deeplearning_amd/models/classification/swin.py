@@ -305,7 +305,10 @@ class SwinTransformer(nn.Module):
                 nn.init.zeros_(m.bias)
 
     def forward_features(self, x):
-        x = self.patch_embed(x).flatten(2).transpose(1, 2)  # B, L, C
+        from .vit import patch_embed_gemm
+        x = patch_embed_gemm(x, self.patch_embed.weight,
+                             self.patch_embed.bias,
+                             self.patch_embed.kernel_size[0])  # B, L, C
         x = self.patch_norm(x)
         if self.ape:
             x = x + self.absolute_pos_embed
