@@ -30,26 +30,32 @@ def find_kernel_view(cur):
             yield cand, cols
 
 
-def summarize(db_path, topn=30):
+def summarize(db_path, topn=30, tail_frac=0.0):
+    """tail_frac > 0: only count dispatches whose start falls in the LAST
+    `tail_frac` of the trace window (drops MIOpen-find / warmup noise)."""
     con = sqlite3.connect(db_path)
     cur = con.cursor()
     rows = None
     for table, cols in find_kernel_view(cur):
         name_col = next(c for c in ("kernel_name", "name", "kernelname")
                         if c in cols)
-        if "start" in cols and "end" in cols:
-            expr = "SUM(end - start)", "COUNT(*)", "AVG(end - start)"
-        else:
-            dur = next(c for c in cols if "duration" in c)
-            expr = f"SUM({dur})", "COUNT(*)", f"AVG({dur})"
+        if not ("start" in cols and "end" in cols):
+            continue
+        where = ""
+        if tail_frac > 0:
+            cur.execute(f"SELECT MIN(start), MAX(end) FROM {table}")
+            t0, t1 = cur.fetchone()
+            cut = t1 - (t1 - t0) * tail_frac
+            where = f" WHERE start >= {cut}"
         try:
             cur.execute(
-                f"SELECT {name_col}, {expr[0]} total, {expr[1]} calls, "
-                f"{expr[2]} avg FROM {table} GROUP BY {name_col} "
-                f"ORDER BY total DESC LIMIT {topn}")
+                f"SELECT {name_col}, SUM(end - start) total, COUNT(*) calls, "
+                f"AVG(end - start) avg FROM {table}{where} "
+                f"GROUP BY {name_col} ORDER BY total DESC LIMIT {topn}")
             rows = cur.fetchall()
             if rows:
-                print(f"# table: {table}", file=sys.stderr)
+                print(f"# table: {table} tail_frac={tail_frac}",
+                      file=sys.stderr)
                 break
         except sqlite3.Error:
             continue
@@ -62,8 +68,9 @@ def main():
     p.add_argument("db")
     p.add_argument("-n", type=int, default=30)
     p.add_argument("-o", default=None)
+    p.add_argument("--tail-frac", type=float, default=0.0)
     args = p.parse_args()
-    rows = summarize(args.db, args.n)
+    rows = summarize(args.db, args.n, args.tail_frac)
     total = sum(r[1] for r in rows) or 1
     out = [("kernel", "total_ns", "calls", "avg_ns", "pct_of_top")]
     for name, tot, calls, avg in rows:
