@@ -51,6 +51,13 @@ torch::Tensor roialign_fwd(torch::Tensor input, torch::Tensor rois, int64_t PH,
 torch::Tensor roialign_bwd(torch::Tensor grad_out, torch::Tensor rois, int64_t N,
                            int64_t C, int64_t H, int64_t W, double spatial_scale,
                            int64_t sampling_ratio, bool aligned);
+// attention.hip
+std::vector<torch::Tensor> attn_fwd(torch::Tensor qkv, int64_t num_heads,
+                                    double scale,
+                                    c10::optional<torch::Tensor> bias,
+                                    c10::optional<torch::Tensor> mask,
+                                    bool save_p);
+torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B);
 // window.hip
 torch::Tensor window_partition_fwd(torch::Tensor x, int64_t ws, int64_t shift);
 torch::Tensor window_partition_bwd(torch::Tensor grad, int64_t B, int64_t H,
@@ -80,6 +87,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("nms", &nms_gpu);
   m.def("roialign_fwd", &roialign_fwd);
   m.def("roialign_bwd", &roialign_bwd);
+  m.def("attn_fwd", &attn_fwd, py::arg("qkv"), py::arg("num_heads"),
+        py::arg("scale"), py::arg("bias") = py::none(),
+        py::arg("mask") = py::none(), py::arg("save_p") = false);
+  m.def("mfma_probe", &mfma_probe);
   m.def("window_partition_fwd", &window_partition_fwd);
   m.def("window_partition_bwd", &window_partition_bwd);
   m.def("window_merge_fwd", &window_merge_fwd);

@@ -95,7 +95,15 @@ class WindowAttention(nn.Module):
         return bias.permute(2, 0, 1).contiguous().unsqueeze(0)  # 1, nH, N, N
 
     def forward(self, x, mask=None):
+        from ...ops.attention import fused_attention
         B_, N, C = x.shape
+        if not self.v2 and (self.attn_drop.p == 0.0 or not self.training):
+            # one fused HIP MFMA kernel: scale*QK^T + rel-pos bias (+window
+            # mask) + softmax + V, straight from the packed qkv projection
+            qkv = self.qkv(x)
+            out = fused_attention(qkv, self.num_heads, self.scale,
+                                  bias=self._bias().squeeze(0), mask=mask)
+            return self.proj_drop(self.proj(out))
         if self.v2:
             qkv_bias = None
             if self.q_bias is not None:

@@ -61,8 +61,14 @@ class Attention(nn.Module):
         self.proj_drop = nn.Dropout(proj_drop)
 
     def forward(self, x):
+        from ...ops.attention import fused_attention
         B, N, C = x.shape
-        qkv = self.qkv(x).reshape(B, N, 3, self.num_heads, self.head_dim)
+        qkv = self.qkv(x)
+        if self.attn_drop.p == 0.0 or not self.training:
+            # one fused HIP MFMA kernel (csrc/attention.hip)
+            out = fused_attention(qkv, self.num_heads, self.scale)
+            return self.proj_drop(self.proj(out))
+        qkv = qkv.reshape(B, N, 3, self.num_heads, self.head_dim)
         qkv = qkv.permute(2, 0, 3, 1, 4)  # 3, B, H, N, d
         q, k, v = qkv.unbind(0)
         attn = (q @ k.transpose(-2, -1)) * self.scale

@@ -67,13 +67,14 @@ def same_layout(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
 
 import os
 
-_FORCE_EAGER = os.environ.get("DLA_FORCE_EAGER", "0") == "1"  # A/B perf testing only
+def _force_eager() -> bool:
+    return os.environ.get("DLA_FORCE_EAGER", "0") == "1"  # A/B testing only
 
 
 def use_hip(*tensors) -> bool:
     """True iff all tensors are on GPU. On GPU the extension is REQUIRED:
     if it is missing this raises instead of falling back."""
-    if _FORCE_EAGER:
+    if _force_eager():
         return False
     on_gpu = all(t.is_cuda for t in tensors if isinstance(t, torch.Tensor))
     if not on_gpu:
