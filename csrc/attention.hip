@@ -50,6 +50,7 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ qkv,  // [B,N,3,H,D]
                      const float* __restrict__ mask,          // [nW,N,N] or null
                      __hip_bfloat16* __restrict__ out,        // [B,N,H*D]
                      __hip_bfloat16* __restrict__ p_out,      // [B,H,N,N] or null
+                     float* __restrict__ stats,               // [2,B,H,N] or null
                      int B, int N, int H, int n_win, float scale) {
   constexpr int KSLICES = D / 32;        // mfma K-steps over head dim
   const int b = blockIdx.x / H;
@@ -158,6 +159,13 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ qkv,  // [B,N,3,H,D]
       }
       row_max[r] = m;
       row_sum[r] = wave16_sum(sum);
+      if (stats != nullptr && (lane & 15) == 0) {
+        const int qrow = q0 + (lane >> 4) * 4 + r;
+        if (qrow < N) {
+          stats[((int64_t)b * H + h) * N + qrow] = m;
+          stats[((int64_t)(B + b) * H + h) * N + qrow] = row_sum[r];
+        }
+      }
     }
 
     // ---- P -> LDS (bf16), per-wave buffer [16 rows][KPAD row stride] --------
@@ -246,6 +254,329 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ qkv,  // [B,N,3,H,D]
   }
 }
 
+// ---- fused attention backward (no bias/mask; flash-style recompute) --------
+// Saved from fwd: row max m and row sum l ([B,H,N] f32 each). D_row =
+// rowsum(dO * O) is computed by the host in one fused reduce.
+//   P = exp(scale*QK^T - m)/l
+//   dV = P^T dO ; dP = dO V^T ; dS = P*(dP - D_row) ; dQ = scale*dS K ;
+//   dK = scale*dS^T Q
+// Kernel A (dK/dV): per (b,h) workgroup; Q and dO staged in LDS both
+// row-major (B-fragments of S^T / dP^T) and transposed (B-fragments of the
+// dK / dV MFMAs); each wave owns 16 keys per iteration, K/V fragments read
+// straight from global.
+// Kernel B (dQ): K and V row-major + K transposed in LDS; each wave owns 16
+// queries.
+
+template <int D>
+__global__ __launch_bounds__(256)
+void attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ qkv,
+                        const __hip_bfloat16* __restrict__ dout,  // [B,N,H*D]
+                        const float* __restrict__ stats,          // [2,B,H,N]
+                        const float* __restrict__ drow,           // [B,H,N]
+                        __hip_bfloat16* __restrict__ dqkv,
+                        int B, int N, int H, float scale) {
+  constexpr int KSLICES = D / 32;
+  const int b = blockIdx.x / H;
+  const int h = blockIdx.x % H;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int Npad = (N + 15) & ~15;
+  const int n_qtiles = Npad / 16;
+
+  extern __shared__ __attribute__((aligned(16))) char lds_raw[];
+  __hip_bfloat16* q_lds = (__hip_bfloat16*)lds_raw;        // [Npad][KPAD]
+  __hip_bfloat16* do_lds = q_lds + Npad * KPAD;            // [Npad][KPAD]
+  const int VROW = Npad + 8;
+  __hip_bfloat16* qt_lds = do_lds + Npad * KPAD;           // [D][VROW]
+  __hip_bfloat16* dot_lds = qt_lds + D * VROW;             // [D][VROW]
+  const int PROW = Npad + 8;
+  __hip_bfloat16* p_lds = dot_lds + D * VROW;              // [4][16][PROW]
+
+  const int64_t bh_stride = (int64_t)3 * H * D;
+  const __hip_bfloat16* q_src = qkv + (int64_t)b * N * 3 * H * D +
+                                (int64_t)h * D;
+  const __hip_bfloat16* do_src = dout + (int64_t)b * N * H * D +
+                                 (int64_t)h * D;
+  for (int n = threadIdx.x / (D / 4); n < Npad; n += 256 / (D / 4)) {
+    const int d0 = (threadIdx.x % (D / 4)) * 4;
+    Vec<__hip_bfloat16, 4> qv, dv;
+    if (n < N) {
+      qv = vload<__hip_bfloat16, 4>(q_src + n * bh_stride + d0);
+      dv = vload<__hip_bfloat16, 4>(do_src + (int64_t)n * H * D + d0);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        qv.v[j] = __hip_bfloat16(0.f);
+        dv.v[j] = __hip_bfloat16(0.f);
+      }
+    }
+    vstore<__hip_bfloat16, 4>(&q_lds[n * KPAD + d0], qv);
+    vstore<__hip_bfloat16, 4>(&do_lds[n * KPAD + d0], dv);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      qt_lds[(d0 + j) * VROW + n] = qv.v[j];
+      dot_lds[(d0 + j) * VROW + n] = dv.v[j];
+    }
+  }
+  __syncthreads();
+
+  const float* m_arr = stats + ((int64_t)b * H + h) * N;
+  const float* l_arr = stats + ((int64_t)(B + b) * H + h) * N;
+  const float* d_arr = drow + ((int64_t)b * H + h) * N;
+  const __hip_bfloat16* k_src = qkv + ((int64_t)b * N * 3 + 1) * H * D +
+                                (int64_t)h * D;
+  const __hip_bfloat16* v_src = qkv + ((int64_t)b * N * 3 + 2) * H * D +
+                                (int64_t)h * D;
+  __hip_bfloat16* p_buf = p_lds + wave * 16 * PROW;
+  const int n_ktiles_own = (N + 15) / 16;
+
+  for (int kt = wave; kt < n_ktiles_own; kt += 4) {
+    const int key0 = kt * 16;
+    // K_tile / V_tile A-fragments from global: row=key, k=d
+    bf16x8 k_frag[KSLICES], v_frag[KSLICES];
+    {
+      const int key = key0 + (lane & 15);
+      const int skey = key < N ? key : N - 1;
+#pragma unroll
+      for (int sl = 0; sl < KSLICES; ++sl) {
+        const int d0 = sl * 32 + (lane >> 4) * 8;
+        k_frag[sl] = *(const bf16x8*)(k_src + skey * bh_stride + d0);
+        v_frag[sl] = *(const bf16x8*)(v_src + skey * bh_stride + d0);
+      }
+    }
+    // per q-tile: S^T and dP^T accumulators
+    f32x4 st_acc[16], dpt_acc[16];
+#pragma unroll 2
+    for (int qt = 0; qt < n_qtiles; ++qt) {
+      f32x4 sacc = {0.f, 0.f, 0.f, 0.f}, dacc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int sl = 0; sl < KSLICES; ++sl) {
+        const int q = qt * 16 + (lane & 15);
+        const int d0 = sl * 32 + (lane >> 4) * 8;
+        bf16x8 qb = *(const bf16x8*)(&q_lds[q * KPAD + d0]);
+        bf16x8 db = *(const bf16x8*)(&do_lds[q * KPAD + d0]);
+        sacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(k_frag[sl], qb, sacc,
+                                                       0, 0, 0);
+        dacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(v_frag[sl], db, dacc,
+                                                       0, 0, 0);
+      }
+      st_acc[qt] = sacc;
+      dpt_acc[qt] = dacc;
+    }
+    // P^T and dS^T: C/D col = q (lane&15), row = key ((lane>>4)*4+reg)
+    const int qcol_base = lane & 15;
+#pragma unroll 2
+    for (int qt = 0; qt < n_qtiles; ++qt) {
+      const int q = qt * 16 + qcol_base;
+      const float m = q < N ? m_arr[q] : 0.f;
+      const float li = q < N ? 1.f / l_arr[q] : 0.f;
+      const float dr = q < N ? d_arr[q] : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const float pt = __expf(st_acc[qt][r] * scale - m) * li;
+        st_acc[qt][r] = pt;                                   // P^T
+        dpt_acc[qt][r] = pt * (dpt_acc[qt][r] - dr) * scale;  // scale*dS^T
+      }
+    }
+    // ---- dV = P^T dO : stage P^T, MFMA over q ------------------------------
+#pragma unroll 2
+    for (int qt = 0; qt < n_qtiles; ++qt)
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        p_buf[((lane >> 4) * 4 + r) * PROW + qt * 16 + qcol_base] =
+            __hip_bfloat16(st_acc[qt][r]);
+    f32x4 dv_acc[D / 16], dk_acc[D / 16];
+#pragma unroll
+    for (int dt = 0; dt < D / 16; ++dt) {
+      dv_acc[dt] = {0.f, 0.f, 0.f, 0.f};
+      dk_acc[dt] = {0.f, 0.f, 0.f, 0.f};
+    }
+    for (int q2 = 0; q2 < Npad; q2 += 32) {
+      const int k0 = q2 + (lane >> 4) * 8;
+      bf16x8 pt_frag = *(const bf16x8*)(&p_buf[(lane & 15) * PROW + k0]);
+#pragma unroll
+      for (int dt = 0; dt < D / 16; ++dt) {
+        const int d = dt * 16 + (lane & 15);
+        bf16x8 dob = *(const bf16x8*)(&dot_lds[d * VROW + k0]);
+        dv_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pt_frag, dob,
+                                                             dv_acc[dt], 0, 0, 0);
+      }
+    }
+    // ---- dK = (scale dS)^T Q : stage dS^T, MFMA over q ---------------------
+    // p_buf rewrite is wave-local; same-wave LDS ops stay ordered (no block
+    // barrier here - waves have different kt trip counts).
+#pragma unroll 2
+    for (int qt = 0; qt < n_qtiles; ++qt)
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        p_buf[((lane >> 4) * 4 + r) * PROW + qt * 16 + qcol_base] =
+            __hip_bfloat16(dpt_acc[qt][r]);
+    for (int q2 = 0; q2 < Npad; q2 += 32) {
+      const int k0 = q2 + (lane >> 4) * 8;
+      bf16x8 ds_frag = *(const bf16x8*)(&p_buf[(lane & 15) * PROW + k0]);
+#pragma unroll
+      for (int dt = 0; dt < D / 16; ++dt) {
+        const int d = dt * 16 + (lane & 15);
+        bf16x8 qb = *(const bf16x8*)(&qt_lds[d * VROW + k0]);
+        dk_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ds_frag, qb,
+                                                             dk_acc[dt], 0, 0, 0);
+      }
+    }
+    // ---- write dK (comp 1) and dV (comp 2): C/D row=key, col=d -------------
+#pragma unroll
+    for (int dt = 0; dt < D / 16; ++dt) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int key = key0 + (lane >> 4) * 4 + r;
+        if (key >= N) continue;
+        const int d = dt * 16 + (lane & 15);
+        const int64_t base = ((int64_t)(b * N + key) * 3) * H * D +
+                             (int64_t)h * D + d;
+        dqkv[base + (int64_t)H * D] = __hip_bfloat16(dk_acc[dt][r]);
+        dqkv[base + (int64_t)2 * H * D] = __hip_bfloat16(dv_acc[dt][r]);
+      }
+    }
+  }
+}
+
+template <int D>
+__global__ __launch_bounds__(256)
+void attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ qkv,
+                       const __hip_bfloat16* __restrict__ dout,
+                       const float* __restrict__ stats,
+                       const float* __restrict__ drow,
+                       __hip_bfloat16* __restrict__ dqkv,
+                       int B, int N, int H, float scale) {
+  constexpr int KSLICES = D / 32;
+  const int b = blockIdx.x / H;
+  const int h = blockIdx.x % H;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int Npad = (N + 15) & ~15;
+  const int n_ktiles = Npad / 16;
+
+  extern __shared__ __attribute__((aligned(16))) char lds_raw[];
+  __hip_bfloat16* k_lds = (__hip_bfloat16*)lds_raw;        // [Npad][KPAD]
+  __hip_bfloat16* v_lds = k_lds + Npad * KPAD;             // [Npad][KPAD]
+  const int VROW = Npad + 8;
+  __hip_bfloat16* kt_lds = v_lds + Npad * KPAD;            // [D][VROW]
+  const int PROW = Npad + 8;
+  __hip_bfloat16* p_lds = kt_lds + D * VROW;               // [4][16][PROW]
+
+  const int64_t bh_stride = (int64_t)3 * H * D;
+  const __hip_bfloat16* k_src = qkv + ((int64_t)b * N * 3 + 1) * H * D +
+                                (int64_t)h * D;
+  const __hip_bfloat16* v_src = qkv + ((int64_t)b * N * 3 + 2) * H * D +
+                                (int64_t)h * D;
+  for (int n = threadIdx.x / (D / 4); n < Npad; n += 256 / (D / 4)) {
+    const int d0 = (threadIdx.x % (D / 4)) * 4;
+    Vec<__hip_bfloat16, 4> kv, vv;
+    if (n < N) {
+      kv = vload<__hip_bfloat16, 4>(k_src + n * bh_stride + d0);
+      vv = vload<__hip_bfloat16, 4>(v_src + n * bh_stride + d0);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        kv.v[j] = __hip_bfloat16(0.f);
+        vv.v[j] = __hip_bfloat16(0.f);
+      }
+    }
+    vstore<__hip_bfloat16, 4>(&k_lds[n * KPAD + d0], kv);
+    vstore<__hip_bfloat16, 4>(&v_lds[n * KPAD + d0], vv);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) kt_lds[(d0 + j) * VROW + n] = kv.v[j];
+  }
+  __syncthreads();
+
+  const float* m_arr = stats + ((int64_t)b * H + h) * N;
+  const float* l_arr = stats + ((int64_t)(B + b) * H + h) * N;
+  const float* d_arr = drow + ((int64_t)b * H + h) * N;
+  const __hip_bfloat16* q_src = qkv + (int64_t)b * N * 3 * H * D +
+                                (int64_t)h * D;
+  const __hip_bfloat16* do_src = dout + (int64_t)b * N * H * D +
+                                 (int64_t)h * D;
+  __hip_bfloat16* p_buf = p_lds + wave * 16 * PROW;
+  const int n_qblocks = (N + 15) / 16;
+
+  for (int qb = wave; qb < n_qblocks; qb += 4) {
+    const int q0 = qb * 16;
+    bf16x8 q_frag[KSLICES], do_frag[KSLICES];
+    {
+      const int row = q0 + (lane & 15);
+      const int srow = row < N ? row : N - 1;
+#pragma unroll
+      for (int sl = 0; sl < KSLICES; ++sl) {
+        const int d0 = sl * 32 + (lane >> 4) * 8;
+        q_frag[sl] = *(const bf16x8*)(q_src + srow * bh_stride + d0);
+        do_frag[sl] = *(const bf16x8*)(do_src + (int64_t)srow * H * D + d0);
+      }
+    }
+    f32x4 s_acc[16], dp_acc[16];
+#pragma unroll 2
+    for (int kt = 0; kt < n_ktiles; ++kt) {
+      f32x4 sacc = {0.f, 0.f, 0.f, 0.f}, dacc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int sl = 0; sl < KSLICES; ++sl) {
+        const int key = kt * 16 + (lane & 15);
+        const int d0 = sl * 32 + (lane >> 4) * 8;
+        bf16x8 kb = *(const bf16x8*)(&k_lds[key * KPAD + d0]);
+        bf16x8 vb = *(const bf16x8*)(&v_lds[key * KPAD + d0]);
+        sacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(q_frag[sl], kb, sacc,
+                                                       0, 0, 0);
+        dacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(do_frag[sl], vb, dacc,
+                                                       0, 0, 0);
+      }
+      s_acc[kt] = sacc;
+      dp_acc[kt] = dacc;
+    }
+    // ds = scale * P * (dP - Drow); C/D row = (lane>>4)*4+reg (query)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int qrow = q0 + (lane >> 4) * 4 + r;
+      const float m = qrow < N ? m_arr[qrow] : 0.f;
+      const float li = qrow < N ? 1.f / l_arr[qrow] : 0.f;
+      const float dr = qrow < N ? d_arr[qrow] : 0.f;
+#pragma unroll 2
+      for (int kt = 0; kt < n_ktiles; ++kt) {
+        const int key = kt * 16 + (lane & 15);
+        float pv = (key < N) ? __expf(s_acc[kt][r] * scale - m) * li : 0.f;
+        s_acc[kt][r] = pv * (dp_acc[kt][r] - dr) * scale;
+      }
+    }
+#pragma unroll 2
+    for (int kt = 0; kt < n_ktiles; ++kt)
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        p_buf[((lane >> 4) * 4 + r) * PROW + kt * 16 + (lane & 15)] =
+            __hip_bfloat16(s_acc[kt][r]);
+    f32x4 dq_acc[D / 16];
+#pragma unroll
+    for (int dt = 0; dt < D / 16; ++dt) dq_acc[dt] = {0.f, 0.f, 0.f, 0.f};
+    for (int k2 = 0; k2 < Npad; k2 += 32) {
+      const int k0 = k2 + (lane >> 4) * 8;
+      bf16x8 ds_frag = *(const bf16x8*)(&p_buf[(lane & 15) * PROW + k0]);
+#pragma unroll
+      for (int dt = 0; dt < D / 16; ++dt) {
+        const int d = dt * 16 + (lane & 15);
+        bf16x8 kb = *(const bf16x8*)(&kt_lds[d * VROW + k0]);
+        dq_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ds_frag, kb,
+                                                             dq_acc[dt], 0, 0, 0);
+      }
+    }
+#pragma unroll
+    for (int dt = 0; dt < D / 16; ++dt) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int qrow = q0 + (lane >> 4) * 4 + r;
+        if (qrow >= N) continue;
+        const int d = dt * 16 + (lane & 15);
+        dqkv[((int64_t)(b * N + qrow) * 3) * H * D + (int64_t)h * D + d] =
+            __hip_bfloat16(dq_acc[dt][r]);
+      }
+    }
+  }
+}
+
 // numerics probe: D[16,16] = A[16,32] @ B[32,16] via one mfma, to pin the
 // fragment layout on real hardware (guide: A=I with asymmetric B).
 __global__ void mfma_probe_kernel(const __hip_bfloat16* A,
@@ -274,7 +605,7 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor qkv, int64_t num_heads,
                                     double scale,
                                     c10::optional<torch::Tensor> bias,
                                     c10::optional<torch::Tensor> mask,
-                                    bool save_p) {
+                                    bool save_p, bool save_stats) {
   // qkv: [B, N, 3*H*D] or [B, N, 3, H, D] contiguous bf16
   TORCH_CHECK(qkv.scalar_type() == at::kBFloat16, "attn_fwd wants bf16 qkv");
   TORCH_CHECK(qkv.is_contiguous(), "attn_fwd wants contiguous qkv");
@@ -288,8 +619,10 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor qkv, int64_t num_heads,
   TORCH_CHECK(N <= 256, "attn_fwd supports N <= 256, got ", N);
 
   auto out = torch::empty({B, N, (int64_t)H * D}, qkv.options());
-  torch::Tensor p;
+  torch::Tensor p, stats;
   if (save_p) p = torch::empty({B, H, N, N}, qkv.options());
+  if (save_stats)
+    stats = torch::empty({2, B, H, N}, qkv.options().dtype(torch::kFloat));
 
   int n_win = 1;
   const float* bias_ptr = nullptr;
@@ -322,6 +655,7 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor qkv, int64_t num_heads,
                        (const __hip_bfloat16*)qkv.data_ptr(), bias_ptr,
                        mask_ptr, (__hip_bfloat16*)out.data_ptr(),
                        save_p ? (__hip_bfloat16*)p.data_ptr() : nullptr,
+                       save_stats ? stats.data_ptr<float>() : nullptr,
                        B, N, H, n_win, (float)scale);
   };
   auto d3 = [&](auto dtag) {
@@ -341,8 +675,10 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor qkv, int64_t num_heads,
   if (D == 64) d3(std::integral_constant<int, 64>{});
   else d3(std::integral_constant<int, 32>{});
   HIP_CHECK_ERR();
-  if (save_p) return {out, p};
-  return {out};
+  std::vector<torch::Tensor> res = {out};
+  if (save_p) res.push_back(p);
+  if (save_stats) res.push_back(stats);
+  return res;
 }
 
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B) {
@@ -354,4 +690,41 @@ torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B) {
                      D.data_ptr<float>());
   HIP_CHECK_ERR();
   return D;
+}
+
+std::vector<torch::Tensor> attn_bwd(torch::Tensor qkv, torch::Tensor dout,
+                                    torch::Tensor stats, torch::Tensor drow,
+                                    int64_t num_heads, double scale) {
+  TORCH_CHECK(qkv.is_contiguous() && dout.is_contiguous());
+  const int B = (int)qkv.size(0);
+  const int N = (int)qkv.size(1);
+  const int H = (int)num_heads;
+  const int D = (int)(qkv.numel() / ((int64_t)B * N * 3 * H));
+  TORCH_CHECK(D == 32 || D == 64);
+  auto dqkv = torch::empty_like(qkv);
+  const int Npad = (N + 15) & ~15;
+  dim3 grid(B * H), block(256);
+  const int lds_kv = (2 * Npad * dla::KPAD + 2 * D * (Npad + 8) +
+                      4 * 16 * (Npad + 8)) * 2;
+  const int lds_q = (2 * Npad * dla::KPAD + D * (Npad + 8) +
+                     4 * 16 * (Npad + 8)) * 2;
+  auto run = [&](auto dtag) {
+    constexpr int DD = decltype(dtag)::value;
+    hipLaunchKernelGGL((dla::attn_bwd_kv_kernel<DD>), grid, block, lds_kv,
+                       dla::stream(), (const __hip_bfloat16*)qkv.data_ptr(),
+                       (const __hip_bfloat16*)dout.data_ptr(),
+                       stats.data_ptr<float>(), drow.data_ptr<float>(),
+                       (__hip_bfloat16*)dqkv.data_ptr(), B, N, H,
+                       (float)scale);
+    hipLaunchKernelGGL((dla::attn_bwd_q_kernel<DD>), grid, block, lds_q,
+                       dla::stream(), (const __hip_bfloat16*)qkv.data_ptr(),
+                       (const __hip_bfloat16*)dout.data_ptr(),
+                       stats.data_ptr<float>(), drow.data_ptr<float>(),
+                       (__hip_bfloat16*)dqkv.data_ptr(), B, N, H,
+                       (float)scale);
+  };
+  if (D == 64) run(std::integral_constant<int, 64>{});
+  else run(std::integral_constant<int, 32>{});
+  HIP_CHECK_ERR();
+  return {dqkv};
 }

@@ -56,7 +56,10 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor qkv, int64_t num_heads,
                                     double scale,
                                     c10::optional<torch::Tensor> bias,
                                     c10::optional<torch::Tensor> mask,
-                                    bool save_p);
+                                    bool save_p, bool save_stats);
+std::vector<torch::Tensor> attn_bwd(torch::Tensor qkv, torch::Tensor dout,
+                                    torch::Tensor stats, torch::Tensor drow,
+                                    int64_t num_heads, double scale);
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B);
 // window.hip
 torch::Tensor window_partition_fwd(torch::Tensor x, int64_t ws, int64_t shift);
@@ -89,7 +92,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("roialign_bwd", &roialign_bwd);
   m.def("attn_fwd", &attn_fwd, py::arg("qkv"), py::arg("num_heads"),
         py::arg("scale"), py::arg("bias") = py::none(),
-        py::arg("mask") = py::none(), py::arg("save_p") = false);
+        py::arg("mask") = py::none(), py::arg("save_p") = false,
+        py::arg("save_stats") = false);
+  m.def("attn_bwd", &attn_bwd);
   m.def("mfma_probe", &mfma_probe);
   m.def("window_partition_fwd", &window_partition_fwd);
   m.def("window_partition_bwd", &window_partition_bwd);

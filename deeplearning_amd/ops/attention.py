@@ -35,13 +35,42 @@ def _eager_attention(qkv: torch.Tensor, num_heads: int, scale: float,
     return out
 
 
+class _AttnFusedBwdFn(torch.autograd.Function):
+    """No-bias/mask path: forward saves only per-row softmax stats; backward
+    recomputes P in-LDS (two HIP kernels, no [B,H,N,N] tensor ever hits HBM)."""
+
+    @staticmethod
+    def forward(ctx, qkv, num_heads, scale):
+        needs = qkv.requires_grad
+        res = ext().attn_fwd(qkv.contiguous(), num_heads, scale, None, None,
+                             False, needs)
+        ctx.num_heads = num_heads
+        ctx.scale = scale
+        if needs:
+            ctx.save_for_backward(qkv, res[0], res[1])
+        return res[0]
+
+    @staticmethod
+    def backward(ctx, dout):
+        qkv, out, stats = ctx.saved_tensors
+        H = ctx.num_heads
+        B, N, _ = qkv.shape
+        dout = dout.contiguous()
+        # D_row[b,h,n] = sum_d dO * O  (one fused reduce)
+        drow = (dout.float() * out.float()).view(B, N, H, -1).sum(-1) \
+            .permute(0, 2, 1).contiguous()
+        (dqkv,) = ext().attn_bwd(qkv, dout.to(qkv.dtype), stats, drow, H,
+                                 ctx.scale)
+        return dqkv, None, None
+
+
 class _AttnFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, qkv, num_heads, scale, bias, mask):
         needs_p = qkv.requires_grad or \
             (bias is not None and bias.requires_grad)
         res = ext().attn_fwd(qkv.contiguous(), num_heads, scale, bias, mask,
-                             needs_p)
+                             needs_p, False)
         ctx.num_heads = num_heads
         ctx.scale = scale
         ctx.has_bias = bias is not None
@@ -85,5 +114,7 @@ def fused_attention(qkv: torch.Tensor, num_heads: int, scale: float,
     d = qkv.shape[2] // (3 * num_heads)
     if (use_hip(qkv) and qkv.dtype == torch.bfloat16 and d in (32, 64)
             and qkv.shape[1] <= 256):
+        if bias is None and mask is None:
+            return _AttnFusedBwdFn.apply(qkv, num_heads, scale)
         return _AttnFn.apply(qkv, num_heads, scale, bias, mask)
     return _eager_attention(qkv, num_heads, scale, bias, mask)
