@@ -391,13 +391,27 @@ void attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ qkv,
       dv_acc[dt] = {0.f, 0.f, 0.f, 0.f};
       dk_acc[dt] = {0.f, 0.f, 0.f, 0.f};
     }
-    for (int q2 = 0; q2 < Npad; q2 += 32) {
+    for (int q2 = 0; q2 + 32 <= Npad; q2 += 32) {
       const int k0 = q2 + (lane >> 4) * 8;
       bf16x8 pt_frag = *(const bf16x8*)(&p_buf[(lane & 15) * PROW + k0]);
 #pragma unroll
       for (int dt = 0; dt < D / 16; ++dt) {
         const int d = dt * 16 + (lane & 15);
         bf16x8 dob = *(const bf16x8*)(&dot_lds[d * VROW + k0]);
+        dv_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pt_frag, dob,
+                                                             dv_acc[dt], 0, 0, 0);
+      }
+    }
+    if (Npad % 32) {  // 16-query half tile: upper k-halves masked to zero
+      const int k0 = (Npad & ~31) + (lane >> 4) * 8;
+      const bool lo = (lane >> 4) * 8 < 16;
+      bf16x8 pt_frag{};
+      if (lo) pt_frag = *(const bf16x8*)(&p_buf[(lane & 15) * PROW + k0]);
+#pragma unroll
+      for (int dt = 0; dt < D / 16; ++dt) {
+        const int d = dt * 16 + (lane & 15);
+        bf16x8 dob{};
+        if (lo) dob = *(const bf16x8*)(&dot_lds[d * VROW + k0]);
         dv_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pt_frag, dob,
                                                              dv_acc[dt], 0, 0, 0);
       }
@@ -411,13 +425,27 @@ void attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ qkv,
       for (int r = 0; r < 4; ++r)
         p_buf[((lane >> 4) * 4 + r) * PROW + qt * 16 + qcol_base] =
             __hip_bfloat16(dpt_acc[qt][r]);
-    for (int q2 = 0; q2 < Npad; q2 += 32) {
+    for (int q2 = 0; q2 + 32 <= Npad; q2 += 32) {
       const int k0 = q2 + (lane >> 4) * 8;
       bf16x8 ds_frag = *(const bf16x8*)(&p_buf[(lane & 15) * PROW + k0]);
 #pragma unroll
       for (int dt = 0; dt < D / 16; ++dt) {
         const int d = dt * 16 + (lane & 15);
         bf16x8 qb = *(const bf16x8*)(&qt_lds[d * VROW + k0]);
+        dk_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ds_frag, qb,
+                                                             dk_acc[dt], 0, 0, 0);
+      }
+    }
+    if (Npad % 32) {
+      const int k0 = (Npad & ~31) + (lane >> 4) * 8;
+      const bool lo = (lane >> 4) * 8 < 16;
+      bf16x8 ds_frag{};
+      if (lo) ds_frag = *(const bf16x8*)(&p_buf[(lane & 15) * PROW + k0]);
+#pragma unroll
+      for (int dt = 0; dt < D / 16; ++dt) {
+        const int d = dt * 16 + (lane & 15);
+        bf16x8 qb{};
+        if (lo) qb = *(const bf16x8*)(&qt_lds[d * VROW + k0]);
         dk_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ds_frag, qb,
                                                              dk_acc[dt], 0, 0, 0);
       }
@@ -552,13 +580,27 @@ void attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ qkv,
     f32x4 dq_acc[D / 16];
 #pragma unroll
     for (int dt = 0; dt < D / 16; ++dt) dq_acc[dt] = {0.f, 0.f, 0.f, 0.f};
-    for (int k2 = 0; k2 < Npad; k2 += 32) {
+    for (int k2 = 0; k2 + 32 <= Npad; k2 += 32) {
       const int k0 = k2 + (lane >> 4) * 8;
       bf16x8 ds_frag = *(const bf16x8*)(&p_buf[(lane & 15) * PROW + k0]);
 #pragma unroll
       for (int dt = 0; dt < D / 16; ++dt) {
         const int d = dt * 16 + (lane & 15);
         bf16x8 kb = *(const bf16x8*)(&kt_lds[d * VROW + k0]);
+        dq_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ds_frag, kb,
+                                                             dq_acc[dt], 0, 0, 0);
+      }
+    }
+    if (Npad % 32) {
+      const int k0 = (Npad & ~31) + (lane >> 4) * 8;
+      const bool lo = (lane >> 4) * 8 < 16;
+      bf16x8 ds_frag{};
+      if (lo) ds_frag = *(const bf16x8*)(&p_buf[(lane & 15) * PROW + k0]);
+#pragma unroll
+      for (int dt = 0; dt < D / 16; ++dt) {
+        const int d = dt * 16 + (lane & 15);
+        bf16x8 kb{};
+        if (lo) kb = *(const bf16x8*)(&kt_lds[d * VROW + k0]);
         dq_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ds_frag, kb,
                                                              dq_acc[dt], 0, 0, 0);
       }
