@@ -113,20 +113,25 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ qkv,  // [B,N,3,H,D]
     }
 
     // ---- S = scale * Q K^T over all key tiles -------------------------------
+    // NOTE: every loop over s_acc is fully unrolled with a compile-time bound
+    // so the accumulator array stays in VGPRs (a runtime-indexed array goes
+    // to scratch = global memory; measured 60x slowdown).
     f32x4 s_acc[16];  // up to 16 key tiles (N<=256)
-#pragma unroll 4
-    for (int kt = 0; kt < n_ktiles; ++kt) {
-      f32x4 acc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-      for (int s = 0; s < KSLICES; ++s) {
-        // B fragment: col=lane&15 (key), k=(lane>>4)*8+j
-        const int key = kt * 16 + (lane & 15);
-        const int d0 = s * 32 + (lane >> 4) * 8;
-        bf16x8 k_frag = *(const bf16x8*)(&k_lds[key * KPAD + d0]);
-        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(q_frag[s], k_frag, acc,
-                                                      0, 0, 0);
+    for (int kt = 0; kt < 16; ++kt) {
+      if (kt < n_ktiles) {
+        f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  #pragma unroll
+        for (int s = 0; s < KSLICES; ++s) {
+          // B fragment: col=lane&15 (key), k=(lane>>4)*8+j
+          const int key = kt * 16 + (lane & 15);
+          const int d0 = s * 32 + (lane >> 4) * 8;
+          bf16x8 k_frag = *(const bf16x8*)(&k_lds[key * KPAD + d0]);
+          acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(q_frag[s], k_frag, acc,
+                                                        0, 0, 0);
+        }
+        s_acc[kt] = acc;
       }
-      s_acc[kt] = acc;
     }
 
     // ---- softmax over each of the 16 rows this lane-group covers ------------
@@ -137,25 +142,29 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ qkv,  // [B,N,3,H,D]
     for (int r = 0; r < 4; ++r) {
       float m = -INFINITY;
       const int qrow = q0 + (lane >> 4) * 4 + r;
-#pragma unroll 4
-      for (int kt = 0; kt < n_ktiles; ++kt) {
-        const int key = kt * 16 + col_in_tile;
-        float s = s_acc[kt][r] * scale;
-        if ((HAS_BIAS || HAS_MASK) && key < N && qrow < N) {
-          if (HAS_BIAS) s += bias[((int64_t)h * N + qrow) * N + key];
-          if (HAS_MASK) s += mask[((int64_t)(b % n_win) * N + qrow) * N + key];
+#pragma unroll
+      for (int kt = 0; kt < 16; ++kt) {
+        if (kt < n_ktiles) {
+          const int key = kt * 16 + col_in_tile;
+          float s = s_acc[kt][r] * scale;
+          if ((HAS_BIAS || HAS_MASK) && key < N && qrow < N) {
+            if (HAS_BIAS) s += bias[((int64_t)h * N + qrow) * N + key];
+            if (HAS_MASK) s += mask[((int64_t)(b % n_win) * N + qrow) * N + key];
+          }
+          if (key >= N) s = -INFINITY;
+          s_acc[kt][r] = s;
+          m = fmaxf(m, s);
         }
-        if (key >= N) s = -INFINITY;
-        s_acc[kt][r] = s;
-        m = fmaxf(m, s);
       }
       m = wave16_max(m);
       float sum = 0.f;
-#pragma unroll 4
-      for (int kt = 0; kt < n_ktiles; ++kt) {
-        const float p = __expf(s_acc[kt][r] - m);
-        s_acc[kt][r] = p;
-        sum += p;
+#pragma unroll
+      for (int kt = 0; kt < 16; ++kt) {
+        if (kt < n_ktiles) {
+          const float p = __expf(s_acc[kt][r] - m);
+          s_acc[kt][r] = p;
+          sum += p;
+        }
       }
       row_max[r] = m;
       row_sum[r] = wave16_sum(sum);
@@ -171,26 +180,30 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ qkv,  // [B,N,3,H,D]
     // ---- P -> LDS (bf16), per-wave buffer [16 rows][KPAD row stride] --------
     __hip_bfloat16* p_buf = p_lds + wave * 16 * PROW;
     // rows are interleaved across lanes; each lane writes its 4 elements/tile
-#pragma unroll 4
-    for (int kt = 0; kt < n_ktiles; ++kt) {
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int row = (lane >> 4) * 4 + r;
-        p_buf[row * PROW + kt * 16 + col_in_tile] =
-            __hip_bfloat16(s_acc[kt][r]);
+    for (int kt = 0; kt < 16; ++kt) {
+      if (kt < n_ktiles) {
+  #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int row = (lane >> 4) * 4 + r;
+          p_buf[row * PROW + kt * 16 + col_in_tile] =
+              __hip_bfloat16(s_acc[kt][r]);
+        }
       }
     }
     if (SAVE_P) {
       const int qrow_base = q0;
-#pragma unroll 4
-      for (int kt = 0; kt < n_ktiles; ++kt) {
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int qrow = qrow_base + (lane >> 4) * 4 + r;
-          const int key = kt * 16 + col_in_tile;
-          if (qrow < N && key < N)
-            p_out[(((int64_t)b * H + h) * N + qrow) * N + key] =
-                __hip_bfloat16(s_acc[kt][r] / row_sum[r]);
+      for (int kt = 0; kt < 16; ++kt) {
+        if (kt < n_ktiles) {
+  #pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const int qrow = qrow_base + (lane >> 4) * 4 + r;
+            const int key = kt * 16 + col_in_tile;
+            if (qrow < N && key < N)
+              p_out[(((int64_t)b * H + h) * N + qrow) * N + key] =
+                  __hip_bfloat16(s_acc[kt][r] / row_sum[r]);
+          }
         }
       }
     }
@@ -344,47 +357,54 @@ void attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ qkv,
         v_frag[sl] = *(const bf16x8*)(v_src + skey * bh_stride + d0);
       }
     }
-    // per q-tile: S^T and dP^T accumulators
+    // per q-tile: S^T and dP^T accumulators (full unroll: keep in VGPRs)
     f32x4 st_acc[16], dpt_acc[16];
-#pragma unroll 2
-    for (int qt = 0; qt < n_qtiles; ++qt) {
-      f32x4 sacc = {0.f, 0.f, 0.f, 0.f}, dacc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-      for (int sl = 0; sl < KSLICES; ++sl) {
-        const int q = qt * 16 + (lane & 15);
-        const int d0 = sl * 32 + (lane >> 4) * 8;
-        bf16x8 qb = *(const bf16x8*)(&q_lds[q * KPAD + d0]);
-        bf16x8 db = *(const bf16x8*)(&do_lds[q * KPAD + d0]);
-        sacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(k_frag[sl], qb, sacc,
-                                                       0, 0, 0);
-        dacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(v_frag[sl], db, dacc,
-                                                       0, 0, 0);
+    for (int qt = 0; qt < 16; ++qt) {
+      if (qt < n_qtiles) {
+        f32x4 sacc = {0.f, 0.f, 0.f, 0.f}, dacc = {0.f, 0.f, 0.f, 0.f};
+  #pragma unroll
+        for (int sl = 0; sl < KSLICES; ++sl) {
+          const int q = qt * 16 + (lane & 15);
+          const int d0 = sl * 32 + (lane >> 4) * 8;
+          bf16x8 qb = *(const bf16x8*)(&q_lds[q * KPAD + d0]);
+          bf16x8 db = *(const bf16x8*)(&do_lds[q * KPAD + d0]);
+          sacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(k_frag[sl], qb, sacc,
+                                                         0, 0, 0);
+          dacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(v_frag[sl], db, dacc,
+                                                         0, 0, 0);
+        }
+        st_acc[qt] = sacc;
+        dpt_acc[qt] = dacc;
       }
-      st_acc[qt] = sacc;
-      dpt_acc[qt] = dacc;
     }
     // P^T and dS^T: C/D col = q (lane&15), row = key ((lane>>4)*4+reg)
     const int qcol_base = lane & 15;
-#pragma unroll 2
-    for (int qt = 0; qt < n_qtiles; ++qt) {
-      const int q = qt * 16 + qcol_base;
-      const float m = q < N ? m_arr[q] : 0.f;
-      const float li = q < N ? 1.f / l_arr[q] : 0.f;
-      const float dr = q < N ? d_arr[q] : 0.f;
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const float pt = __expf(st_acc[qt][r] * scale - m) * li;
-        st_acc[qt][r] = pt;                                   // P^T
-        dpt_acc[qt][r] = pt * (dpt_acc[qt][r] - dr) * scale;  // scale*dS^T
+    for (int qt = 0; qt < 16; ++qt) {
+      if (qt < n_qtiles) {
+        const int q = qt * 16 + qcol_base;
+        const float m = q < N ? m_arr[q] : 0.f;
+        const float li = q < N ? 1.f / l_arr[q] : 0.f;
+        const float dr = q < N ? d_arr[q] : 0.f;
+  #pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const float pt = __expf(st_acc[qt][r] * scale - m) * li;
+          st_acc[qt][r] = pt;                                   // P^T
+          dpt_acc[qt][r] = pt * (dpt_acc[qt][r] - dr) * scale;  // scale*dS^T
+        }
       }
     }
     // ---- dV = P^T dO : stage P^T, MFMA over q ------------------------------
-#pragma unroll 2
-    for (int qt = 0; qt < n_qtiles; ++qt)
 #pragma unroll
-      for (int r = 0; r < 4; ++r)
-        p_buf[((lane >> 4) * 4 + r) * PROW + qt * 16 + qcol_base] =
-            __hip_bfloat16(st_acc[qt][r]);
+    for (int qt = 0; qt < 16; ++qt) {
+      if (qt < n_qtiles) {
+  #pragma unroll
+        for (int r = 0; r < 4; ++r)
+          p_buf[((lane >> 4) * 4 + r) * PROW + qt * 16 + qcol_base] =
+              __hip_bfloat16(st_acc[qt][r]);
+      }
+    }
     f32x4 dv_acc[D / 16], dk_acc[D / 16];
 #pragma unroll
     for (int dt = 0; dt < D / 16; ++dt) {
@@ -419,12 +439,15 @@ void attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ qkv,
     // ---- dK = (scale dS)^T Q : stage dS^T, MFMA over q ---------------------
     // p_buf rewrite is wave-local; same-wave LDS ops stay ordered (no block
     // barrier here - waves have different kt trip counts).
-#pragma unroll 2
-    for (int qt = 0; qt < n_qtiles; ++qt)
 #pragma unroll
-      for (int r = 0; r < 4; ++r)
-        p_buf[((lane >> 4) * 4 + r) * PROW + qt * 16 + qcol_base] =
-            __hip_bfloat16(dpt_acc[qt][r]);
+    for (int qt = 0; qt < 16; ++qt) {
+      if (qt < n_qtiles) {
+  #pragma unroll
+        for (int r = 0; r < 4; ++r)
+          p_buf[((lane >> 4) * 4 + r) * PROW + qt * 16 + qcol_base] =
+              __hip_bfloat16(dpt_acc[qt][r]);
+      }
+    }
     for (int q2 = 0; q2 + 32 <= Npad; q2 += 32) {
       const int k0 = q2 + (lane >> 4) * 8;
       bf16x8 ds_frag = *(const bf16x8*)(&p_buf[(lane & 15) * PROW + k0]);
@@ -540,22 +563,24 @@ void attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ qkv,
       }
     }
     f32x4 s_acc[16], dp_acc[16];
-#pragma unroll 2
-    for (int kt = 0; kt < n_ktiles; ++kt) {
-      f32x4 sacc = {0.f, 0.f, 0.f, 0.f}, dacc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-      for (int sl = 0; sl < KSLICES; ++sl) {
-        const int key = kt * 16 + (lane & 15);
-        const int d0 = sl * 32 + (lane >> 4) * 8;
-        bf16x8 kb = *(const bf16x8*)(&k_lds[key * KPAD + d0]);
-        bf16x8 vb = *(const bf16x8*)(&v_lds[key * KPAD + d0]);
-        sacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(q_frag[sl], kb, sacc,
-                                                       0, 0, 0);
-        dacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(do_frag[sl], vb, dacc,
-                                                       0, 0, 0);
+    for (int kt = 0; kt < 16; ++kt) {
+      if (kt < n_ktiles) {
+        f32x4 sacc = {0.f, 0.f, 0.f, 0.f}, dacc = {0.f, 0.f, 0.f, 0.f};
+  #pragma unroll
+        for (int sl = 0; sl < KSLICES; ++sl) {
+          const int key = kt * 16 + (lane & 15);
+          const int d0 = sl * 32 + (lane >> 4) * 8;
+          bf16x8 kb = *(const bf16x8*)(&k_lds[key * KPAD + d0]);
+          bf16x8 vb = *(const bf16x8*)(&v_lds[key * KPAD + d0]);
+          sacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(q_frag[sl], kb, sacc,
+                                                         0, 0, 0);
+          dacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(do_frag[sl], vb, dacc,
+                                                         0, 0, 0);
+        }
+        s_acc[kt] = sacc;
+        dp_acc[kt] = dacc;
       }
-      s_acc[kt] = sacc;
-      dp_acc[kt] = dacc;
     }
     // ds = scale * P * (dP - Drow); C/D row = (lane>>4)*4+reg (query)
 #pragma unroll
@@ -564,19 +589,24 @@ void attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ qkv,
       const float m = qrow < N ? m_arr[qrow] : 0.f;
       const float li = qrow < N ? 1.f / l_arr[qrow] : 0.f;
       const float dr = qrow < N ? d_arr[qrow] : 0.f;
-#pragma unroll 2
-      for (int kt = 0; kt < n_ktiles; ++kt) {
-        const int key = kt * 16 + (lane & 15);
-        float pv = (key < N) ? __expf(s_acc[kt][r] * scale - m) * li : 0.f;
-        s_acc[kt][r] = pv * (dp_acc[kt][r] - dr) * scale;
+#pragma unroll
+      for (int kt = 0; kt < 16; ++kt) {
+        if (kt < n_ktiles) {
+          const int key = kt * 16 + (lane & 15);
+          float pv = (key < N) ? __expf(s_acc[kt][r] * scale - m) * li : 0.f;
+          s_acc[kt][r] = pv * (dp_acc[kt][r] - dr) * scale;
+        }
       }
     }
-#pragma unroll 2
-    for (int kt = 0; kt < n_ktiles; ++kt)
 #pragma unroll
-      for (int r = 0; r < 4; ++r)
-        p_buf[((lane >> 4) * 4 + r) * PROW + kt * 16 + (lane & 15)] =
-            __hip_bfloat16(s_acc[kt][r]);
+    for (int kt = 0; kt < 16; ++kt) {
+      if (kt < n_ktiles) {
+  #pragma unroll
+        for (int r = 0; r < 4; ++r)
+          p_buf[((lane >> 4) * 4 + r) * PROW + kt * 16 + (lane & 15)] =
+              __hip_bfloat16(s_acc[kt][r]);
+      }
+    }
     f32x4 dq_acc[D / 16];
 #pragma unroll
     for (int dt = 0; dt < D / 16; ++dt) dq_acc[dt] = {0.f, 0.f, 0.f, 0.f};

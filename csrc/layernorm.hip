@@ -271,14 +271,15 @@ __global__ void ln_bwd_smem_kernel(const dev_t* __restrict__ dy,
 // vs ~30us of traffic at ViT-B shapes).
 
 template <typename dev_t, int V>
-__global__ void ln_fwd_wave_kernel(const dev_t* __restrict__ x,
+__global__ __launch_bounds__(256) void ln_fwd_wave_kernel(const dev_t* __restrict__ x,
                                    const dev_t* __restrict__ w,
                                    const dev_t* __restrict__ b,
                                    dev_t* __restrict__ y,
                                    float* __restrict__ mean_out,
                                    float* __restrict__ rstd_out,
                                    int M, int C, float eps) {
-  constexpr int MAX_PL = 8;  // max vectors per lane
+  constexpr int MAX_PL = 8;  // max vectors per lane (compile-time bound:
+                             // runtime-indexed arrays would spill to scratch)
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int nwaves = blockDim.x >> 6;
@@ -288,14 +289,17 @@ __global__ void ln_fwd_wave_kernel(const dev_t* __restrict__ x,
     const dev_t* xr = x + (int64_t)r * C;
     dev_t* yr = y + (int64_t)r * C;
     float sum = 0.f, sumsq = 0.f;
-    int np = 0;
-    for (int i = lane * V; i < C; i += 64 * V, ++np) {
-      xv[np] = vload<dev_t, V>(xr + i);
 #pragma unroll
-      for (int j = 0; j < V; ++j) {
-        const float f = to_f32(xv[np].v[j]);
-        sum += f;
-        sumsq += f * f;
+    for (int p = 0; p < MAX_PL; ++p) {
+      const int i = lane * V + p * 64 * V;
+      if (i < C) {
+        xv[p] = vload<dev_t, V>(xr + i);
+  #pragma unroll
+        for (int j = 0; j < V; ++j) {
+          const float f = to_f32(xv[p].v[j]);
+          sum += f;
+          sumsq += f * f;
+        }
       }
     }
     sum = wave_reduce_sum(sum);
@@ -306,23 +310,26 @@ __global__ void ln_fwd_wave_kernel(const dev_t* __restrict__ x,
       mean_out[r] = mu;
       rstd_out[r] = rs;
     }
-    np = 0;
-    for (int i = lane * V; i < C; i += 64 * V, ++np) {
-      Vec<dev_t, V> wv = vload<dev_t, V>(w + i);
-      Vec<dev_t, V> bv = vload<dev_t, V>(b + i);
-      Vec<dev_t, V> yv;
 #pragma unroll
-      for (int j = 0; j < V; ++j)
-        yv.v[j] = from_f32<dev_t>((to_f32(xv[np].v[j]) - mu) * rs *
-                                      to_f32(wv.v[j]) +
-                                  to_f32(bv.v[j]));
-      vstore<dev_t, V>(yr + i, yv);
+    for (int p = 0; p < MAX_PL; ++p) {
+      const int i = lane * V + p * 64 * V;
+      if (i < C) {
+        Vec<dev_t, V> wv = vload<dev_t, V>(w + i);
+        Vec<dev_t, V> bv = vload<dev_t, V>(b + i);
+        Vec<dev_t, V> yv;
+  #pragma unroll
+        for (int j = 0; j < V; ++j)
+          yv.v[j] = from_f32<dev_t>((to_f32(xv[p].v[j]) - mu) * rs *
+                                        to_f32(wv.v[j]) +
+                                    to_f32(bv.v[j]));
+        vstore<dev_t, V>(yr + i, yv);
+      }
     }
   }
 }
 
 template <typename dev_t, int V>
-__global__ void ln_bwd_wave_kernel(const dev_t* __restrict__ dy,
+__global__ __launch_bounds__(256) void ln_bwd_wave_kernel(const dev_t* __restrict__ dy,
                                    const dev_t* __restrict__ x,
                                    const dev_t* __restrict__ w,
                                    const float* __restrict__ mean,
@@ -330,21 +337,27 @@ __global__ void ln_bwd_wave_kernel(const dev_t* __restrict__ dy,
                                    dev_t* __restrict__ dx,
                                    float* __restrict__ dw,
                                    float* __restrict__ db, int M, int C) {
-  constexpr int MAX_PL = 8;
+  // dgamma/dbeta accumulate in per-lane REGISTERS (each lane owns fixed
+  // channels); one partial row per (block, wave) is written at the end and
+  // summed by the host (no LDS or global atomics anywhere).
+  constexpr int MAX_PL = 4;  // caps C at 64*V*4 (bf16: 2048); keeps the
+                             // 2*MAX_PL*V accumulator regs affordable
   extern __shared__ __attribute__((aligned(16))) float lds[];
-  float* wsh = lds;      // C (fp32 weight, shared by all waves)
-  float* dgs = lds + C;  // C block dgamma
-  float* dbs = lds + 2 * C;  // C block dbeta
+  float* wsh = lds;  // C (fp32 weight, shared by all waves)
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int nwaves = blockDim.x >> 6;
-  for (int i = threadIdx.x; i < C; i += blockDim.x) {
-    wsh[i] = to_f32(w[i]);
-    dgs[i] = 0.f;
-    dbs[i] = 0.f;
-  }
+  for (int i = threadIdx.x; i < C; i += blockDim.x) wsh[i] = to_f32(w[i]);
   __syncthreads();
   Vec<dev_t, V> dyv[MAX_PL], xv[MAX_PL];
+  float acc_dg[MAX_PL][V], acc_db[MAX_PL][V];
+#pragma unroll
+  for (int p = 0; p < MAX_PL; ++p)
+#pragma unroll
+    for (int j = 0; j < V; ++j) {
+      acc_dg[p][j] = 0.f;
+      acc_db[p][j] = 0.f;
+    }
   for (int r = blockIdx.x * nwaves + wave; r < M;
        r += gridDim.x * nwaves) {
     const dev_t* dyr = dy + (int64_t)r * C;
@@ -352,40 +365,52 @@ __global__ void ln_bwd_wave_kernel(const dev_t* __restrict__ dy,
     dev_t* dxr = dx + (int64_t)r * C;
     const float mu = mean[r], rs = rstd[r];
     float s1 = 0.f, s2 = 0.f;
-    int np = 0;
-    for (int i = lane * V; i < C; i += 64 * V, ++np) {
-      dyv[np] = vload<dev_t, V>(dyr + i);
-      xv[np] = vload<dev_t, V>(xr + i);
 #pragma unroll
-      for (int j = 0; j < V; ++j) {
-        const float gw = to_f32(dyv[np].v[j]) * wsh[i + j];
-        const float xh = (to_f32(xv[np].v[j]) - mu) * rs;
-        s1 += gw;
-        s2 += gw * xh;
+    for (int p = 0; p < MAX_PL; ++p) {
+      const int i = lane * V + p * 64 * V;
+      if (i < C) {
+        dyv[p] = vload<dev_t, V>(dyr + i);
+        xv[p] = vload<dev_t, V>(xr + i);
+  #pragma unroll
+        for (int j = 0; j < V; ++j) {
+          const float gw = to_f32(dyv[p].v[j]) * wsh[i + j];
+          const float xh = (to_f32(xv[p].v[j]) - mu) * rs;
+          s1 += gw;
+          s2 += gw * xh;
+        }
       }
     }
     s1 = wave_reduce_sum(s1) / C;
     s2 = wave_reduce_sum(s2) / C;
-    np = 0;
-    for (int i = lane * V; i < C; i += 64 * V, ++np) {
-      Vec<dev_t, V> dxv;
 #pragma unroll
-      for (int j = 0; j < V; ++j) {
-        const float g = to_f32(dyv[np].v[j]);
-        const float xh = (to_f32(xv[np].v[j]) - mu) * rs;
-        dxv.v[j] = from_f32<dev_t>(rs * (g * wsh[i + j] - s1 - xh * s2));
-        atomicAdd(&dgs[i + j], g * xh);  // LDS atomic, cross-wave
-        atomicAdd(&dbs[i + j], g);
+    for (int p = 0; p < MAX_PL; ++p) {
+      const int i = lane * V + p * 64 * V;
+      if (i < C) {
+        Vec<dev_t, V> dxv;
+  #pragma unroll
+        for (int j = 0; j < V; ++j) {
+          const float g = to_f32(dyv[p].v[j]);
+          const float xh = (to_f32(xv[p].v[j]) - mu) * rs;
+          dxv.v[j] = from_f32<dev_t>(rs * (g * wsh[i + j] - s1 - xh * s2));
+          acc_dg[p][j] += g * xh;
+          acc_db[p][j] += g;
+        }
+        vstore<dev_t, V>(dxr + i, dxv);
       }
-      vstore<dev_t, V>(dxr + i, dxv);
     }
   }
-  __syncthreads();
-  // atomic-free flush: per-block partials, reduced by col_sum_2_kernel.
-  // (global atomicAdd from kMaxGrid blocks serializes per cache line.)
-  for (int i = threadIdx.x; i < C; i += blockDim.x) {
-    dw[(int64_t)blockIdx.x * C + i] = dgs[i];
-    db[(int64_t)blockIdx.x * C + i] = dbs[i];
+  // flush one partial row per (block, wave)
+  const int64_t prow = ((int64_t)blockIdx.x * nwaves + wave) * C;
+#pragma unroll
+  for (int p = 0; p < MAX_PL; ++p) {
+    const int i = lane * V + p * 64 * V;
+    if (i < C) {
+  #pragma unroll
+      for (int j = 0; j < V; ++j) {
+        dw[prow + i + j] = acc_dg[p][j];
+        db[prow + i + j] = acc_db[p][j];
+      }
+    }
   }
 }
 
@@ -471,13 +496,13 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
     constexpr int VMAX = 16 / (int)sizeof(dev_t);
     auto launch = [&](auto vtag) {
       constexpr int V = decltype(vtag)::value;
-      if (C <= 64 * V * 8 && C <= 8192) {  // wave-per-row fused path
+      if (C <= 64 * V * 4) {  // wave-per-row fused path
         const int nwaves = 4;
         const int g = (int)std::min<int64_t>((M + nwaves - 1) / nwaves,
                                              dla::kMaxGrid);
-        const int lds = 3 * C * sizeof(float);
-        auto part_dw = torch::empty({g, C}, dw.options());
-        auto part_db = torch::empty({g, C}, db.options());
+        const int lds = C * sizeof(float);
+        auto part_dw = torch::empty({g * nwaves, C}, dw.options());
+        auto part_db = torch::empty({g * nwaves, C}, db.options());
         hipLaunchKernelGGL((dla::ln_bwd_wave_kernel<dev_t, V>), dim3(g),
                            dim3(nwaves * 64), lds, dla::stream(),
                            (const dev_t*)dy.data_ptr(), (const dev_t*)x.data_ptr(),
@@ -485,10 +510,8 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
                            rstd.data_ptr<float>(), (dev_t*)dx.data_ptr(),
                            part_dw.data_ptr<float>(), part_db.data_ptr<float>(),
                            (int)M, C);
-        hipLaunchKernelGGL((dla::col_sum_2_kernel), dim3((C + 255) / 256),
-                           dim3(256), 0, dla::stream(),
-                           part_dw.data_ptr<float>(), part_db.data_ptr<float>(),
-                           dw.data_ptr<float>(), db.data_ptr<float>(), g, C);
+        at::sum_out(dw, part_dw, {0});  // torch's tuned column reduce
+        at::sum_out(db, part_db, {0});
         return;
       }
       if (C <= 3072) {  // fused dx + dgamma/dbeta path (5C+16 floats of LDS)
