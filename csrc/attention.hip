@@ -1,4 +1,4 @@
-// Fused multi-head attention forward for CDNA4 (gfx950).
+// Fused multi-head attention forward + backward for CDNA4 (gfx950).
 //
 // Replaces the eager softmax(Q K^T / sqrt(d) + bias + mask) V chain in
 // ViT (classification/vision_transformer/vit_model.py:88-113), Swin window
@@ -7,15 +7,21 @@
 //  - consumes the packed qkv projection [B, N, 3, H, d] directly (no permute
 //    copies, no [B,H,N,N] fp32 score tensor, no separate softmax kernels);
 //  - writes O in [B, N, H*d] (the layout the output projection wants);
-//  - optionally writes P = softmax(S) in bf16 for the backward pass.
+//  - backward recomputes P in-LDS from saved per-row softmax stats (m, l) —
+//    no [B,H,N,N] tensor ever reaches HBM (optionally saves P in bf16 for
+//    the bias-gradient fallback path).
 //
 // Shape domain: d in {32, 64}, N <= 256 (ViT 197, Swin windows 49, MAE 50).
-// Design: one 256-thread workgroup (4 waves) per (batch, head). K is staged
-// in LDS [N][d] (row stride padded to dodge bank conflicts), V transposed in
-// LDS [d][Npad]. Each wave owns 16 query rows at a time:
-//   S(16xN) via mfma_f32_16x16x32_bf16 (A = Q rows, B = K^T from LDS),
-//   row softmax with shuffle reductions over the 16-lane C/D columns,
-//   P -> LDS bf16, O(16xd) via mfma (A = P, B = V^T from LDS).
+// Design notes (rocprof-driven):
+//  - one workgroup per (b, h); 8 waves (4 for small N), each owning 16
+//    output rows at a time;
+//  - K staged in LDS [Npad][72] (72*2B = 36-dword row stride: 16-lane
+//    B-fragment reads hit 16 distinct banks), V transposed in LDS [d][Npad+8];
+//  - P/dS round-trips through a tiny per-wave LDS chunk (16x72) and the
+//    O/dV/dK MFMAs accumulate chunk-by-chunk, keeping LDS small enough for
+//    multiple waves per SIMD;
+//  - every accumulator loop is fully unrolled with compile-time bounds:
+//    a runtime-indexed register array spills to scratch (measured 60x).
 #include "common.h"
 #include "vec.h"
 
@@ -24,13 +30,11 @@ namespace dla {
 using bf16x8 = __attribute__((ext_vector_type(8))) short;
 using f32x4 = __attribute__((ext_vector_type(4))) float;
 
-// K LDS row padding (bf16 elements) so B-fragment reads (16 lanes x 16B,
-// row stride apart) hit distinct banks: stride 72*2B=144B -> 36 dwords,
-// 36*k mod 64 distinct for k in [0,16).
-constexpr int KPAD = 72;
+constexpr int KPAD = 72;         // K/Q/dO LDS row stride (bf16 elems)
+constexpr int CHUNK = 64;        // keys/queries per P-staging chunk (4 tiles)
+constexpr int PBUF = 16 * KPAD;  // per-wave P chunk buffer elems
 
 __device__ __forceinline__ float wave16_max(float v) {
-  // max over the 16-lane group (lane bits 0..3)
 #pragma unroll
   for (int off = 8; off > 0; off >>= 1) v = fmaxf(v, __shfl_xor(v, off, 64));
   return v;
@@ -42,102 +46,101 @@ __device__ __forceinline__ float wave16_sum(float v) {
   return v;
 }
 
-// One workgroup per (b, h); template D in {32, 64}.
+// Stage rows [N][D] of a (possibly strided) bf16 source into LDS row-major
+// (row stride KPAD) and/or transposed (dst_t[d][VROW]).
+template <int D>
+__device__ void stage_rows(const __hip_bfloat16* src, int64_t row_stride,
+                           __hip_bfloat16* dst_rows, __hip_bfloat16* dst_t,
+                           int N, int Npad, int VROW) {
+  for (int n = threadIdx.x / (D / 4); n < Npad;
+       n += blockDim.x / (D / 4)) {
+    const int d0 = (threadIdx.x % (D / 4)) * 4;
+    Vec<__hip_bfloat16, 4> v;
+    if (n < N) {
+      v = vload<__hip_bfloat16, 4>(src + n * row_stride + d0);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) v.v[j] = __hip_bfloat16(0.f);
+    }
+    if (dst_rows) vstore<__hip_bfloat16, 4>(&dst_rows[n * KPAD + d0], v);
+    if (dst_t) {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) dst_t[(d0 + j) * VROW + n] = v.v[j];
+    }
+  }
+}
+
+// ---------------------------------------------------------------- forward --
 template <int D, bool HAS_BIAS, bool HAS_MASK, bool SAVE_P>
-__global__ __launch_bounds__(256)
+__global__ __launch_bounds__(512)
 void attn_fwd_kernel(const __hip_bfloat16* __restrict__ qkv,  // [B,N,3,H,D]
-                     const float* __restrict__ bias,          // [H,N,N] or null
-                     const float* __restrict__ mask,          // [nW,N,N] or null
+                     const float* __restrict__ bias,          // [H,N,N]|null
+                     const float* __restrict__ mask,          // [nW,N,N]|null
                      __hip_bfloat16* __restrict__ out,        // [B,N,H*D]
-                     __hip_bfloat16* __restrict__ p_out,      // [B,H,N,N] or null
-                     float* __restrict__ stats,               // [2,B,H,N] or null
+                     __hip_bfloat16* __restrict__ p_out,      // [B,H,N,N]|null
+                     float* __restrict__ stats,               // [2,B,H,N]|null
                      int B, int N, int H, int n_win, float scale) {
-  constexpr int KSLICES = D / 32;        // mfma K-steps over head dim
+  constexpr int KSLICES = D / 32;
   const int b = blockIdx.x / H;
   const int h = blockIdx.x % H;
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  const int Npad = (N + 15) & ~15;       // key tiles padded to 16
+  const int nwaves = blockDim.x >> 6;
+  const int Npad = (N + 15) & ~15;
   const int n_ktiles = Npad / 16;
 
   extern __shared__ __attribute__((aligned(16))) char lds_raw[];
-  __hip_bfloat16* k_lds = (__hip_bfloat16*)lds_raw;            // [Npad][KPAD]
-  __hip_bfloat16* vt_lds = k_lds + Npad * KPAD;                // [D][Npad+8]
+  __hip_bfloat16* k_lds = (__hip_bfloat16*)lds_raw;       // [Npad][KPAD]
+  __hip_bfloat16* vt_lds = k_lds + Npad * KPAD;           // [D][VROW]
   const int VROW = Npad + 8;
-  const int PROW = Npad + 8;  // P row stride (Npad keys + bank-dodge pad)
-  __hip_bfloat16* p_lds = vt_lds + D * VROW;                   // [4][16][PROW]
+  __hip_bfloat16* p_lds = vt_lds + D * VROW;              // [nwaves][PBUF]
 
-  // ---- stage K ([N][D] rows) and V transposed ([D][Npad]) -------------------
-  // qkv element (b, n, c, h, d) at (((b*N + n)*3 + c)*H + h)*D + d
   const int64_t bh_stride = (int64_t)3 * H * D;
-  const __hip_bfloat16* k_src = qkv + ((int64_t)b * N * 3 + 1) * H * D +
-                                (int64_t)h * D;
-  const __hip_bfloat16* v_src = qkv + ((int64_t)b * N * 3 + 2) * H * D +
-                                (int64_t)h * D;
-  for (int n = threadIdx.x / (D / 4); n < Npad; n += 256 / (D / 4)) {
-    const int d0 = (threadIdx.x % (D / 4)) * 4;
-    Vec<__hip_bfloat16, 4> kv, vv;
-    if (n < N) {
-      kv = vload<__hip_bfloat16, 4>(k_src + n * bh_stride + d0);
-      vv = vload<__hip_bfloat16, 4>(v_src + n * bh_stride + d0);
-    } else {
-#pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        kv.v[j] = __hip_bfloat16(0.f);
-        vv.v[j] = __hip_bfloat16(0.f);
-      }
-    }
-    vstore<__hip_bfloat16, 4>(&k_lds[n * KPAD + d0], kv);
-#pragma unroll
-    for (int j = 0; j < 4; ++j) vt_lds[(d0 + j) * VROW + n] = vv.v[j];
-  }
+  stage_rows<D>(qkv + ((int64_t)b * N * 3 + 1) * H * D + (int64_t)h * D,
+                bh_stride, k_lds, nullptr, N, Npad, VROW);
+  stage_rows<D>(qkv + ((int64_t)b * N * 3 + 2) * H * D + (int64_t)h * D,
+                bh_stride, nullptr, vt_lds, N, Npad, VROW);
   __syncthreads();
 
   const __hip_bfloat16* q_src = qkv + (int64_t)b * N * 3 * H * D +
                                 (int64_t)h * D;
+  __hip_bfloat16* p_buf = p_lds + wave * PBUF;
   const int n_qblocks = (N + 15) / 16;
 
-  for (int qb = wave; qb < n_qblocks; qb += 4) {
+  for (int qb = wave; qb < n_qblocks; qb += nwaves) {
     const int q0 = qb * 16;
-    // ---- load Q fragment: A of mfma = Q[16 rows][32 k], row=lane&15,
-    // k=(lane>>4)*8+j  -> per lane 8 consecutive d -> one 16B load per slice
     bf16x8 q_frag[KSLICES];
     {
       const int row = q0 + (lane & 15);
       const int srow = row < N ? row : N - 1;
 #pragma unroll
-      for (int s = 0; s < KSLICES; ++s) {
-        const int d0 = s * 32 + (lane >> 4) * 8;
-        q_frag[s] = *(const bf16x8*)(q_src + srow * bh_stride + d0);
+      for (int sl = 0; sl < KSLICES; ++sl) {
+        const int d0 = sl * 32 + (lane >> 4) * 8;
+        q_frag[sl] = *(const bf16x8*)(q_src + srow * bh_stride + d0);
       }
     }
 
-    // ---- S = scale * Q K^T over all key tiles -------------------------------
-    // NOTE: every loop over s_acc is fully unrolled with a compile-time bound
-    // so the accumulator array stays in VGPRs (a runtime-indexed array goes
-    // to scratch = global memory; measured 60x slowdown).
-    f32x4 s_acc[16];  // up to 16 key tiles (N<=256)
+    // S = Q K^T over all key tiles (kept in VGPRs)
+    f32x4 s_acc[16];
 #pragma unroll
     for (int kt = 0; kt < 16; ++kt) {
       if (kt < n_ktiles) {
         f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-  #pragma unroll
-        for (int s = 0; s < KSLICES; ++s) {
-          // B fragment: col=lane&15 (key), k=(lane>>4)*8+j
+#pragma unroll
+        for (int sl = 0; sl < KSLICES; ++sl) {
           const int key = kt * 16 + (lane & 15);
-          const int d0 = s * 32 + (lane >> 4) * 8;
+          const int d0 = sl * 32 + (lane >> 4) * 8;
           bf16x8 k_frag = *(const bf16x8*)(&k_lds[key * KPAD + d0]);
-          acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(q_frag[s], k_frag, acc,
-                                                        0, 0, 0);
+          acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(q_frag[sl], k_frag,
+                                                        acc, 0, 0, 0);
         }
         s_acc[kt] = acc;
       }
     }
 
-    // ---- softmax over each of the 16 rows this lane-group covers ------------
-    // C/D: col = lane&15 (key within tile), row = (lane>>4)*4 + reg
-    const int col_in_tile = lane & 15;
-    float row_max[4], row_sum[4];
+    // softmax rows (C/D: col = key = lane&15, row = (lane>>4)*4 + reg)
+    const int col = lane & 15;
+    float row_sum[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       float m = -INFINITY;
@@ -145,15 +148,16 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ qkv,  // [B,N,3,H,D]
 #pragma unroll
       for (int kt = 0; kt < 16; ++kt) {
         if (kt < n_ktiles) {
-          const int key = kt * 16 + col_in_tile;
-          float s = s_acc[kt][r] * scale;
+          const int key = kt * 16 + col;
+          float sv = s_acc[kt][r] * scale;
           if ((HAS_BIAS || HAS_MASK) && key < N && qrow < N) {
-            if (HAS_BIAS) s += bias[((int64_t)h * N + qrow) * N + key];
-            if (HAS_MASK) s += mask[((int64_t)(b % n_win) * N + qrow) * N + key];
+            if (HAS_BIAS) sv += bias[((int64_t)h * N + qrow) * N + key];
+            if (HAS_MASK)
+              sv += mask[((int64_t)(b % n_win) * N + qrow) * N + key];
           }
-          if (key >= N) s = -INFINITY;
-          s_acc[kt][r] = s;
-          m = fmaxf(m, s);
+          if (key >= N) sv = -INFINITY;
+          s_acc[kt][r] = sv;
+          m = fmaxf(m, sv);
         }
       }
       m = wave16_max(m);
@@ -166,40 +170,20 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ qkv,  // [B,N,3,H,D]
           sum += p;
         }
       }
-      row_max[r] = m;
       row_sum[r] = wave16_sum(sum);
-      if (stats != nullptr && (lane & 15) == 0) {
-        const int qrow = q0 + (lane >> 4) * 4 + r;
-        if (qrow < N) {
-          stats[((int64_t)b * H + h) * N + qrow] = m;
-          stats[((int64_t)(B + b) * H + h) * N + qrow] = row_sum[r];
-        }
-      }
-    }
-
-    // ---- P -> LDS (bf16), per-wave buffer [16 rows][KPAD row stride] --------
-    __hip_bfloat16* p_buf = p_lds + wave * 16 * PROW;
-    // rows are interleaved across lanes; each lane writes its 4 elements/tile
-#pragma unroll
-    for (int kt = 0; kt < 16; ++kt) {
-      if (kt < n_ktiles) {
-  #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int row = (lane >> 4) * 4 + r;
-          p_buf[row * PROW + kt * 16 + col_in_tile] =
-              __hip_bfloat16(s_acc[kt][r]);
-        }
+      if (stats != nullptr && col == 0 && qrow < N) {
+        stats[((int64_t)b * H + h) * N + qrow] = m;
+        stats[((int64_t)(B + b) * H + h) * N + qrow] = row_sum[r];
       }
     }
     if (SAVE_P) {
-      const int qrow_base = q0;
 #pragma unroll
       for (int kt = 0; kt < 16; ++kt) {
         if (kt < n_ktiles) {
-  #pragma unroll
+#pragma unroll
           for (int r = 0; r < 4; ++r) {
-            const int qrow = qrow_base + (lane >> 4) * 4 + r;
-            const int key = kt * 16 + col_in_tile;
+            const int qrow = q0 + (lane >> 4) * 4 + r;
+            const int key = kt * 16 + col;
             if (qrow < N && key < N)
               p_out[(((int64_t)b * H + h) * N + qrow) * N + key] =
                   __hip_bfloat16(s_acc[kt][r] / row_sum[r]);
@@ -207,52 +191,48 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ qkv,  // [B,N,3,H,D]
         }
       }
     }
-    // no cross-wave LDS sharing of p_buf: same wave writes then reads.
-    // s_waitcnt lgkmcnt is enough; compiler inserts it for LDS dependences.
 
-    // ---- O = P V : A = P[16 rows][32 keys], B = V^T[32 keys][16 d] ----------
+    // O = P V, accumulated chunk by chunk through the small P buffer
     f32x4 o_acc[D / 16];
 #pragma unroll
     for (int dt = 0; dt < D / 16; ++dt) o_acc[dt] = {0.f, 0.f, 0.f, 0.f};
-    for (int kt2 = 0; kt2 < Npad / 32; ++kt2) {
-      // A fragment: row=lane&15, k=(lane>>4)*8+j  (keys)
-      const int row = lane & 15;
-      const int k0 = kt2 * 32 + (lane >> 4) * 8;
-      bf16x8 p_frag = *(const bf16x8*)(&p_buf[row * PROW + k0]);
 #pragma unroll
-      for (int dt = 0; dt < D / 16; ++dt) {
-        // B fragment: col=lane&15 (d), k=(lane>>4)*8+j (keys) from vt_lds
-        const int d = dt * 16 + (lane & 15);
-        bf16x8 v_frag = *(const bf16x8*)(&vt_lds[d * VROW + k0]);
-        o_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(p_frag, v_frag,
-                                                            o_acc[dt], 0, 0, 0);
-      }
-    }
-    // Npad may have a 16-key tail not covered by the 32-key loop
-    if (Npad % 32) {
-      const int row = lane & 15;
-      const int k0 = (Npad / 32) * 32 + (lane >> 4) * 8;
-      bf16x8 p_frag{};
-      if ((lane >> 4) * 8 < 16)  // only first 16 keys valid in this half tile
-        p_frag = *(const bf16x8*)(&p_buf[row * PROW + k0]);
-      else
+    for (int c = 0; c < 4; ++c) {  // chunks of 4 key tiles (64 keys)
+      const int t0 = c * 4;
+      if (t0 < n_ktiles) {
+        const int nt = n_ktiles - t0 < 4 ? n_ktiles - t0 : 4;
 #pragma unroll
-        for (int j = 0; j < 8; ++j) p_frag[j] = 0;
+        for (int tt = 0; tt < 4; ++tt) {
+          if (tt < nt) {
 #pragma unroll
-      for (int dt = 0; dt < D / 16; ++dt) {
-        const int d = dt * 16 + (lane & 15);
-        bf16x8 v_frag{};
-        if ((lane >> 4) * 8 < 16)
-          v_frag = *(const bf16x8*)(&vt_lds[d * VROW + k0]);
-        else
+            for (int r = 0; r < 4; ++r)
+              p_buf[((lane >> 4) * 4 + r) * KPAD + tt * 16 + col] =
+                  __hip_bfloat16(s_acc[t0 + tt][r]);
+          }
+        }
+        // MFMA over the chunk: pairs of tiles = 32-key K-steps
 #pragma unroll
-          for (int j = 0; j < 8; ++j) v_frag[j] = 0;
-        o_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(p_frag, v_frag,
-                                                            o_acc[dt], 0, 0, 0);
+        for (int kk = 0; kk < 2; ++kk) {
+          if (kk * 32 < nt * 16) {
+            const int k0 = kk * 32 + (lane >> 4) * 8;
+            const bool valid = k0 < nt * 16;
+            bf16x8 p_frag{};
+            if (valid)
+              p_frag = *(const bf16x8*)(&p_buf[(lane & 15) * KPAD + k0]);
+#pragma unroll
+            for (int dt = 0; dt < D / 16; ++dt) {
+              const int d = dt * 16 + (lane & 15);
+              bf16x8 v_frag{};
+              if (valid)
+                v_frag = *(const bf16x8*)(&vt_lds[d * VROW + c * CHUNK + k0]);
+              o_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  p_frag, v_frag, o_acc[dt], 0, 0, 0);
+            }
+          }
+        }
       }
     }
 
-    // ---- write O / row_sum normalize: out[b][qrow][h*D + d] -----------------
 #pragma unroll
     for (int dt = 0; dt < D / 16; ++dt) {
 #pragma unroll
@@ -267,25 +247,19 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ qkv,  // [B,N,3,H,D]
   }
 }
 
-// ---- fused attention backward (no bias/mask; flash-style recompute) --------
-// Saved from fwd: row max m and row sum l ([B,H,N] f32 each). D_row =
-// rowsum(dO * O) is computed by the host in one fused reduce.
-//   P = exp(scale*QK^T - m)/l
-//   dV = P^T dO ; dP = dO V^T ; dS = P*(dP - D_row) ; dQ = scale*dS K ;
-//   dK = scale*dS^T Q
-// Kernel A (dK/dV): per (b,h) workgroup; Q and dO staged in LDS both
-// row-major (B-fragments of S^T / dP^T) and transposed (B-fragments of the
-// dK / dV MFMAs); each wave owns 16 keys per iteration, K/V fragments read
-// straight from global.
-// Kernel B (dQ): K and V row-major + K transposed in LDS; each wave owns 16
-// queries.
+// --------------------------------------------------------------- backward --
+// dV = P^T dO ; dP = dO V^T ; dS = P*(dP - D_row) ; dQ = scale*dS K ;
+// dK = scale*dS^T Q.  P recomputed from (m, l); D_row precomputed by host.
 
+// Kernel A: dK and dV. Each wave owns 16 keys; S^T / dP^T computed per
+// 64-query chunk and pushed through the small P buffer into the dV / dK
+// MFMA accumulators.
 template <int D>
-__global__ __launch_bounds__(256)
+__global__ __launch_bounds__(512)
 void attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ qkv,
                         const __hip_bfloat16* __restrict__ dout,  // [B,N,H*D]
-                        const float* __restrict__ stats,          // [2,B,H,N]
-                        const float* __restrict__ drow,           // [B,H,N]
+                        const float* __restrict__ stats,
+                        const float* __restrict__ drow,
                         __hip_bfloat16* __restrict__ dqkv,
                         int B, int N, int H, float scale) {
   constexpr int KSLICES = D / 32;
@@ -293,44 +267,23 @@ void attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ qkv,
   const int h = blockIdx.x % H;
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
+  const int nwaves = blockDim.x >> 6;
   const int Npad = (N + 15) & ~15;
   const int n_qtiles = Npad / 16;
 
   extern __shared__ __attribute__((aligned(16))) char lds_raw[];
-  __hip_bfloat16* q_lds = (__hip_bfloat16*)lds_raw;        // [Npad][KPAD]
-  __hip_bfloat16* do_lds = q_lds + Npad * KPAD;            // [Npad][KPAD]
+  __hip_bfloat16* q_lds = (__hip_bfloat16*)lds_raw;       // [Npad][KPAD]
+  __hip_bfloat16* do_lds = q_lds + Npad * KPAD;           // [Npad][KPAD]
   const int VROW = Npad + 8;
-  __hip_bfloat16* qt_lds = do_lds + Npad * KPAD;           // [D][VROW]
-  __hip_bfloat16* dot_lds = qt_lds + D * VROW;             // [D][VROW]
-  const int PROW = Npad + 8;
-  __hip_bfloat16* p_lds = dot_lds + D * VROW;              // [4][16][PROW]
+  __hip_bfloat16* qt_lds = do_lds + Npad * KPAD;          // [D][VROW]
+  __hip_bfloat16* dot_lds = qt_lds + D * VROW;            // [D][VROW]
+  __hip_bfloat16* p_lds = dot_lds + D * VROW;             // [nwaves][PBUF]
 
   const int64_t bh_stride = (int64_t)3 * H * D;
-  const __hip_bfloat16* q_src = qkv + (int64_t)b * N * 3 * H * D +
-                                (int64_t)h * D;
-  const __hip_bfloat16* do_src = dout + (int64_t)b * N * H * D +
-                                 (int64_t)h * D;
-  for (int n = threadIdx.x / (D / 4); n < Npad; n += 256 / (D / 4)) {
-    const int d0 = (threadIdx.x % (D / 4)) * 4;
-    Vec<__hip_bfloat16, 4> qv, dv;
-    if (n < N) {
-      qv = vload<__hip_bfloat16, 4>(q_src + n * bh_stride + d0);
-      dv = vload<__hip_bfloat16, 4>(do_src + (int64_t)n * H * D + d0);
-    } else {
-#pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        qv.v[j] = __hip_bfloat16(0.f);
-        dv.v[j] = __hip_bfloat16(0.f);
-      }
-    }
-    vstore<__hip_bfloat16, 4>(&q_lds[n * KPAD + d0], qv);
-    vstore<__hip_bfloat16, 4>(&do_lds[n * KPAD + d0], dv);
-#pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      qt_lds[(d0 + j) * VROW + n] = qv.v[j];
-      dot_lds[(d0 + j) * VROW + n] = dv.v[j];
-    }
-  }
+  stage_rows<D>(qkv + (int64_t)b * N * 3 * H * D + (int64_t)h * D,
+                bh_stride, q_lds, qt_lds, N, Npad, VROW);
+  stage_rows<D>(dout + (int64_t)b * N * H * D + (int64_t)h * D,
+                (int64_t)H * D, do_lds, dot_lds, N, Npad, VROW);
   __syncthreads();
 
   const float* m_arr = stats + ((int64_t)b * H + h) * N;
@@ -340,12 +293,10 @@ void attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ qkv,
                                 (int64_t)h * D;
   const __hip_bfloat16* v_src = qkv + ((int64_t)b * N * 3 + 2) * H * D +
                                 (int64_t)h * D;
-  __hip_bfloat16* p_buf = p_lds + wave * 16 * PROW;
-  const int n_ktiles_own = (N + 15) / 16;
+  __hip_bfloat16* p_buf = p_lds + wave * PBUF;
 
-  for (int kt = wave; kt < n_ktiles_own; kt += 4) {
+  for (int kt = wave; kt < (N + 15) / 16; kt += nwaves) {
     const int key0 = kt * 16;
-    // K_tile / V_tile A-fragments from global: row=key, k=d
     bf16x8 k_frag[KSLICES], v_frag[KSLICES];
     {
       const int key = key0 + (lane & 15);
@@ -357,123 +308,109 @@ void attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ qkv,
         v_frag[sl] = *(const bf16x8*)(v_src + skey * bh_stride + d0);
       }
     }
-    // per q-tile: S^T and dP^T accumulators (full unroll: keep in VGPRs)
-    f32x4 st_acc[16], dpt_acc[16];
-#pragma unroll
-    for (int qt = 0; qt < 16; ++qt) {
-      if (qt < n_qtiles) {
-        f32x4 sacc = {0.f, 0.f, 0.f, 0.f}, dacc = {0.f, 0.f, 0.f, 0.f};
-  #pragma unroll
-        for (int sl = 0; sl < KSLICES; ++sl) {
-          const int q = qt * 16 + (lane & 15);
-          const int d0 = sl * 32 + (lane >> 4) * 8;
-          bf16x8 qb = *(const bf16x8*)(&q_lds[q * KPAD + d0]);
-          bf16x8 db = *(const bf16x8*)(&do_lds[q * KPAD + d0]);
-          sacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(k_frag[sl], qb, sacc,
-                                                         0, 0, 0);
-          dacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(v_frag[sl], db, dacc,
-                                                         0, 0, 0);
-        }
-        st_acc[qt] = sacc;
-        dpt_acc[qt] = dacc;
-      }
-    }
-    // P^T and dS^T: C/D col = q (lane&15), row = key ((lane>>4)*4+reg)
-    const int qcol_base = lane & 15;
-#pragma unroll
-    for (int qt = 0; qt < 16; ++qt) {
-      if (qt < n_qtiles) {
-        const int q = qt * 16 + qcol_base;
-        const float m = q < N ? m_arr[q] : 0.f;
-        const float li = q < N ? 1.f / l_arr[q] : 0.f;
-        const float dr = q < N ? d_arr[q] : 0.f;
-  #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const float pt = __expf(st_acc[qt][r] * scale - m) * li;
-          st_acc[qt][r] = pt;                                   // P^T
-          dpt_acc[qt][r] = pt * (dpt_acc[qt][r] - dr) * scale;  // scale*dS^T
-        }
-      }
-    }
-    // ---- dV = P^T dO : stage P^T, MFMA over q ------------------------------
-#pragma unroll
-    for (int qt = 0; qt < 16; ++qt) {
-      if (qt < n_qtiles) {
-  #pragma unroll
-        for (int r = 0; r < 4; ++r)
-          p_buf[((lane >> 4) * 4 + r) * PROW + qt * 16 + qcol_base] =
-              __hip_bfloat16(st_acc[qt][r]);
-      }
-    }
     f32x4 dv_acc[D / 16], dk_acc[D / 16];
 #pragma unroll
     for (int dt = 0; dt < D / 16; ++dt) {
       dv_acc[dt] = {0.f, 0.f, 0.f, 0.f};
       dk_acc[dt] = {0.f, 0.f, 0.f, 0.f};
     }
-    for (int q2 = 0; q2 + 32 <= Npad; q2 += 32) {
-      const int k0 = q2 + (lane >> 4) * 8;
-      bf16x8 pt_frag = *(const bf16x8*)(&p_buf[(lane & 15) * PROW + k0]);
+
 #pragma unroll
-      for (int dt = 0; dt < D / 16; ++dt) {
-        const int d = dt * 16 + (lane & 15);
-        bf16x8 dob = *(const bf16x8*)(&dot_lds[d * VROW + k0]);
-        dv_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pt_frag, dob,
-                                                             dv_acc[dt], 0, 0, 0);
+    for (int c = 0; c < 4; ++c) {  // 64-query chunks
+      const int t0 = c * 4;
+      if (t0 < n_qtiles) {
+        const int nt = n_qtiles - t0 < 4 ? n_qtiles - t0 : 4;
+        f32x4 st[4], dpt[4];
+#pragma unroll
+        for (int tt = 0; tt < 4; ++tt) {
+          if (tt < nt) {
+            f32x4 sacc = {0.f, 0.f, 0.f, 0.f}, dacc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+            for (int sl = 0; sl < KSLICES; ++sl) {
+              const int q = (t0 + tt) * 16 + (lane & 15);
+              const int d0 = sl * 32 + (lane >> 4) * 8;
+              bf16x8 qb = *(const bf16x8*)(&q_lds[q * KPAD + d0]);
+              bf16x8 db = *(const bf16x8*)(&do_lds[q * KPAD + d0]);
+              sacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(k_frag[sl], qb,
+                                                             sacc, 0, 0, 0);
+              dacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(v_frag[sl], db,
+                                                             dacc, 0, 0, 0);
+            }
+            st[tt] = sacc;
+            dpt[tt] = dacc;
+          }
+        }
+        // P^T and scale*dS^T for this chunk; stage P^T, accumulate dV
+#pragma unroll
+        for (int tt = 0; tt < 4; ++tt) {
+          if (tt < nt) {
+            const int q = (t0 + tt) * 16 + (lane & 15);
+            const float m = q < N ? m_arr[q] : 0.f;
+            const float li = q < N ? 1.f / l_arr[q] : 0.f;
+            const float dr = q < N ? d_arr[q] : 0.f;
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+              const float pt = __expf(st[tt][r] * scale - m) * li;
+              st[tt][r] = pt;
+              dpt[tt][r] = pt * (dpt[tt][r] - dr) * scale;
+            }
+#pragma unroll
+            for (int r = 0; r < 4; ++r)
+              p_buf[((lane >> 4) * 4 + r) * KPAD + tt * 16 + (lane & 15)] =
+                  __hip_bfloat16(st[tt][r]);
+          }
+        }
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk) {
+          if (kk * 32 < nt * 16) {
+            const int k0 = kk * 32 + (lane >> 4) * 8;
+            const bool valid = k0 < nt * 16;
+            bf16x8 pt_frag{};
+            if (valid)
+              pt_frag = *(const bf16x8*)(&p_buf[(lane & 15) * KPAD + k0]);
+#pragma unroll
+            for (int dt = 0; dt < D / 16; ++dt) {
+              const int d = dt * 16 + (lane & 15);
+              bf16x8 dob{};
+              if (valid)
+                dob = *(const bf16x8*)(&dot_lds[d * VROW + c * CHUNK + k0]);
+              dv_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  pt_frag, dob, dv_acc[dt], 0, 0, 0);
+            }
+          }
+        }
+        // stage dS^T, accumulate dK
+#pragma unroll
+        for (int tt = 0; tt < 4; ++tt) {
+          if (tt < nt) {
+#pragma unroll
+            for (int r = 0; r < 4; ++r)
+              p_buf[((lane >> 4) * 4 + r) * KPAD + tt * 16 + (lane & 15)] =
+                  __hip_bfloat16(dpt[tt][r]);
+          }
+        }
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk) {
+          if (kk * 32 < nt * 16) {
+            const int k0 = kk * 32 + (lane >> 4) * 8;
+            const bool valid = k0 < nt * 16;
+            bf16x8 ds_frag{};
+            if (valid)
+              ds_frag = *(const bf16x8*)(&p_buf[(lane & 15) * KPAD + k0]);
+#pragma unroll
+            for (int dt = 0; dt < D / 16; ++dt) {
+              const int d = dt * 16 + (lane & 15);
+              bf16x8 qb{};
+              if (valid)
+                qb = *(const bf16x8*)(&qt_lds[d * VROW + c * CHUNK + k0]);
+              dk_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  ds_frag, qb, dk_acc[dt], 0, 0, 0);
+            }
+          }
+        }
       }
     }
-    if (Npad % 32) {  // 16-query half tile: upper k-halves masked to zero
-      const int k0 = (Npad & ~31) + (lane >> 4) * 8;
-      const bool lo = (lane >> 4) * 8 < 16;
-      bf16x8 pt_frag{};
-      if (lo) pt_frag = *(const bf16x8*)(&p_buf[(lane & 15) * PROW + k0]);
-#pragma unroll
-      for (int dt = 0; dt < D / 16; ++dt) {
-        const int d = dt * 16 + (lane & 15);
-        bf16x8 dob{};
-        if (lo) dob = *(const bf16x8*)(&dot_lds[d * VROW + k0]);
-        dv_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pt_frag, dob,
-                                                             dv_acc[dt], 0, 0, 0);
-      }
-    }
-    // ---- dK = (scale dS)^T Q : stage dS^T, MFMA over q ---------------------
-    // p_buf rewrite is wave-local; same-wave LDS ops stay ordered (no block
-    // barrier here - waves have different kt trip counts).
-#pragma unroll
-    for (int qt = 0; qt < 16; ++qt) {
-      if (qt < n_qtiles) {
-  #pragma unroll
-        for (int r = 0; r < 4; ++r)
-          p_buf[((lane >> 4) * 4 + r) * PROW + qt * 16 + qcol_base] =
-              __hip_bfloat16(dpt_acc[qt][r]);
-      }
-    }
-    for (int q2 = 0; q2 + 32 <= Npad; q2 += 32) {
-      const int k0 = q2 + (lane >> 4) * 8;
-      bf16x8 ds_frag = *(const bf16x8*)(&p_buf[(lane & 15) * PROW + k0]);
-#pragma unroll
-      for (int dt = 0; dt < D / 16; ++dt) {
-        const int d = dt * 16 + (lane & 15);
-        bf16x8 qb = *(const bf16x8*)(&qt_lds[d * VROW + k0]);
-        dk_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ds_frag, qb,
-                                                             dk_acc[dt], 0, 0, 0);
-      }
-    }
-    if (Npad % 32) {
-      const int k0 = (Npad & ~31) + (lane >> 4) * 8;
-      const bool lo = (lane >> 4) * 8 < 16;
-      bf16x8 ds_frag{};
-      if (lo) ds_frag = *(const bf16x8*)(&p_buf[(lane & 15) * PROW + k0]);
-#pragma unroll
-      for (int dt = 0; dt < D / 16; ++dt) {
-        const int d = dt * 16 + (lane & 15);
-        bf16x8 qb{};
-        if (lo) qb = *(const bf16x8*)(&qt_lds[d * VROW + k0]);
-        dk_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ds_frag, qb,
-                                                             dk_acc[dt], 0, 0, 0);
-      }
-    }
-    // ---- write dK (comp 1) and dV (comp 2): C/D row=key, col=d -------------
+
 #pragma unroll
     for (int dt = 0; dt < D / 16; ++dt) {
 #pragma unroll
@@ -490,8 +427,10 @@ void attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ qkv,
   }
 }
 
+// Kernel B: dQ. Each wave owns 16 queries; per 64-key chunk: S and dP
+// recomputed, dS staged, dQ accumulated.
 template <int D>
-__global__ __launch_bounds__(256)
+__global__ __launch_bounds__(512)
 void attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ qkv,
                        const __hip_bfloat16* __restrict__ dout,
                        const float* __restrict__ stats,
@@ -503,40 +442,22 @@ void attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ qkv,
   const int h = blockIdx.x % H;
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
+  const int nwaves = blockDim.x >> 6;
   const int Npad = (N + 15) & ~15;
   const int n_ktiles = Npad / 16;
 
   extern __shared__ __attribute__((aligned(16))) char lds_raw[];
-  __hip_bfloat16* k_lds = (__hip_bfloat16*)lds_raw;        // [Npad][KPAD]
-  __hip_bfloat16* v_lds = k_lds + Npad * KPAD;             // [Npad][KPAD]
+  __hip_bfloat16* k_lds = (__hip_bfloat16*)lds_raw;       // [Npad][KPAD]
+  __hip_bfloat16* v_lds = k_lds + Npad * KPAD;            // [Npad][KPAD]
   const int VROW = Npad + 8;
-  __hip_bfloat16* kt_lds = v_lds + Npad * KPAD;            // [D][VROW]
-  const int PROW = Npad + 8;
-  __hip_bfloat16* p_lds = kt_lds + D * VROW;               // [4][16][PROW]
+  __hip_bfloat16* kt_lds = v_lds + Npad * KPAD;           // [D][VROW]
+  __hip_bfloat16* p_lds = kt_lds + D * VROW;              // [nwaves][PBUF]
 
   const int64_t bh_stride = (int64_t)3 * H * D;
-  const __hip_bfloat16* k_src = qkv + ((int64_t)b * N * 3 + 1) * H * D +
-                                (int64_t)h * D;
-  const __hip_bfloat16* v_src = qkv + ((int64_t)b * N * 3 + 2) * H * D +
-                                (int64_t)h * D;
-  for (int n = threadIdx.x / (D / 4); n < Npad; n += 256 / (D / 4)) {
-    const int d0 = (threadIdx.x % (D / 4)) * 4;
-    Vec<__hip_bfloat16, 4> kv, vv;
-    if (n < N) {
-      kv = vload<__hip_bfloat16, 4>(k_src + n * bh_stride + d0);
-      vv = vload<__hip_bfloat16, 4>(v_src + n * bh_stride + d0);
-    } else {
-#pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        kv.v[j] = __hip_bfloat16(0.f);
-        vv.v[j] = __hip_bfloat16(0.f);
-      }
-    }
-    vstore<__hip_bfloat16, 4>(&k_lds[n * KPAD + d0], kv);
-    vstore<__hip_bfloat16, 4>(&v_lds[n * KPAD + d0], vv);
-#pragma unroll
-    for (int j = 0; j < 4; ++j) kt_lds[(d0 + j) * VROW + n] = kv.v[j];
-  }
+  stage_rows<D>(qkv + ((int64_t)b * N * 3 + 1) * H * D + (int64_t)h * D,
+                bh_stride, k_lds, kt_lds, N, Npad, VROW);
+  stage_rows<D>(qkv + ((int64_t)b * N * 3 + 2) * H * D + (int64_t)h * D,
+                bh_stride, v_lds, nullptr, N, Npad, VROW);
   __syncthreads();
 
   const float* m_arr = stats + ((int64_t)b * H + h) * N;
@@ -546,10 +467,9 @@ void attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ qkv,
                                 (int64_t)h * D;
   const __hip_bfloat16* do_src = dout + (int64_t)b * N * H * D +
                                  (int64_t)h * D;
-  __hip_bfloat16* p_buf = p_lds + wave * 16 * PROW;
-  const int n_qblocks = (N + 15) / 16;
+  __hip_bfloat16* p_buf = p_lds + wave * PBUF;
 
-  for (int qb = wave; qb < n_qblocks; qb += 4) {
+  for (int qb = wave; qb < (N + 15) / 16; qb += nwaves) {
     const int q0 = qb * 16;
     bf16x8 q_frag[KSLICES], do_frag[KSLICES];
     {
@@ -562,79 +482,81 @@ void attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ qkv,
         do_frag[sl] = *(const bf16x8*)(do_src + (int64_t)srow * H * D + d0);
       }
     }
-    f32x4 s_acc[16], dp_acc[16];
-#pragma unroll
-    for (int kt = 0; kt < 16; ++kt) {
-      if (kt < n_ktiles) {
-        f32x4 sacc = {0.f, 0.f, 0.f, 0.f}, dacc = {0.f, 0.f, 0.f, 0.f};
-  #pragma unroll
-        for (int sl = 0; sl < KSLICES; ++sl) {
-          const int key = kt * 16 + (lane & 15);
-          const int d0 = sl * 32 + (lane >> 4) * 8;
-          bf16x8 kb = *(const bf16x8*)(&k_lds[key * KPAD + d0]);
-          bf16x8 vb = *(const bf16x8*)(&v_lds[key * KPAD + d0]);
-          sacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(q_frag[sl], kb, sacc,
-                                                         0, 0, 0);
-          dacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(do_frag[sl], vb, dacc,
-                                                         0, 0, 0);
-        }
-        s_acc[kt] = sacc;
-        dp_acc[kt] = dacc;
-      }
-    }
-    // ds = scale * P * (dP - Drow); C/D row = (lane>>4)*4+reg (query)
+    float mr[4], lr[4], drr[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int qrow = q0 + (lane >> 4) * 4 + r;
-      const float m = qrow < N ? m_arr[qrow] : 0.f;
-      const float li = qrow < N ? 1.f / l_arr[qrow] : 0.f;
-      const float dr = qrow < N ? d_arr[qrow] : 0.f;
-#pragma unroll
-      for (int kt = 0; kt < 16; ++kt) {
-        if (kt < n_ktiles) {
-          const int key = kt * 16 + (lane & 15);
-          float pv = (key < N) ? __expf(s_acc[kt][r] * scale - m) * li : 0.f;
-          s_acc[kt][r] = pv * (dp_acc[kt][r] - dr) * scale;
-        }
-      }
-    }
-#pragma unroll
-    for (int kt = 0; kt < 16; ++kt) {
-      if (kt < n_ktiles) {
-  #pragma unroll
-        for (int r = 0; r < 4; ++r)
-          p_buf[((lane >> 4) * 4 + r) * PROW + kt * 16 + (lane & 15)] =
-              __hip_bfloat16(s_acc[kt][r]);
-      }
+      mr[r] = qrow < N ? m_arr[qrow] : 0.f;
+      lr[r] = qrow < N ? 1.f / l_arr[qrow] : 0.f;
+      drr[r] = qrow < N ? d_arr[qrow] : 0.f;
     }
     f32x4 dq_acc[D / 16];
 #pragma unroll
     for (int dt = 0; dt < D / 16; ++dt) dq_acc[dt] = {0.f, 0.f, 0.f, 0.f};
-    for (int k2 = 0; k2 + 32 <= Npad; k2 += 32) {
-      const int k0 = k2 + (lane >> 4) * 8;
-      bf16x8 ds_frag = *(const bf16x8*)(&p_buf[(lane & 15) * PROW + k0]);
+
 #pragma unroll
-      for (int dt = 0; dt < D / 16; ++dt) {
-        const int d = dt * 16 + (lane & 15);
-        bf16x8 kb = *(const bf16x8*)(&kt_lds[d * VROW + k0]);
-        dq_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ds_frag, kb,
-                                                             dq_acc[dt], 0, 0, 0);
+    for (int c = 0; c < 4; ++c) {  // 64-key chunks
+      const int t0 = c * 4;
+      if (t0 < n_ktiles) {
+        const int nt = n_ktiles - t0 < 4 ? n_ktiles - t0 : 4;
+        f32x4 st[4], dpt[4];
+#pragma unroll
+        for (int tt = 0; tt < 4; ++tt) {
+          if (tt < nt) {
+            f32x4 sacc = {0.f, 0.f, 0.f, 0.f}, dacc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+            for (int sl = 0; sl < KSLICES; ++sl) {
+              const int key = (t0 + tt) * 16 + (lane & 15);
+              const int d0 = sl * 32 + (lane >> 4) * 8;
+              bf16x8 kb = *(const bf16x8*)(&k_lds[key * KPAD + d0]);
+              bf16x8 vb = *(const bf16x8*)(&v_lds[key * KPAD + d0]);
+              sacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(q_frag[sl], kb,
+                                                             sacc, 0, 0, 0);
+              dacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(do_frag[sl], vb,
+                                                             dacc, 0, 0, 0);
+            }
+            st[tt] = sacc;
+            dpt[tt] = dacc;
+          }
+        }
+#pragma unroll
+        for (int tt = 0; tt < 4; ++tt) {
+          if (tt < nt) {
+            const int key = (t0 + tt) * 16 + (lane & 15);
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+              const float pv =
+                  key < N ? __expf(st[tt][r] * scale - mr[r]) * lr[r] : 0.f;
+              st[tt][r] = pv * (dpt[tt][r] - drr[r]) * scale;
+            }
+#pragma unroll
+            for (int r = 0; r < 4; ++r)
+              p_buf[((lane >> 4) * 4 + r) * KPAD + tt * 16 + (lane & 15)] =
+                  __hip_bfloat16(st[tt][r]);
+          }
+        }
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk) {
+          if (kk * 32 < nt * 16) {
+            const int k0 = kk * 32 + (lane >> 4) * 8;
+            const bool valid = k0 < nt * 16;
+            bf16x8 ds_frag{};
+            if (valid)
+              ds_frag = *(const bf16x8*)(&p_buf[(lane & 15) * KPAD + k0]);
+#pragma unroll
+            for (int dt = 0; dt < D / 16; ++dt) {
+              const int d = dt * 16 + (lane & 15);
+              bf16x8 kb{};
+              if (valid)
+                kb = *(const bf16x8*)(&kt_lds[d * VROW + c * CHUNK + k0]);
+              dq_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  ds_frag, kb, dq_acc[dt], 0, 0, 0);
+            }
+          }
+        }
       }
     }
-    if (Npad % 32) {
-      const int k0 = (Npad & ~31) + (lane >> 4) * 8;
-      const bool lo = (lane >> 4) * 8 < 16;
-      bf16x8 ds_frag{};
-      if (lo) ds_frag = *(const bf16x8*)(&p_buf[(lane & 15) * PROW + k0]);
-#pragma unroll
-      for (int dt = 0; dt < D / 16; ++dt) {
-        const int d = dt * 16 + (lane & 15);
-        bf16x8 kb{};
-        if (lo) kb = *(const bf16x8*)(&kt_lds[d * VROW + k0]);
-        dq_acc[dt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ds_frag, kb,
-                                                             dq_acc[dt], 0, 0, 0);
-      }
-    }
+
 #pragma unroll
     for (int dt = 0; dt < D / 16; ++dt) {
 #pragma unroll
@@ -678,7 +600,6 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor qkv, int64_t num_heads,
                                     c10::optional<torch::Tensor> bias,
                                     c10::optional<torch::Tensor> mask,
                                     bool save_p, bool save_stats) {
-  // qkv: [B, N, 3*H*D] or [B, N, 3, H, D] contiguous bf16
   TORCH_CHECK(qkv.scalar_type() == at::kBFloat16, "attn_fwd wants bf16 qkv");
   TORCH_CHECK(qkv.is_contiguous(), "attn_fwd wants contiguous qkv");
   const int B = (int)qkv.size(0);
@@ -702,20 +623,23 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor qkv, int64_t num_heads,
   torch::Tensor bias_f, mask_f;
   if (bias.has_value()) {
     bias_f = bias->to(torch::kFloat).contiguous();
-    TORCH_CHECK(bias_f.dim() == 3 && bias_f.size(1) == N, "bias must be [H,N,N]");
+    TORCH_CHECK(bias_f.dim() == 3 && bias_f.size(1) == N,
+                "bias must be [H,N,N]");
     bias_ptr = bias_f.data_ptr<float>();
   }
   if (mask.has_value()) {
     mask_f = mask->to(torch::kFloat).contiguous();
-    TORCH_CHECK(mask_f.dim() == 3 && mask_f.size(1) == N, "mask must be [nW,N,N]");
+    TORCH_CHECK(mask_f.dim() == 3 && mask_f.size(1) == N,
+                "mask must be [nW,N,N]");
     n_win = (int)mask_f.size(0);
     mask_ptr = mask_f.data_ptr<float>();
   }
 
   const int Npad = (N + 15) & ~15;
-  const int lds = (Npad * dla::KPAD + D * (Npad + 8) + 4 * 16 * (Npad + 8)) *
-                  sizeof(__hip_bfloat16);
-  dim3 grid(B * H), block(256);
+  const int nwaves = N > 96 ? 8 : 4;  // small-N: 4 waves cover all q blocks
+  const int lds = (Npad * dla::KPAD + D * (Npad + 8) + nwaves * dla::PBUF) *
+                  (int)sizeof(__hip_bfloat16);
+  dim3 grid(B * H), block(nwaves * 64);
 
   auto launch = [&](auto dtag, auto btag, auto mtag, auto ptag) {
     constexpr int DD = decltype(dtag)::value;
@@ -731,7 +655,6 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor qkv, int64_t num_heads,
                        B, N, H, n_win, (float)scale);
   };
   auto d3 = [&](auto dtag) {
-    constexpr int DD = decltype(dtag)::value;
     using T = std::true_type;
     using F = std::false_type;
     const bool bb = bias_ptr != nullptr, mm = mask_ptr != nullptr;
@@ -753,17 +676,6 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor qkv, int64_t num_heads,
   return res;
 }
 
-torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B) {
-  TORCH_CHECK(A.scalar_type() == at::kBFloat16 && B.scalar_type() == at::kBFloat16);
-  auto D = torch::zeros({16, 16}, A.options().dtype(torch::kFloat));
-  hipLaunchKernelGGL(dla::mfma_probe_kernel, dim3(1), dim3(64), 0,
-                     dla::stream(), (const __hip_bfloat16*)A.data_ptr(),
-                     (const __hip_bfloat16*)B.data_ptr(),
-                     D.data_ptr<float>());
-  HIP_CHECK_ERR();
-  return D;
-}
-
 std::vector<torch::Tensor> attn_bwd(torch::Tensor qkv, torch::Tensor dout,
                                     torch::Tensor stats, torch::Tensor drow,
                                     int64_t num_heads, double scale) {
@@ -775,11 +687,12 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor qkv, torch::Tensor dout,
   TORCH_CHECK(D == 32 || D == 64);
   auto dqkv = torch::empty_like(qkv);
   const int Npad = (N + 15) & ~15;
-  dim3 grid(B * H), block(256);
+  const int nwaves = N > 96 ? 8 : 4;
+  dim3 grid(B * H), block(nwaves * 64);
   const int lds_kv = (2 * Npad * dla::KPAD + 2 * D * (Npad + 8) +
-                      4 * 16 * (Npad + 8)) * 2;
+                      nwaves * dla::PBUF) * 2;
   const int lds_q = (2 * Npad * dla::KPAD + D * (Npad + 8) +
-                     4 * 16 * (Npad + 8)) * 2;
+                     nwaves * dla::PBUF) * 2;
   auto run = [&](auto dtag) {
     constexpr int DD = decltype(dtag)::value;
     hipLaunchKernelGGL((dla::attn_bwd_kv_kernel<DD>), grid, block, lds_kv,
@@ -799,4 +712,16 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor qkv, torch::Tensor dout,
   else run(std::integral_constant<int, 32>{});
   HIP_CHECK_ERR();
   return {dqkv};
+}
+
+torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B) {
+  TORCH_CHECK(A.scalar_type() == at::kBFloat16 &&
+              B.scalar_type() == at::kBFloat16);
+  auto D = torch::zeros({16, 16}, A.options().dtype(torch::kFloat));
+  hipLaunchKernelGGL(dla::mfma_probe_kernel, dim3(1), dim3(64), 0,
+                     dla::stream(), (const __hip_bfloat16*)A.data_ptr(),
+                     (const __hip_bfloat16*)B.data_ptr(),
+                     D.data_ptr<float>());
+  HIP_CHECK_ERR();
+  return {D};
 }
