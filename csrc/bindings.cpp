@@ -61,6 +61,13 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor qkv, torch::Tensor dout,
                                     torch::Tensor stats, torch::Tensor drow,
                                     int64_t num_heads, double scale);
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B);
+// cocoeval.cpp (CPU)
+std::vector<torch::Tensor> cocoeval_match_image(torch::Tensor det_boxes,
+                                                torch::Tensor det_scores,
+                                                torch::Tensor gt_boxes,
+                                                torch::Tensor gt_crowd,
+                                                torch::Tensor iou_thrs,
+                                                int64_t max_dets);
 // window.hip
 torch::Tensor window_partition_fwd(torch::Tensor x, int64_t ws, int64_t shift);
 torch::Tensor window_partition_bwd(torch::Tensor grad, int64_t B, int64_t H,
@@ -96,6 +103,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("save_stats") = false);
   m.def("attn_bwd", &attn_bwd);
   m.def("mfma_probe", &mfma_probe);
+  m.def("cocoeval_match_image", &cocoeval_match_image);
   m.def("window_partition_fwd", &window_partition_fwd);
   m.def("window_partition_bwd", &window_partition_bwd);
   m.def("window_merge_fwd", &window_merge_fwd);

@@ -1,0 +1,100 @@
+// Fast COCO-style detection matching (CPU, C++).
+//
+// Reference parity: detection/YOLOX/yolox/layers/csrc/cocoeval/cocoeval.{h,cpp}
+// (EvaluateImages / Accumulate, ~10x pycocotools) — re-designed to back this
+// repo's DetEvaluator (deeplearning_amd/engine/det_eval.py): the greedy
+// per-IoU-threshold matcher below is the hot loop; PR accumulation stays in
+// Python (vectorized torch). Python match_image is the reference
+// implementation the parity test compares against.
+#include <torch/extension.h>
+
+#include <algorithm>
+#include <cmath>
+#include <vector>
+
+namespace {
+
+inline float iou_xyxy(const float* a, const float* b) {
+  const float ix1 = std::max(a[0], b[0]);
+  const float iy1 = std::max(a[1], b[1]);
+  const float ix2 = std::min(a[2], b[2]);
+  const float iy2 = std::min(a[3], b[3]);
+  const float iw = std::max(0.f, ix2 - ix1);
+  const float ih = std::max(0.f, iy2 - iy1);
+  const float inter = iw * ih;
+  const float area_a = std::max(0.f, a[2] - a[0]) * std::max(0.f, a[3] - a[1]);
+  const float area_b = std::max(0.f, b[2] - b[0]) * std::max(0.f, b[3] - b[1]);
+  const float uni = area_a + area_b - inter;
+  return uni > 0.f ? inter / uni : 0.f;
+}
+
+}  // namespace
+
+// Greedy COCO matching for one image+class. Inputs are float32 CPU tensors:
+// det_boxes [D,4] xyxy, det_scores [D], gt_boxes [G,4], gt_crowd [G] (bool).
+// Returns (matched [T,D] bool, ignored [T,D] bool, sorted_scores [Dk], n_gt)
+// with detections sorted score-descending and truncated to max_dets —
+// identical protocol to DetEvaluator.match_image.
+std::vector<torch::Tensor> cocoeval_match_image(torch::Tensor det_boxes,
+                                                torch::Tensor det_scores,
+                                                torch::Tensor gt_boxes,
+                                                torch::Tensor gt_crowd,
+                                                torch::Tensor iou_thrs,
+                                                int64_t max_dets) {
+  TORCH_CHECK(det_boxes.device().is_cpu(), "cocoeval runs on CPU tensors");
+  auto order = det_scores.argsort(0, /*descending=*/true);
+  if (order.numel() > max_dets) order = order.slice(0, 0, max_dets);
+  auto boxes = det_boxes.index_select(0, order).contiguous();
+  auto scores = det_scores.index_select(0, order).contiguous();
+  auto gt = gt_boxes.contiguous();
+  auto crowd = gt_crowd.to(torch::kBool).contiguous();
+  auto thrs = iou_thrs.to(torch::kFloat).contiguous();
+
+  const int64_t D = boxes.size(0);
+  const int64_t G = gt.size(0);
+  const int64_t T = thrs.size(0);
+  auto matched = torch::zeros({T, D}, torch::kBool);
+  auto ignored = torch::zeros({T, D}, torch::kBool);
+  int64_t n_gt = G > 0 ? (G - crowd.sum().item<int64_t>()) : 0;
+  if (D == 0 || G == 0)
+    return {matched, ignored, scores, torch::tensor(n_gt)};
+
+  // IoU matrix once
+  std::vector<float> ious((size_t)D * G);
+  const float* bp = boxes.data_ptr<float>();
+  const float* gp = gt.data_ptr<float>();
+  for (int64_t d = 0; d < D; ++d)
+    for (int64_t g = 0; g < G; ++g)
+      ious[d * G + g] = iou_xyxy(bp + d * 4, gp + g * 4);
+
+  const bool* cp = crowd.data_ptr<bool>();
+  const float* tp = thrs.data_ptr<float>();
+  auto m_acc = matched.accessor<bool, 2>();
+  auto i_acc = ignored.accessor<bool, 2>();
+  std::vector<char> taken(G);
+  for (int64_t t = 0; t < T; ++t) {
+    std::fill(taken.begin(), taken.end(), 0);
+    for (int64_t d = 0; d < D; ++d) {
+      float best_iou = tp[t];
+      int64_t best_g = -1;
+      for (int64_t g = 0; g < G; ++g) {
+        if (taken[g] && !cp[g]) continue;
+        // prefer non-crowd matches; crowd only if nothing else found
+        if (best_g >= 0 && !cp[best_g] && cp[g]) continue;
+        if (ious[d * G + g] >= best_iou) {
+          best_iou = ious[d * G + g];
+          best_g = g;
+        }
+      }
+      if (best_g >= 0) {
+        if (cp[best_g]) {
+          i_acc[t][d] = true;
+        } else {
+          m_acc[t][d] = true;
+          taken[best_g] = 1;
+        }
+      }
+    }
+  }
+  return {matched, ignored, scores, torch::tensor(n_gt)};
+}

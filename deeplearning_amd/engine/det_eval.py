@@ -16,8 +16,19 @@ import torch
 from ..core.dist import all_gather_object_list as all_gather_object
 from ..core.dist import get_world_size
 from ..ops import box_iou
+from ..ops._ext import has_ext
 
 COCO_IOU_THRS = [0.5 + 0.05 * i for i in range(10)]
+
+
+def match_image_native(det_boxes, det_scores, gt_boxes, gt_crowd, iou_thrs,
+                       max_dets=100):
+    """C++ fast path (csrc/cocoeval.cpp, ref YOLOX native cocoeval)."""
+    from ..ops._ext import ext
+    matched, ignored, scores, n_gt = ext().cocoeval_match_image(
+        det_boxes.float(), det_scores.float(), gt_boxes.float(),
+        gt_crowd.to(torch.bool), torch.tensor(iou_thrs), max_dets)
+    return matched, ignored, scores, int(n_gt)
 
 
 def match_image(det_boxes, det_scores, gt_boxes, gt_crowd, iou_thrs,
@@ -89,10 +100,11 @@ class DetEvaluator:
             classes = torch.cat([pred["labels"], gt["labels"]]).unique()
             crowd = gt.get("iscrowd",
                            torch.zeros_like(gt["labels"]))
+            match = match_image_native if has_ext() else match_image
             for c in classes.tolist():
                 dm = pred["labels"] == c
                 gm = gt["labels"] == c
-                matched, ignored, scores, n_gt = match_image(
+                matched, ignored, scores, n_gt = match(
                     pred["boxes"][dm].cpu(), pred["scores"][dm].cpu(),
                     gt["boxes"][gm].cpu(), crowd[gm].cpu().bool(),
                     self.iou_thrs, self.max_dets)

@@ -155,3 +155,26 @@ def test_unet_project_cli(tmp_path):
              "--output", str(tmp_path))
     assert r.returncode == 0, r.stderr[-2000:]
     assert (tmp_path / "U-Net" / "weights" / "model_0.pth").exists()
+
+
+def test_cocoeval_native_matches_python():
+    """C++ fast matcher (csrc/cocoeval.cpp) vs the Python reference."""
+    from deeplearning_amd.engine.det_eval import (COCO_IOU_THRS, match_image,
+                                                  match_image_native)
+    from deeplearning_amd.ops import has_ext
+    if not has_ext():
+        pytest.skip("extension not built")
+    torch.manual_seed(1)
+    for _ in range(8):
+        D = int(torch.randint(0, 30, (1,)))
+        G = int(torch.randint(0, 8, (1,)))
+        db = torch.rand(D, 4) * 100
+        db[:, 2:] += db[:, :2] + 2
+        ds = torch.rand(D)
+        gb = torch.rand(G, 4) * 100
+        gb[:, 2:] += gb[:, :2] + 2
+        gc = torch.rand(G) > 0.8
+        m1, i1, s1, n1 = match_image(db, ds, gb, gc, COCO_IOU_THRS)
+        m2, i2, s2, n2 = match_image_native(db, ds, gb, gc, COCO_IOU_THRS)
+        assert torch.equal(m1, m2) and torch.equal(i1, i2)
+        assert torch.allclose(s1, s2) and n1 == n2
