@@ -1,3 +1,4 @@
 from . import (coatnet, convnext, efficientnet, googlenet, lenet,  # noqa: F401
-               repvgg, resnest, resnet, senet, shufflenet, swin, transfg, vgg,
+               repvgg, resnest, resnet, senet, shufflenet, swin, swin_moe,
+               transfg, vgg,
                vit)
