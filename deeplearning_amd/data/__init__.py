@@ -1,6 +1,6 @@
 from .datasets import (COCODetectionDataset, ClassificationDataset,  # noqa: F401
-                       SegmentationDataset, VOCDetectionDataset, mosaic4,
-                       read_split_data)
+                       MosaicDetection, SegmentationDataset,
+                       VOCDetectionDataset, mosaic4, read_split_data)
 from .prefetcher import DataPrefetcher  # noqa: F401
 from .samplers import (GroupedBatchSampler, InfiniteSampler,  # noqa: F401
                        SubsetRandomSampler, YoloBatchSampler,

@@ -234,3 +234,33 @@ def mosaic4(images, targets, out_size=640):
     labels = torch.cat(all_labels) if all_labels else \
         torch.zeros(0, dtype=torch.int64)
     return canvas, {"boxes": boxes, "labels": labels}
+
+
+class MosaicDetection(Dataset):
+    """Wrap a detection dataset with 4-image mosaic augmentation
+    (ref detection/YOLOX/yolox/data/datasets/mosaicdetection.py). Each item
+    draws 3 extra random indices and composes mosaic4; `enabled=False` (or a
+    YoloBatchSampler (mosaic, idx) tuple with mosaic=False) passes through —
+    the trainer flips it off for the last no-aug epochs."""
+
+    def __init__(self, dataset, out_size=640, enabled=True):
+        self.dataset = dataset
+        self.out_size = out_size
+        self.enabled = enabled
+
+    def __len__(self):
+        return len(self.dataset)
+
+    def __getitem__(self, index):
+        enabled = self.enabled
+        if isinstance(index, tuple):  # YoloBatchSampler item
+            enabled, index = index
+        if not enabled:
+            return self.dataset[index]
+        idxs = [index] + [random.randrange(len(self.dataset))
+                          for _ in range(3)]
+        items = [self.dataset[i] for i in idxs]
+        return mosaic4([im for im, _ in items], [t for _, t in items],
+                       self.out_size)
+
+    collate_fn = staticmethod(VOCDetectionDataset.collate_fn)

@@ -1,4 +1,4 @@
 from . import (coatnet, convnext, efficientnet, googlenet, lenet,  # noqa: F401
                repvgg, resnest, resnet, senet, shufflenet, swin, swin_moe,
-               transfg, vgg,
+               transfg, vgg, zoo_extra,
                vit)

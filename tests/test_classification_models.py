@@ -82,3 +82,14 @@ def test_transfg_contrastive_loss():
     loss = contrastive_loss(f, labels)
     loss.backward()
     assert loss.item() >= 0 and f.grad is not None
+
+
+@pytest.mark.parametrize("name,size", [("dpn68", 224), ("inception_v4", 299),
+                                       ("swin_moe_t", 224)])
+def test_zoo_extra_forward_backward(name, size):
+    torch.manual_seed(0)
+    m = build_model(name, num_classes=10)
+    m.train()
+    out = m(torch.randn(2, 3, size, size))
+    out.sum().backward()
+    assert out.shape == (2, 10)
