@@ -70,6 +70,13 @@ def classification_argparser(default_model: str, **defaults):
     p.add_argument("--clip-grad", type=float, default=0.0)
     p.add_argument("--synthetic-size", type=int, default=256,
                    help="synthetic dataset length when --data-path is empty")
+    p.add_argument("--throughput", action="store_true",
+                   help="50 warmup + 30 timed forward passes, then exit "
+                        "(ref swin main.py:280-297)")
+    p.add_argument("--cfg", default="", help="YAML config (CfgNode, supports "
+                   "_BASE_ inheritance); keys override CLI defaults")
+    p.add_argument("--opts", nargs="*", default=None,
+                   help="dotted-key config overrides, e.g. train.lr 0.1")
     return p
 
 
@@ -100,7 +107,24 @@ def build_classification_loaders(args):
     return train_loader, val_loader, train_sampler
 
 
+def apply_cfg(args):
+    """Merge --cfg YAML + --opts into args (one config system for all
+    subprojects; ref swin yacs config.py semantics)."""
+    if not getattr(args, "cfg", "") and not getattr(args, "opts", None):
+        return args
+    from ..core.config import load_config
+    defaults = {k.replace("-", "_"): v for k, v in vars(args).items()}
+    cfg = load_config(defaults, args.cfg or None, args.opts, freeze=False)
+    for k, v in cfg.items():
+        if k not in ("cfg", "opts"):
+            setattr(args, k, v)
+    return args
+
+
 def classification_train_main(args) -> dict:
+    args = apply_cfg(args)
+    if getattr(args, "throughput", False):
+        return throughput_main(args)
     info = init_distributed()
     if torch.cuda.is_available() and get_world_size() > 1:
         device = torch.device("cuda", info["local_rank"])
@@ -254,3 +278,32 @@ def predict_main(default_model: str, num_classes: int = 1000,
     for score, idx in zip(topk.values.tolist(), topk.indices.tolist()):
         print(f"class {idx}: {score:.4f}")
     return topk
+
+
+@torch.no_grad()
+def throughput_main(args) -> dict:
+    """Throughput mode: 50 warmup + 30 timed forward passes
+    (ref swin main.py:280-297)."""
+    device = select_device(args.device)
+    model = build_model(args.model, num_classes=args.num_classes).to(device)
+    model.eval()
+    c = getattr(args, "in_channels", 3)
+    x = torch.randn(args.batch_size, c, args.img_size, args.img_size,
+                    device=device)
+    amp = args.amp and device.type == "cuda"
+    warmup, iters = (50, 30) if device.type == "cuda" else (3, 5)
+    with torch.autocast(device.type, dtype=torch.bfloat16, enabled=amp):
+        for _ in range(warmup):
+            model(x)
+        if device.type == "cuda":
+            torch.cuda.synchronize()
+        t0 = time.time()
+        for _ in range(iters):
+            model(x)
+        if device.type == "cuda":
+            torch.cuda.synchronize()
+    elapsed = time.time() - t0
+    ips = args.batch_size * iters / elapsed
+    print(f"throughput: {ips:.1f} images/s "
+          f"({elapsed / iters * 1000:.2f} ms/batch of {args.batch_size})")
+    return {"images_per_sec": ips}
