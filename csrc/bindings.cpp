@@ -61,6 +61,8 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor qkv, torch::Tensor dout,
                                     torch::Tensor stats, torch::Tensor drow,
                                     int64_t num_heads, double scale);
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B);
+// my_add.cpp (custom-op export tutorial)
+torch::Tensor my_add(torch::Tensor a, torch::Tensor b);
 // cocoeval.cpp (CPU)
 std::vector<torch::Tensor> cocoeval_match_image(torch::Tensor det_boxes,
                                                 torch::Tensor det_scores,
@@ -104,6 +106,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_bwd", &attn_bwd);
   m.def("mfma_probe", &mfma_probe);
   m.def("cocoeval_match_image", &cocoeval_match_image);
+  m.def("my_add", &my_add);
   m.def("window_partition_fwd", &window_partition_fwd);
   m.def("window_partition_bwd", &window_partition_bwd);
   m.def("window_merge_fwd", &window_merge_fwd);

@@ -16,7 +16,7 @@ from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E40
 ROOT = Path(__file__).parent
 CSRC = ROOT / "csrc"
 
-sources = [str(CSRC / "bindings.cpp"), str(CSRC / "cocoeval.cpp")] + sorted(
+sources = [str(CSRC / "bindings.cpp"), str(CSRC / "cocoeval.cpp"), str(CSRC / "my_add.cpp")] + sorted(
     str(p) for p in CSRC.glob("*.hip") if not p.name.endswith("_hip.hip")
 )
 
