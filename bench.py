@@ -39,6 +39,9 @@ def parse_args():
     p.add_argument("--bucket-mb", type=float, default=64.0)
     p.add_argument("--no-channels-last", action="store_true")
     p.add_argument("--lr", type=float, default=0.1)
+    p.add_argument("--graph", default="auto", choices=["auto", "on", "off"],
+                   help="capture the train step in a hipGraph (single-GPU "
+                        "only; removes per-kernel launch overhead)")
     return p.parse_args()
 
 
@@ -78,7 +81,10 @@ def main():
     amp_enabled = has_gpu
     finalize = getattr(model, "finalize", None)
 
-    def one_step(i):
+    use_graph = args.graph == "on" or (
+        args.graph == "auto" and world == 1 and has_gpu)
+
+    def one_step(i, set_to_none=True):
         x, y = batches[i % len(batches)]
         with torch.autocast("cuda", dtype=amp_dtype, enabled=amp_enabled):
             logits = model(x)
@@ -87,19 +93,49 @@ def main():
         if finalize is not None:
             finalize()
         optimizer.step()
-        optimizer.zero_grad(set_to_none=True)
+        optimizer.zero_grad(set_to_none=set_to_none)
         return loss
 
-    # warmup
+    # warmup (also absorbs MIOpen find); grads stay allocated for capture
     for i in range(args.warmup):
-        one_step(i)
+        one_step(i, set_to_none=not use_graph)
+
+    graph = None
+    if use_graph:
+        # capture one full train step (fwd + loss + bwd + SGD) as a hipGraph;
+        # replay re-executes every kernel on the same static batch buffers
+        try:
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                for _ in range(2):
+                    one_step(0, set_to_none=False)
+            torch.cuda.current_stream().wait_stream(side)
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
+                one_step(0, set_to_none=False)
+        except Exception as e:  # capture unsupported for this model: run eager
+            import sys
+
+            print(f"[bench] graph capture failed ({e}); eager path",
+                  file=sys.stderr)
+            graph = None
 
     barrier()
     if has_gpu:
         torch.cuda.synchronize()
     t0 = time.perf_counter()
+    static_x, static_y = batches[0]
     for i in range(args.steps):
-        one_step(args.warmup + i)
+        if graph is not None:
+            # rotate fresh data into the captured static buffers
+            src_x, src_y = batches[(args.warmup + i) % len(batches)]
+            if src_x.data_ptr() != static_x.data_ptr():
+                static_x.copy_(src_x)
+                static_y.copy_(src_y)
+            graph.replay()
+        else:
+            one_step(args.warmup + i)
     barrier()
     if has_gpu:
         torch.cuda.synchronize()
