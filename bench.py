@@ -39,7 +39,7 @@ def parse_args():
     p.add_argument("--bucket-mb", type=float, default=64.0)
     p.add_argument("--no-channels-last", action="store_true")
     p.add_argument("--lr", type=float, default=0.1)
-    p.add_argument("--graph", default="auto", choices=["auto", "on", "off"],
+    p.add_argument("--graph", default="off", choices=["auto", "on", "off"],
                    help="capture the train step in a hipGraph (single-GPU "
                         "only; removes per-kernel launch overhead)")
     return p.parse_args()
