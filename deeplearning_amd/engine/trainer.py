@@ -117,7 +117,7 @@ class Trainer:
                  criterion=None, lr_scheduler=None, val_loader=None, ema=None,
                  accum_steps=1, clip_grad=0.0, amp=True,
                  amp_dtype=torch.bfloat16, output_dir=None, logger=None,
-                 eval_interval=1, save_interval=1):
+                 eval_interval=1, save_interval=1, callbacks=None):
         self.model = model
         self.optimizer = optimizer
         self.train_loader = train_loader
@@ -137,6 +137,8 @@ class Trainer:
         self.save_interval = save_interval
         self.epoch = 0
         self.best_metric = 0.0
+        from .callbacks import Callbacks
+        self.callbacks = callbacks or Callbacks()
 
     # hooks ------------------------------------------------------------
     def before_train(self):
@@ -176,16 +178,20 @@ class Trainer:
 
     # loop -------------------------------------------------------------
     def train(self):
+        self.callbacks.run("on_train_start", self)
         self.before_train()
         try:
             for self.epoch in range(self.epoch, self.max_epoch):
+                self.callbacks.run("on_train_epoch_start", self)
                 self.before_epoch()
                 train_one_epoch(self.model, self.criterion, self.train_loader,
                                 self.optimizer, self.device, self.epoch,
                                 self.lr_scheduler, self.accum_steps,
                                 self.clip_grad, self.amp, self.amp_dtype,
                                 self.ema, logger=self.logger)
-                self.after_epoch()
+                stats = self.after_epoch()
+                self.callbacks.run("on_fit_epoch_end", self, stats)
         finally:
             self.after_train()
+            self.callbacks.run("on_train_end", self)
         return self.best_metric

@@ -5,7 +5,6 @@ from pathlib import Path
 
 sys.path.insert(0, str(Path(__file__).resolve().parents[3]))
 
-import torch
 
 from deeplearning_amd.core.checkpoint import (load_pretrained, save_weights,
                                               strip_module_prefix)

@@ -1,5 +1,4 @@
 """CPU tests for deeplearning_amd.core: config, meters, checkpoint, env."""
-import os
 
 import pytest
 import torch

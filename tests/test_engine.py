@@ -84,3 +84,22 @@ def test_trainer_class(tmp_path):
 
     assert any(f.startswith("ckpt_epoch_") or f == "best.pth"
                for f in os.listdir(tmp_path))
+
+
+def test_callbacks_registry():
+    from deeplearning_amd.engine.callbacks import Callbacks
+    cb = Callbacks()
+    seen = []
+    cb.register_action("on_train_start", "rec", lambda t: seen.append("s"))
+
+    @cb.on("on_train_end")
+    def done(t):
+        seen.append("e")
+
+    cb.run("on_train_start", None)
+    cb.run("on_train_end", None)
+    assert seen == ["s", "e"]
+    assert len(cb.get_registered_actions("on_train_end")) == 1
+    import pytest as _pytest
+    with _pytest.raises(AssertionError):
+        cb.register_action("nope", callback=lambda: None)
