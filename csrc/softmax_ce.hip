@@ -87,6 +87,125 @@ __global__ void ce_bwd_kernel(const dev_t* __restrict__ logits,
   }
 }
 
+// ---- wave-per-row variants (C % V == 0, C <= 64*V*MAX_PL) ------------------
+// One 64-lane wavefront per row: shuffle-only reductions, 16B vector loads,
+// no __syncthreads (the block-per-row kernels above stay as the fallback).
+template <typename dev_t, int V>
+__global__ __launch_bounds__(256)
+void ce_fwd_wave_kernel(const dev_t* __restrict__ logits,
+                        const int64_t* __restrict__ target,
+                        const dev_t* __restrict__ soft_target,
+                        float* __restrict__ loss, float* __restrict__ lse_out,
+                        int B, int C, float smoothing, int64_t ignore_index) {
+  constexpr int MAX_PL = 8;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int nwaves = blockDim.x >> 6;
+  Vec<dev_t, V> zv[MAX_PL];
+  for (int row = blockIdx.x * nwaves + wave; row < B;
+       row += gridDim.x * nwaves) {
+    const dev_t* lr = logits + (int64_t)row * C;
+    float mx = -INFINITY;
+#pragma unroll
+    for (int p = 0; p < MAX_PL; ++p) {
+      const int i = lane * V + p * 64 * V;
+      if (i < C) {
+        zv[p] = vload<dev_t, V>(lr + i);
+#pragma unroll
+        for (int j = 0; j < V; ++j) mx = fmaxf(mx, to_f32(zv[p].v[j]));
+      }
+    }
+    mx = wave_reduce_max(mx);
+    float se = 0.f, ssum = 0.f;
+#pragma unroll
+    for (int p = 0; p < MAX_PL; ++p) {
+      const int i = lane * V + p * 64 * V;
+      if (i < C) {
+        Vec<dev_t, V> qv;
+        if (soft_target != nullptr)
+          qv = vload<dev_t, V>(soft_target + (int64_t)row * C + i);
+#pragma unroll
+        for (int j = 0; j < V; ++j) {
+          const float z = to_f32(zv[p].v[j]);
+          se += __expf(z - mx);
+          if (soft_target != nullptr) ssum += to_f32(qv.v[j]) * z;
+          else if (smoothing > 0.f) ssum += z;
+        }
+      }
+    }
+    se = wave_reduce_sum(se);
+    const float lse = mx + __logf(se);
+    if (soft_target != nullptr) {
+      ssum = wave_reduce_sum(ssum);
+      if (lane == 0) {
+        loss[row] = lse - ssum;
+        lse_out[row] = lse;
+      }
+    } else {
+      if (smoothing > 0.f) ssum = wave_reduce_sum(ssum);
+      if (lane == 0) {
+        const int64_t t = target[row];
+        if (t == ignore_index) {
+          loss[row] = 0.f;
+        } else {
+          const float zt = to_f32(lr[t]);
+          loss[row] = (1.f - smoothing) * (lse - zt) +
+                      smoothing * (lse - ssum / C);
+        }
+        lse_out[row] = lse;
+      }
+    }
+  }
+}
+
+template <typename dev_t, int V>
+__global__ __launch_bounds__(256)
+void ce_bwd_wave_kernel(const dev_t* __restrict__ logits,
+                        const int64_t* __restrict__ target,
+                        const dev_t* __restrict__ soft_target,
+                        const float* __restrict__ lse,
+                        dev_t* __restrict__ dlogits, int B, int C,
+                        float smoothing, int64_t ignore_index,
+                        float grad_scale) {
+  constexpr int MAX_PL = 8;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int nwaves = blockDim.x >> 6;
+  for (int row = blockIdx.x * nwaves + wave; row < B;
+       row += gridDim.x * nwaves) {
+    const dev_t* lr = logits + (int64_t)row * C;
+    dev_t* dr = dlogits + (int64_t)row * C;
+    const float l = lse[row];
+    const int64_t t = soft_target == nullptr ? target[row] : -1;
+    const bool ignored = (soft_target == nullptr) && (t == ignore_index);
+#pragma unroll
+    for (int p = 0; p < MAX_PL; ++p) {
+      const int i = lane * V + p * 64 * V;
+      if (i < C) {
+        Vec<dev_t, V> out;
+        if (ignored) {
+#pragma unroll
+          for (int j = 0; j < V; ++j) out.v[j] = from_f32<dev_t>(0.f);
+        } else {
+          Vec<dev_t, V> zv = vload<dev_t, V>(lr + i);
+          Vec<dev_t, V> qv;
+          if (soft_target != nullptr)
+            qv = vload<dev_t, V>(soft_target + (int64_t)row * C + i);
+#pragma unroll
+          for (int j = 0; j < V; ++j) {
+            const float prob = __expf(to_f32(zv.v[j]) - l);
+            const float q = soft_target != nullptr
+                ? to_f32(qv.v[j])
+                : ((i + j == (int)t ? 1.f - smoothing : 0.f) + smoothing / C);
+            out.v[j] = from_f32<dev_t>((prob - q) * grad_scale);
+          }
+        }
+        vstore<dev_t, V>(dr + i, out);
+      }
+    }
+  }
+}
+
 }  // namespace dla
 
 // Returns {loss[B] fp32, lse[B] fp32}
@@ -100,6 +219,19 @@ std::vector<torch::Tensor> softmax_ce_fwd(torch::Tensor logits,
   auto lse = torch::empty({B}, logits.options().dtype(torch::kFloat));
   const int grid = (int)std::min<int64_t>(B, dla::kMaxGrid);
   DLA_DISPATCH_FLOAT_TYPES(logits.scalar_type(), "softmax_ce_fwd", [&] {
+    constexpr int VMAX = 16 / (int)sizeof(dev_t);
+    if (C % VMAX == 0 && C <= 64 * VMAX * 8) {
+      const int g = (int)std::min<int64_t>((B + 3) / 4, dla::kMaxGrid);
+      hipLaunchKernelGGL(
+          (dla::ce_fwd_wave_kernel<dev_t, VMAX>), dim3(g), dim3(256), 0,
+          dla::stream(), (const dev_t*)logits.data_ptr(),
+          target.has_value() ? target->data_ptr<int64_t>() : nullptr,
+          soft_target.has_value() ? (const dev_t*)soft_target->data_ptr()
+                                  : nullptr,
+          loss.data_ptr<float>(), lse.data_ptr<float>(), B, C,
+          (float)smoothing, ignore_index);
+      return;
+    }
     hipLaunchKernelGGL(
         (dla::ce_fwd_kernel<dev_t>), dim3(grid), dim3(256), 0, dla::stream(),
         (const dev_t*)logits.data_ptr(),
@@ -122,6 +254,19 @@ torch::Tensor softmax_ce_bwd(torch::Tensor logits,
   auto dlogits = torch::empty_like(logits);
   const int grid = (int)std::min<int64_t>(B, dla::kMaxGrid);
   DLA_DISPATCH_FLOAT_TYPES(logits.scalar_type(), "softmax_ce_bwd", [&] {
+    constexpr int VMAX = 16 / (int)sizeof(dev_t);
+    if (C % VMAX == 0 && C <= 64 * VMAX * 8) {
+      const int g = (int)std::min<int64_t>((B + 3) / 4, dla::kMaxGrid);
+      hipLaunchKernelGGL(
+          (dla::ce_bwd_wave_kernel<dev_t, VMAX>), dim3(g), dim3(256), 0,
+          dla::stream(), (const dev_t*)logits.data_ptr(),
+          target.has_value() ? target->data_ptr<int64_t>() : nullptr,
+          soft_target.has_value() ? (const dev_t*)soft_target->data_ptr()
+                                  : nullptr,
+          lse.data_ptr<float>(), (dev_t*)dlogits.data_ptr(), B, C,
+          (float)smoothing, ignore_index, (float)grad_scale);
+      return;
+    }
     hipLaunchKernelGGL(
         (dla::ce_bwd_kernel<dev_t>), dim3(grid), dim3(256), 0, dla::stream(),
         (const dev_t*)logits.data_ptr(),
