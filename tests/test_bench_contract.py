@@ -55,3 +55,25 @@ def test_bench_torchrun_world2_cpu():
     assert res["config"]["parallelism"] == "dp2"
     # whole-job value: global batch = 2x per-rank batch
     assert res["config"]["global_batch"] == 16  # 8 per rank on CPU
+
+
+@pytest.mark.slow
+@pytest.mark.timeout(600)
+def test_classification_cli_torchrun_world2(tmp_path):
+    """Full classification training CLI under torchrun world=2 (gloo):
+    DDP wrap, DistributedSampler, rank-0 checkpointing."""
+    env = dict(os.environ)
+    env.pop("RANK", None)
+    env.pop("WORLD_SIZE", None)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29533",
+         "projects/classification/mnist/train.py", "--epochs", "1",
+         "--batch-size", "8", "--synthetic-size", "32", "--workers", "0",
+         "--device", "cpu", "--output", str(tmp_path)],
+        capture_output=True, text=True, timeout=560, cwd=REPO, env=env)
+    assert r.returncode == 0, (r.stdout[-1000:], r.stderr[-2000:])
+    weights = tmp_path / "mnist" / "weights"
+    assert (weights / "model_0.pth").exists()
+    assert (weights / "best_model.pth").exists()
