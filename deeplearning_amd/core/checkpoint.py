@@ -68,6 +68,7 @@ def load_pretrained(model, path: str | Path, map_location: str = "cpu",
     mnist/train.py:110-117, train_with_DDP:167-169). Returns dropped keys."""
     ckpt = torch.load(path, map_location=map_location, weights_only=False)
     state = strip_module_prefix(ckpt.get("model", ckpt))
+    state = interpolate_rel_pos_tables(state, model)
     own = unwrap_model(model).state_dict()
     dropped = []
     filtered = {}
@@ -96,3 +97,31 @@ def auto_resume_helper(output_dir: str | Path, prefix: str = "ckpt_epoch_") -> s
         return int(m.group(1)) if m else -1
 
     return max(files, key=epoch_of)
+
+
+def interpolate_rel_pos_tables(state_dict: dict, model) -> dict:
+    """Shape-aware load: bicubic-resize Swin relative_position_bias_table
+    entries whose window size differs, and drop stale relative_position_index
+    buffers (reference swin load_pretrained, utils/torch_utils.py:143-231)."""
+    import torch.nn.functional as F
+
+    own = unwrap_model(model).state_dict()
+    out = dict(state_dict)
+    for k in list(out.keys()):
+        if k.endswith("relative_position_index"):
+            out.pop(k)  # buffer, recomputed at construction
+            continue
+        if not k.endswith("relative_position_bias_table") or k not in own:
+            continue
+        src, dst = out[k], own[k]
+        if src.shape == dst.shape:
+            continue
+        L1, heads = src.shape
+        L2 = dst.shape[0]
+        s1 = int(L1 ** 0.5)
+        s2 = int(L2 ** 0.5)
+        table = src.permute(1, 0).reshape(1, heads, s1, s1)
+        table = F.interpolate(table, size=(s2, s2), mode="bicubic",
+                              align_corners=False)
+        out[k] = table.reshape(heads, L2).permute(1, 0)
+    return out
