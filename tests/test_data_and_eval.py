@@ -178,3 +178,21 @@ def test_cocoeval_native_matches_python():
         m2, i2, s2, n2 = match_image_native(db, ds, gb, gc, COCO_IOU_THRS)
         assert torch.equal(m1, m2) and torch.equal(i1, i2)
         assert torch.allclose(s1, s2) and n1 == n2
+
+
+@pytest.mark.slow
+def test_predict_cli(tmp_path):
+    """predict.py surface: saved weights + image -> top-k prediction lines."""
+    import torch as _torch
+    from deeplearning_amd.core.checkpoint import save_weights
+    from deeplearning_amd.models import build_model
+    m = build_model("resnet18", num_classes=5)
+    save_weights(m, tmp_path / "w.pth")
+    Image.fromarray(np.random.randint(0, 255, (64, 64, 3),
+                                      dtype=np.uint8)).save(tmp_path / "x.jpg")
+    r = _run("projects/classification/resnet/predict.py",
+             str(tmp_path / "x.jpg"), "--model", "resnet18",
+             "--weights", str(tmp_path / "w.pth"), "--num-classes", "5",
+             "--img-size", "64", "--device", "cpu")
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert "class" in r.stdout
