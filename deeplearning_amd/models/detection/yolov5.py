@@ -360,3 +360,21 @@ def yolov5l(num_classes=80, **kw):
 @register_model
 def yolov5x(num_classes=80, **kw):
     return YoloV5(nc=num_classes, gd=1.33, gw=1.25)
+
+
+def yolov5_from_yaml(cfg_path, nc=80, gd=None, gw=None):
+    """Build a YoloV5 from a yaml model file (ref models/yolo.py Model(cfg)).
+
+    The yaml uses the reference schema: nc, depth_multiple, width_multiple,
+    anchors, backbone, head.
+    """
+    import yaml
+
+    with open(cfg_path) as f:
+        d = yaml.safe_load(f)
+    cfg = {"anchors": d["anchors"], "backbone": d["backbone"],
+           "head": d["head"]}
+    return YoloV5(nc=d.get("nc", nc) if nc == 80 else nc,
+                  gd=gd if gd is not None else d.get("depth_multiple", 0.33),
+                  gw=gw if gw is not None else d.get("width_multiple", 0.5),
+                  cfg=cfg)
