@@ -118,8 +118,9 @@ class MoEMlp(nn.Module):
                 toks = tokens[flat_tok[sel]]
                 h = F.gelu(toks @ self.w1[e] + self.b1[e])
                 y = h @ self.w2[e] + self.b2[e]
+                # gate comes from an fp32 softmax under autocast
                 out.index_add_(0, flat_tok[sel],
-                               y * flat_gate[sel, None])
+                               (y * flat_gate[sel, None]).to(out.dtype))
             return out.reshape(B, N, C), self.aux_loss
 
         # ---- expert parallel: pad per-expert to capacity, all-to-all -------
@@ -140,7 +141,8 @@ class MoEMlp(nn.Module):
         back = torch.empty_like(y)
         dist.all_to_all_single(back, y, group=self.ep_group)
         back = back.reshape(E, cap, C)
-        contrib = back[flat_expert, slot] * flat_gate[:, None]
+        contrib = (back[flat_expert, slot] *
+                   flat_gate[:, None]).to(out.dtype)
         out.index_add_(0, flat_tok, contrib)
         return out.reshape(B, N, C), self.aux_loss
 
