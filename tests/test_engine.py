@@ -103,3 +103,18 @@ def test_callbacks_registry():
     import pytest as _pytest
     with _pytest.raises(AssertionError):
         cb.register_action("nope", callback=lambda: None)
+
+
+def test_exp_config_as_code():
+    from deeplearning_amd.engine.exp import YoloxExp, get_exp
+    exp = get_exp(exp_name="yolox_m")
+    assert exp.depth == 0.67 and exp.model_name == "yolox_m"
+    exp.merge(["max_epoch", "10", "mosaic", "False"])
+    assert exp.max_epoch == 10 and exp.mosaic is False
+    m = exp.get_model()
+    opt = exp.get_optimizer(m, batch_size=8)
+    assert len(opt.param_groups) == 2
+    assert opt.param_groups[1]["weight_decay"] == 0.0
+    import pytest as _p
+    with _p.raises(AttributeError):
+        YoloxExp().merge(["nope", "1"])
