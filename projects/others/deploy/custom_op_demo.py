@@ -26,11 +26,26 @@ class MyAdd(torch.autograd.Function):
         return g.op("custom::MyAdd", a, b)
 
 
+@torch.jit.script
+def my_add_torchscript(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    """Way 3: TorchScript op (ref support_TorchScript_ops.py)."""
+    return 3 * a + 2 * b
+
+
+def my_add_aten(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    """Way 2: composed from ATen ops only (ref support_Aten_ops.py)."""
+    return torch.add(torch.mul(a, 3), torch.mul(b, 2))
+
+
 if __name__ == "__main__":
     a = torch.randn(4, requires_grad=True)
     b = torch.randn(4, requires_grad=True)
+    # way 1: native extension + autograd.Function + ONNX symbolic
     y = MyAdd.apply(a, b)
     assert torch.allclose(y, 3 * a + 2 * b)
     y.sum().backward()
     assert torch.allclose(a.grad, torch.full((4,), 3.0))
-    print("my_add custom op OK:", y.tolist())
+    # ways 2 & 3 agree
+    assert torch.allclose(my_add_aten(a, b), y)
+    assert torch.allclose(my_add_torchscript(a, b), y)
+    print("my_add custom op OK (native / aten / torchscript):", y.tolist())
