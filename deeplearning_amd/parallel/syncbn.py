@@ -16,6 +16,42 @@ from ..core.dist import get_world_size, is_dist
 
 
 def convert_sync_batchnorm(model: nn.Module) -> nn.Module:
+    """Like torch's convert_sync_batchnorm, but preserves this repo's fused
+    BatchNorm2d(relu=True): those become SyncBatchNorm + ReLU (torch's
+    converter would silently DROP the fused activation)."""
+    from ..ops import BatchNorm2d as FusedBN
+
+    def convert(module):
+        if isinstance(module, FusedBN):
+            sync = nn.SyncBatchNorm(module.num_features, module.eps,
+                                    module.momentum, module.affine,
+                                    module.track_running_stats)
+            with torch.no_grad():
+                if module.affine:
+                    sync.weight.copy_(module.weight)
+                    sync.bias.copy_(module.bias)
+                sync.running_mean.copy_(module.running_mean)
+                sync.running_var.copy_(module.running_var)
+                sync.num_batches_tracked.copy_(module.num_batches_tracked)
+            if module.relu:
+                return nn.Sequential(sync, nn.ReLU(inplace=True))
+            return sync
+        return None
+
+    def walk(parent):
+        for name, child in parent.named_children():
+            repl = convert(child)
+            if repl is not None:
+                setattr(parent, name, repl)
+            else:
+                walk(child)
+        return parent
+
+    top = convert(model)
+    if top is not None:
+        return top
+    walk(model)
+    # plain nn.BatchNorm2d (if any slipped in) via torch's converter
     return nn.SyncBatchNorm.convert_sync_batchnorm(model)
 
 

@@ -123,3 +123,23 @@ def test_no_sync_accumulation():
 def test_all_reduce_norm_single_proc():
     m = nn.Sequential(nn.Conv2d(3, 8, 3), nn.BatchNorm2d(8))
     all_reduce_norm(m)  # no-op without dist, must not raise
+
+
+def test_syncbn_conversion_preserves_fused_relu():
+    """Torch's converter silently drops BatchNorm2d(relu=True)'s activation;
+    ours must keep it as SyncBN + ReLU with copied stats."""
+    import torch.nn as nn
+
+    from deeplearning_amd.ops import BatchNorm2d
+    from deeplearning_amd.parallel.syncbn import convert_sync_batchnorm
+
+    net = nn.Sequential(nn.Conv2d(3, 4, 3), BatchNorm2d(4, relu=True),
+                        BatchNorm2d(4))
+    with torch.no_grad():
+        net[1].running_mean.fill_(0.5)
+    conv = convert_sync_batchnorm(net)
+    assert isinstance(conv[1], nn.Sequential)
+    assert isinstance(conv[1][0], nn.SyncBatchNorm)
+    assert isinstance(conv[1][1], nn.ReLU)
+    assert isinstance(conv[2], nn.SyncBatchNorm)
+    assert float(conv[1][0].running_mean[0]) == 0.5
