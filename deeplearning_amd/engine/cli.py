@@ -307,3 +307,48 @@ def throughput_main(args) -> dict:
     print(f"throughput: {ips:.1f} images/s "
           f"({elapsed / iters * 1000:.2f} ms/batch of {args.batch_size})")
     return {"images_per_sec": ips}
+
+
+def evaluate_main(default_model: str, num_classes: int = 1000,
+                  img_size: int = 224, in_channels: int = 3):
+    """Dataset evaluation CLI: load a checkpoint, run the val split, print
+    accuracy + confusion matrix (ref classification/*/test.py pattern)."""
+    p = argparse.ArgumentParser()
+    p.add_argument("--model", default=default_model)
+    p.add_argument("--weights", required=True)
+    p.add_argument("--data-path", default="", help="empty = synthetic")
+    p.add_argument("--num-classes", type=int, default=num_classes)
+    p.add_argument("--img-size", type=int, default=img_size)
+    p.add_argument("--in-channels", type=int, default=in_channels)
+    p.add_argument("--batch-size", type=int, default=32)
+    p.add_argument("--device", default="cuda")
+    p.add_argument("--workers", type=int, default=2)
+    p.add_argument("--synthetic-size", type=int, default=64)
+    args = p.parse_args()
+
+    device = select_device(args.device)
+    model = build_model(args.model, num_classes=args.num_classes).to(device)
+    load_pretrained(model, args.weights)
+    model.eval()
+    _, val_loader, _ = build_classification_loaders(args)
+
+    from .metrics import ConfusionMatrix
+    cm = ConfusionMatrix(args.num_classes)
+    correct = top5 = total = 0
+    with torch.no_grad():
+        for x, y in val_loader:
+            x, y = x.to(device), y.to(device)
+            with torch.autocast(device.type, dtype=torch.bfloat16,
+                                enabled=device.type == "cuda"):
+                logits = model(x).float()
+            pred = logits.argmax(1)
+            correct += int((pred == y).sum())
+            k = min(5, logits.shape[1])
+            top5 += int((logits.topk(k, dim=1).indices ==
+                         y[:, None]).any(1).sum())
+            total += y.numel()
+            cm.update(y.flatten(), pred.flatten())
+    print(f"top1 {100 * correct / max(total, 1):.2f}%  "
+          f"top5 {100 * top5 / max(total, 1):.2f}%  ({total} images)")
+    print(cm)
+    return correct / max(total, 1)
