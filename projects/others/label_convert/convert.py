@@ -73,16 +73,46 @@ def voc_to_coco(xml_files, class_names, out_json):
     return coco
 
 
+def yolo_to_voc_xml(txt_file, class_names, img_w, img_h, filename="img.jpg"):
+    """yolo txt -> VOC annotation XML string."""
+    from xml.dom import minidom
+    from xml.etree.ElementTree import Element, SubElement, tostring
+
+    root = Element("annotation")
+    SubElement(root, "filename").text = filename
+    size = SubElement(root, "size")
+    SubElement(size, "width").text = str(img_w)
+    SubElement(size, "height").text = str(img_h)
+    SubElement(size, "depth").text = "3"
+    with open(txt_file) as f:
+        lines = [l for l in f if l.strip()]
+    for cls, x1, y1, x2, y2 in yolo_to_voc_boxes(lines, img_w, img_h):
+        obj = SubElement(root, "object")
+        SubElement(obj, "name").text = class_names[cls]
+        SubElement(obj, "difficult").text = "0"
+        bb = SubElement(obj, "bndbox")
+        SubElement(bb, "xmin").text = str(int(round(x1)))
+        SubElement(bb, "ymin").text = str(int(round(y1)))
+        SubElement(bb, "xmax").text = str(int(round(x2)))
+        SubElement(bb, "ymax").text = str(int(round(y2)))
+    return minidom.parseString(tostring(root)).toprettyxml(indent="  ")
+
+
 if __name__ == "__main__":
     p = argparse.ArgumentParser()
-    p.add_argument("mode", choices=["voc2yolo", "voc2coco"])
+    p.add_argument("mode", choices=["voc2yolo", "voc2coco", "yolo2voc"])
     p.add_argument("inputs", nargs="+")
     p.add_argument("--classes", nargs="+", required=True)
     p.add_argument("--out", default="out.json")
+    p.add_argument("--img-size", type=int, nargs=2, default=[640, 640],
+                   help="width height (yolo2voc)")
     args = p.parse_args()
     if args.mode == "voc2yolo":
         for xf in args.inputs:
             print("\n".join(voc_to_yolo(xf, args.classes)))
+    elif args.mode == "yolo2voc":
+        for tf in args.inputs:
+            print(yolo_to_voc_xml(tf, args.classes, *args.img_size))
     else:
         voc_to_coco(args.inputs, args.classes, args.out)
         print(f"wrote {args.out}")
