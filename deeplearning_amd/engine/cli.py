@@ -141,7 +141,12 @@ def classification_train_main(args) -> dict:
                            dist_rank=get_rank())
     writer = SummaryWriter(str(run_dir)) if is_main_process() else None
 
-    model = build_model(args.model, num_classes=args.num_classes).to(device)
+    try:  # size-dependent models (swin, vit, transfg) take img_size
+        model = build_model(args.model, num_classes=args.num_classes,
+                            img_size=args.img_size)
+    except TypeError:
+        model = build_model(args.model, num_classes=args.num_classes)
+    model = model.to(device)
     if args.weights:
         load_pretrained(model, args.weights, logger=logger)
     if args.syncbn and get_world_size() > 1:
