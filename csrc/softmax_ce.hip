@@ -2,10 +2,22 @@
 // Covers the reference's nn.CrossEntropyLoss / timm LabelSmoothingCrossEntropy /
 // SoftTargetCrossEntropy call sites (swin main.py:111-117, everywhere else).
 // Hard-label path (int64 targets) and soft-target path (B,C probabilities).
+#include <cstdlib>
+
 #include "common.h"
 #include "vec.h"
 
 namespace dla {
+
+// The wave-per-row CE variants are gated behind DLA_CE_WAVE=1 until they get
+// hardware numerics validation (ROADMAP item; __expf/__logf are fast-math).
+inline bool ce_wave_enabled() {
+  static const bool v = []{
+    const char* e = std::getenv("DLA_CE_WAVE");
+    return e != nullptr && e[0] == '1';
+  }();
+  return v;
+}
 
 // one block per row; returns per-row loss and saves lse for backward
 template <typename dev_t>
@@ -220,7 +232,7 @@ std::vector<torch::Tensor> softmax_ce_fwd(torch::Tensor logits,
   const int grid = (int)std::min<int64_t>(B, dla::kMaxGrid);
   DLA_DISPATCH_FLOAT_TYPES(logits.scalar_type(), "softmax_ce_fwd", [&] {
     constexpr int VMAX = 16 / (int)sizeof(dev_t);
-    if (C % VMAX == 0 && C <= 64 * VMAX * 8) {
+    if (dla::ce_wave_enabled() && C % VMAX == 0 && C <= 64 * VMAX * 8) {
       const int g = (int)std::min<int64_t>((B + 3) / 4, dla::kMaxGrid);
       hipLaunchKernelGGL(
           (dla::ce_fwd_wave_kernel<dev_t, VMAX>), dim3(g), dim3(256), 0,
@@ -255,7 +267,7 @@ torch::Tensor softmax_ce_bwd(torch::Tensor logits,
   const int grid = (int)std::min<int64_t>(B, dla::kMaxGrid);
   DLA_DISPATCH_FLOAT_TYPES(logits.scalar_type(), "softmax_ce_bwd", [&] {
     constexpr int VMAX = 16 / (int)sizeof(dev_t);
-    if (C % VMAX == 0 && C <= 64 * VMAX * 8) {
+    if (dla::ce_wave_enabled() && C % VMAX == 0 && C <= 64 * VMAX * 8) {
       const int g = (int)std::min<int64_t>((B + 3) / 4, dla::kMaxGrid);
       hipLaunchKernelGGL(
           (dla::ce_bwd_wave_kernel<dev_t, VMAX>), dim3(g), dim3(256), 0,
