@@ -114,3 +114,52 @@ def test_prefetcher_requires_gpu_shapes():
     # CPU-side structural check only: DataPrefetcher is GPU-only
     from deeplearning_amd.data import DataPrefetcher
     assert DataPrefetcher is not None
+
+
+def test_rocpd_stats_tool(tmp_path):
+    """tools/rocpd_stats.py summarizes a rocprofv3-style sqlite db."""
+    import sqlite3
+    import subprocess
+    import sys
+    from pathlib import Path
+
+    db = tmp_path / "r.db"
+    con = sqlite3.connect(db)
+    con.execute("CREATE TABLE kernels (kernel_name TEXT, start INT, end INT)")
+    rows = [("slow_kernel", 0, 1000), ("slow_kernel", 2000, 3200),
+            ("fast_kernel", 1000, 1100), ("late_kernel", 9000, 9500)]
+    con.executemany("INSERT INTO kernels VALUES (?,?,?)", rows)
+    con.commit()
+    con.close()
+    repo = Path(__file__).resolve().parents[1]
+    out = tmp_path / "s.csv"
+    r = subprocess.run(
+        [sys.executable, str(repo / "tools/rocpd_stats.py"), str(db),
+         "-o", str(out)], capture_output=True, text=True)
+    assert r.returncode == 0, r.stderr
+    lines = out.read_text().strip().splitlines()
+    assert lines[0].startswith("kernel,")
+    assert lines[1].startswith("slow_kernel,2200,2")
+    # tail filter keeps only the late kernel
+    r2 = subprocess.run(
+        [sys.executable, str(repo / "tools/rocpd_stats.py"), str(db),
+         "--tail-frac", "0.2", "-o", str(out)],
+        capture_output=True, text=True)
+    lines2 = out.read_text().strip().splitlines()
+    assert len(lines2) == 2 and lines2[1].startswith("late_kernel")
+
+
+def test_generic_export_covers_pose_model(tmp_path):
+    """others/deploy export handles any registry model (here: pose HRNet)."""
+    import subprocess
+    import sys
+    from pathlib import Path
+
+    repo = Path(__file__).resolve().parents[1]
+    r = subprocess.run(
+        [sys.executable, str(repo / "projects/others/deploy/export_onnx.py"),
+         "--model", "hrnet_w18_pose", "--img-size", "64",
+         "--out", str(tmp_path / "pose.onnx")],
+        capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-1500:]
+    assert "exported" in r.stdout
