@@ -192,3 +192,54 @@ def det_train_main(args, model_kwargs=None) -> dict:
     logger.info(f"final mAP {stats['mAP']:.4f} mAP50 {stats['mAP50']:.4f}")
     cleanup()
     return {"mAP": stats["mAP"], "run_dir": str(run_dir)}
+
+
+def detect_main(default_model: str, num_classes: int = 21):
+    """Single-image detection CLI: load weights, print + optionally draw
+    detections (ref detection/yolov5/detect.py, fasterRcnn predict.py)."""
+    import argparse as _ap
+
+    from PIL import Image, ImageDraw
+
+    from ..core.checkpoint import load_pretrained
+    from ..core.env import select_device
+    from ..data.transforms import pil_to_tensor
+    from ..models import build_model
+
+    p = _ap.ArgumentParser()
+    p.add_argument("image")
+    p.add_argument("--model", default=default_model)
+    p.add_argument("--weights", required=True)
+    p.add_argument("--num-classes", type=int, default=num_classes)
+    p.add_argument("--score-thresh", type=float, default=0.5)
+    p.add_argument("--device", default="cuda")
+    p.add_argument("--save", default="", help="write annotated PNG here")
+    p.add_argument("--min-size", type=int, default=800)
+    p.add_argument("--max-size", type=int, default=1333)
+    args = p.parse_args()
+
+    device = select_device(args.device)
+    model = build_model(args.model, num_classes=args.num_classes,
+                        min_size=args.min_size,
+                        max_size=args.max_size).to(device)
+    load_pretrained(model, args.weights)
+    model.eval()
+    img = Image.open(args.image).convert("RGB")
+    x = pil_to_tensor(img).to(device)
+    with torch.no_grad():
+        det = model([x])[0]
+    keep = det["scores"] > args.score_thresh
+    boxes = det["boxes"][keep].cpu()
+    scores = det["scores"][keep].cpu()
+    labels = det["labels"][keep].cpu()
+    for b, s, l in zip(boxes.tolist(), scores.tolist(), labels.tolist()):
+        print(f"class {l}  score {s:.3f}  box "
+              f"[{b[0]:.1f}, {b[1]:.1f}, {b[2]:.1f}, {b[3]:.1f}]")
+    if args.save:
+        draw = ImageDraw.Draw(img)
+        for b, l in zip(boxes.tolist(), labels.tolist()):
+            draw.rectangle(b, outline="red", width=2)
+            draw.text((b[0], max(b[1] - 10, 0)), str(int(l)), fill="red")
+        img.save(args.save)
+        print(f"saved {args.save}")
+    return boxes, scores, labels

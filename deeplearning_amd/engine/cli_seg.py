@@ -171,3 +171,47 @@ def seg_train_main(args) -> dict:
                             model, optimizer, scheduler, epoch)
     cleanup()
     return {"miou": miou, "run_dir": str(run_dir)}
+
+
+def seg_predict_main(default_model: str, num_classes: int = 21):
+    """Single-image segmentation CLI: write the predicted mask as a palette
+    PNG (ref Image_segmentation/U-Net predict, DeepLabV3 predict)."""
+    import argparse as _ap
+
+    import numpy as np
+    from PIL import Image
+
+    from ..core.checkpoint import load_pretrained
+    from ..core.env import select_device
+    from ..data.transforms import pil_to_tensor
+    from ..models import build_model
+
+    p = _ap.ArgumentParser()
+    p.add_argument("image")
+    p.add_argument("--model", default=default_model)
+    p.add_argument("--weights", required=True)
+    p.add_argument("--num-classes", type=int, default=num_classes)
+    p.add_argument("--device", default="cuda")
+    p.add_argument("--out", default="mask.png")
+    args = p.parse_args()
+
+    device = select_device(args.device)
+    model = build_model(args.model, num_classes=args.num_classes).to(device)
+    load_pretrained(model, args.weights)
+    model.eval()
+    img = Image.open(args.image).convert("RGB")
+    x = pil_to_tensor(img)[None].to(device)
+    with torch.no_grad():
+        out = model(x)
+        logits = out["out"] if isinstance(out, dict) else out
+    mask = logits.argmax(1)[0].byte().cpu().numpy()
+    pal = Image.fromarray(mask, mode="P")
+    # simple deterministic palette
+    palette = []
+    for c in range(256):
+        palette += [(c * 37) % 256, (c * 91) % 256, (c * 173) % 256]
+    pal.putpalette(palette)
+    pal.save(args.out)
+    classes = sorted(int(v) for v in np.unique(mask))
+    print(f"classes present: {classes}; wrote {args.out}")
+    return mask
