@@ -6,6 +6,7 @@ from .samplers import (GroupedBatchSampler, InfiniteSampler,  # noqa: F401
                        SubsetRandomSampler, YoloBatchSampler,
                        create_aspect_ratio_groups)
 from .synthetic import DeviceBatchLoader, SyntheticClassification  # noqa: F401
-from .transforms import (Compose, Mixup, Normalize,  # noqa: F401
-                         RandomResizedCrop, classification_eval_transform,
+from .transforms import (ColorJitter, Compose, Mixup,  # noqa: F401
+                         Normalize, RandomErasing, RandomResizedCrop,
+                         classification_eval_transform,
                          classification_train_transform)

@@ -154,3 +154,54 @@ class Mixup:
             lam = float(np.random.beta(self.mixup_alpha, self.mixup_alpha))
             x = lam * x + (1 - lam) * x[perm]
         return x, self._one_hot(target, lam, target[perm])
+
+
+class RandomErasing:
+    """Erase a random rectangle with noise (timm RandomErasing semantics used
+    by swin dataLoader/build.py; operates on a CHW tensor post-normalize)."""
+
+    def __init__(self, p=0.25, scale=(0.02, 0.33), ratio=(0.3, 3.3)):
+        self.p = p
+        self.scale = scale
+        self.ratio = ratio
+
+    def __call__(self, t: torch.Tensor) -> torch.Tensor:
+        if random.random() > self.p:
+            return t
+        C, H, W = t.shape
+        area = H * W
+        for _ in range(10):
+            target = random.uniform(*self.scale) * area
+            ar = random.uniform(*self.ratio)
+            eh = int(round((target * ar) ** 0.5))
+            ew = int(round((target / ar) ** 0.5))
+            if eh < H and ew < W:
+                y = random.randint(0, H - eh)
+                x = random.randint(0, W - ew)
+                t = t.clone()
+                t[:, y:y + eh, x:x + ew] = torch.randn(C, eh, ew)
+                return t
+        return t
+
+
+class ColorJitter:
+    """Brightness/contrast/saturation jitter on a CHW float tensor in [0,1]
+    (applied before Normalize)."""
+
+    def __init__(self, brightness=0.4, contrast=0.4, saturation=0.4):
+        self.brightness = brightness
+        self.contrast = contrast
+        self.saturation = saturation
+
+    def __call__(self, t: torch.Tensor) -> torch.Tensor:
+        if self.brightness:
+            t = t * (1.0 + random.uniform(-self.brightness, self.brightness))
+        if self.contrast:
+            mean = t.mean()
+            f = 1.0 + random.uniform(-self.contrast, self.contrast)
+            t = (t - mean) * f + mean
+        if self.saturation:
+            gray = t.mean(0, keepdim=True)
+            f = 1.0 + random.uniform(-self.saturation, self.saturation)
+            t = (t - gray) * f + gray
+        return t.clamp(0.0, 1.0)
