@@ -1,0 +1,92 @@
+"""Subprocess smoke tests for every projects/others tool and the light
+train scripts not covered elsewhere — keeps the whole CLI surface green."""
+import subprocess
+import sys
+from pathlib import Path
+
+import pytest
+
+REPO = Path(__file__).resolve().parents[1]
+
+
+def _run(script, *args, timeout=300):
+    return subprocess.run([sys.executable, str(REPO / script), *args],
+                          capture_output=True, text=True, timeout=timeout,
+                          cwd=REPO)
+
+
+def test_normalization_rederivations():
+    r = _run("projects/others/normalization/normalization.py")
+    assert r.returncode == 0, r.stderr[-1000:]
+    assert "group_norm" in r.stdout
+
+
+def test_tensorboard_demo():
+    r = _run("projects/others/tensorboard_test/tb_demo.py")
+    assert r.returncode == 0, r.stderr[-1000:]
+
+
+def test_visualize_feature_maps(tmp_path):
+    r = _run("projects/others/visual_weight_feature_map_test/visualize.py",
+             "--out-prefix", str(tmp_path / "v"))
+    assert r.returncode == 0, r.stderr[-1000:]
+    assert (tmp_path / "v_kernels.png").exists()
+    assert (tmp_path / "v_featmaps.png").exists()
+
+
+def test_custom_op_demo():
+    r = _run("projects/others/deploy/custom_op_demo.py")
+    assert r.returncode == 0, r.stderr[-1000:]
+    assert "native / aten / torchscript" in r.stdout
+
+
+def test_load_weights_walkthrough():
+    r = _run("projects/others/load_weights_test/partial_load.py")
+    assert r.returncode == 0, r.stderr[-1000:]
+    assert "module.-prefix strip OK" in r.stdout
+
+
+def test_fpn_demo():
+    r = _run("projects/detection/FPN/fpn_demo.py")
+    assert r.returncode == 0, r.stderr[-1000:]
+    assert "P0:" in r.stdout and "Ppool:" in r.stdout
+
+
+def test_label_convert_roundtrip(tmp_path):
+    xml = ("<annotation><filename>a.jpg</filename><size><width>100</width>"
+           "<height>80</height></size><object><name>cat</name><bndbox>"
+           "<xmin>10</xmin><ymin>20</ymin><xmax>50</xmax><ymax>60</ymax>"
+           "</bndbox></object></annotation>")
+    xf = tmp_path / "a.xml"
+    xf.write_text(xml)
+    r = _run("projects/others/label_convert/convert.py", "voc2yolo",
+             str(xf), "--classes", "cat")
+    assert r.returncode == 0 and r.stdout.startswith("0 0.3")
+    yf = tmp_path / "a.txt"
+    yf.write_text(r.stdout)
+    r2 = _run("projects/others/label_convert/convert.py", "yolo2voc",
+              str(yf), "--classes", "cat", "--img-size", "100", "80")
+    assert r2.returncode == 0
+    assert "<xmin>10</xmin>" in r2.stdout and "<ymax>60</ymax>" in r2.stdout
+
+
+@pytest.mark.slow
+def test_madnet_cli():
+    r = _run("projects/deep_stereo/MadNet/train.py", "--steps", "2",
+             "--device", "cpu", "--height", "64", "--width", "128")
+    assert r.returncode == 0, r.stderr[-1000:]
+    assert "photometric loss" in r.stdout
+
+
+@pytest.mark.slow
+def test_fewshot_cli():
+    r = _run("projects/Image_segmentation/few_shot_segmentation/train.py",
+             "--episodes", "1", "--img-size", "64", "--device", "cpu")
+    assert r.returncode == 0, r.stderr[-1000:]
+
+
+@pytest.mark.slow
+def test_happy_whale_cli():
+    r = _run("projects/metric_learning/Happy-Whale/train.py", "--epochs",
+             "1", "--batch-size", "4", "--device", "cpu")
+    assert r.returncode == 0, r.stderr[-1000:]
