@@ -62,6 +62,9 @@ def classification_argparser(default_model: str, **defaults):
                    "'auto'")
     p.add_argument("--amp", action="store_true", default=True)
     p.add_argument("--no-amp", dest="amp", action="store_false")
+    p.add_argument("--amp-dtype", default="bf16", choices=["bf16", "fp16"],
+                   help="fp16 adds a GradScaler (reference NativeScaler "
+                        "semantics, swin utils/torch_utils.py:297-323)")
     p.add_argument("--seed", type=int, default=0)
     p.add_argument("--output", default="runs")
     p.add_argument("--name", default=defaults.get("name", "exp"))
@@ -183,6 +186,12 @@ def classification_train_main(args) -> dict:
         logger.info(f"resumed from {resume} at epoch {start_epoch}")
 
     amp = args.amp and device.type == "cuda"
+    amp_dtype = torch.float16 if getattr(args, "amp_dtype", "bf16") == "fp16" \
+        else torch.bfloat16
+    # fp16 needs a loss scaler (reference NativeScalerWithGradNormCount);
+    # bf16 has the dynamic range to train unscaled
+    scaler = torch.amp.GradScaler(
+        "cuda", enabled=amp and amp_dtype == torch.float16)
     for epoch in range(start_epoch, args.epochs):
         if train_sampler is not None:
             train_sampler.set_epoch(epoch)
@@ -193,8 +202,7 @@ def classification_train_main(args) -> dict:
         for it, (x, y) in enumerate(train_loader):
             x = x.to(device, non_blocking=True)
             y = y.to(device, non_blocking=True)
-            with torch.autocast(device.type, dtype=torch.bfloat16,
-                                enabled=amp):
+            with torch.autocast(device.type, dtype=amp_dtype, enabled=amp):
                 out = model(x)
                 logits = out[0] if isinstance(out, tuple) else out
                 loss = cross_entropy(logits, y)
@@ -202,14 +210,16 @@ def classification_train_main(args) -> dict:
                     for aux in out[1:]:
                         if aux is not None:
                             loss = loss + 0.3 * cross_entropy(aux, y)
-            (loss / args.accumulate_steps).backward()
+            scaler.scale(loss / args.accumulate_steps).backward()
             if (it + 1) % args.accumulate_steps == 0:
                 if args.clip_grad > 0:
+                    scaler.unscale_(optimizer)
                     torch.nn.utils.clip_grad_norm_(params, args.clip_grad)
                 finalize = getattr(model, "finalize", None)
                 if finalize is not None:
                     finalize()
-                optimizer.step()
+                scaler.step(optimizer)
+                scaler.update()
                 optimizer.zero_grad(set_to_none=True)
                 scheduler.step()
             with torch.no_grad():
