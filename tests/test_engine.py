@@ -118,3 +118,13 @@ def test_exp_config_as_code():
     import pytest as _p
     with _p.raises(AttributeError):
         YoloxExp().merge(["nope", "1"])
+
+
+def test_count_flops_resnet18():
+    from deeplearning_amd.engine.metrics import count_flops
+    from deeplearning_amd.models import build_model
+    m = build_model("resnet18", num_classes=1000)
+    flops, params = count_flops(m, (1, 3, 224, 224))
+    # published resnet18: ~1.8 GMac = 3.6 GFLOPs, 11.7 M params
+    assert 3.2e9 < flops < 4.2e9, flops
+    assert 11e6 < params < 12.5e6, params
