@@ -66,6 +66,9 @@ def classification_argparser(default_model: str, **defaults):
                    help="fp16 adds a GradScaler (reference NativeScaler "
                         "semantics, swin utils/torch_utils.py:297-323)")
     p.add_argument("--seed", type=int, default=0)
+    p.add_argument("--deterministic", action="store_true",
+                   help="cudnn.deterministic + fixed seeds "
+                        "(ref FCOS trainers/trainer.py:57-66)")
     p.add_argument("--output", default="runs")
     p.add_argument("--name", default=defaults.get("name", "exp"))
     p.add_argument("--syncbn", action="store_true")
@@ -134,7 +137,8 @@ def classification_train_main(args) -> dict:
         torch.cuda.set_device(device)
     else:
         device = select_device(args.device)
-    seed_everything(args.seed, rank=get_rank())
+    seed_everything(args.seed, rank=get_rank(),
+                    deterministic=getattr(args, 'deterministic', False))
 
     run_dir = Path(increment_path(Path(args.output) / args.name,
                                   exist_ok=False)) \

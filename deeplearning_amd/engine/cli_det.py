@@ -85,6 +85,9 @@ def det_argparser(default_model: str, **defaults):
     p.add_argument("--device", default="cuda")
     p.add_argument("--workers", type=int, default=2)
     p.add_argument("--seed", type=int, default=0)
+    p.add_argument("--deterministic", action="store_true",
+                   help="cudnn.deterministic + fixed seeds "
+                        "(ref FCOS trainers/trainer.py:57-66)")
     p.add_argument("--amp", action="store_true", default=True)
     p.add_argument("--no-amp", dest="amp", action="store_false")
     p.add_argument("--output", default="runs")
@@ -115,7 +118,8 @@ def det_train_main(args, model_kwargs=None) -> dict:
         torch.cuda.set_device(device)
     else:
         device = select_device(args.device)
-    seed_everything(args.seed, rank=get_rank())
+    seed_everything(args.seed, rank=get_rank(),
+                    deterministic=getattr(args, 'deterministic', False))
 
     run_dir = Path(increment_path(Path(args.output) / args.name)) \
         if is_main_process() else Path(args.output) / args.name
