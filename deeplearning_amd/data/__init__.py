@@ -1,6 +1,7 @@
-from .datasets import (COCODetectionDataset, ClassificationDataset,  # noqa: F401
-                       MosaicDetection, SegmentationDataset,
-                       VOCDetectionDataset, mosaic4, read_split_data)
+from .datasets import (CachedImageFolder, COCODetectionDataset,  # noqa: F401
+                       ClassificationDataset, MosaicDetection,
+                       SegmentationDataset, VOCDetectionDataset,
+                       ZipImageDataset, mosaic4, read_split_data)
 from .prefetcher import DataPrefetcher  # noqa: F401
 from .samplers import (GroupedBatchSampler, InfiniteSampler,  # noqa: F401
                        SubsetRandomSampler, YoloBatchSampler,

@@ -264,3 +264,79 @@ class MosaicDetection(Dataset):
                        self.out_size)
 
     collate_fn = staticmethod(VOCDetectionDataset.collate_fn)
+
+
+class CachedImageFolder(Dataset):
+    """Class-per-folder dataset with in-memory caching.
+
+    Reference parity: classification/swin_transformer/dataLoader/
+    cached_image_folder.py:64 — cache_mode: 'no' (read from disk every time),
+    'full' (decode once, keep tensors), 'part' (each worker caches its shard
+    lazily).
+    """
+
+    def __init__(self, root, transform=None, cache_mode="no", val_rate=0.0,
+                 split="train", seed=0):
+        tp, tl, vp, vl, self.classes = read_split_data(root, val_rate, seed)
+        self.paths, self.labels = (tp, tl) if split == "train" else (vp, vl)
+        self.transform = transform
+        assert cache_mode in ("no", "part", "full")
+        self.cache_mode = cache_mode
+        self._cache = {}
+        if cache_mode == "full":
+            for i in range(len(self.paths)):
+                self._cache[i] = self._load(i)
+
+    def _load(self, i):
+        return Image.open(self.paths[i]).convert("RGB")
+
+    def __len__(self):
+        return len(self.paths)
+
+    def __getitem__(self, i):
+        if self.cache_mode == "no":
+            img = self._load(i)
+        elif i in self._cache:
+            img = self._cache[i]
+        else:
+            img = self._load(i)
+            self._cache[i] = img  # 'part': lazily fill this worker's shard
+        out = self.transform(img) if self.transform else pil_to_tensor(img)
+        return out, self.labels[i]
+
+
+class ZipImageDataset(Dataset):
+    """Images inside a zip archive + a (member, label) index list.
+
+    Reference parity: swin dataLoader/zipreader.py:23 (ZipReader) — one zip
+    handle per worker process (zipfile handles are not fork-safe).
+    """
+
+    def __init__(self, zip_path, index, transform=None):
+        """index: list of (member_name, label) pairs."""
+        self.zip_path = str(zip_path)
+        self.index = list(index)
+        self.transform = transform
+        self._zf = None
+        self._pid = None
+
+    def _zip(self):
+        pid = os.getpid()
+        if self._zf is None or self._pid != pid:
+            import zipfile
+
+            self._zf = zipfile.ZipFile(self.zip_path, "r")
+            self._pid = pid
+        return self._zf
+
+    def __len__(self):
+        return len(self.index)
+
+    def __getitem__(self, i):
+        name, label = self.index[i]
+        import io
+
+        with self._zip().open(name) as f:
+            img = Image.open(io.BytesIO(f.read())).convert("RGB")
+        out = self.transform(img) if self.transform else pil_to_tensor(img)
+        return out, label

@@ -196,3 +196,38 @@ def test_predict_cli(tmp_path):
              "--img-size", "64", "--device", "cpu")
     assert r.returncode == 0, r.stderr[-2000:]
     assert "class" in r.stdout
+
+
+def test_cached_image_folder_modes(tmp_path):
+    from deeplearning_amd.data import CachedImageFolder
+    for c in ["a", "b"]:
+        (tmp_path / c).mkdir()
+        for i in range(4):
+            Image.fromarray(np.random.randint(
+                0, 255, (16, 16, 3), dtype=np.uint8)).save(
+                    tmp_path / c / f"{i}.png")
+    for mode in ("no", "part", "full"):
+        ds = CachedImageFolder(tmp_path, cache_mode=mode)
+        x, y = ds[0]
+        assert x.shape == (3, 16, 16) and y in (0, 1)
+        x2, _ = ds[0]
+        assert torch.equal(x, x2)
+    full = CachedImageFolder(tmp_path, cache_mode="full")
+    assert len(full._cache) == len(full)
+
+
+def test_zip_image_dataset(tmp_path):
+    import zipfile
+
+    from deeplearning_amd.data import ZipImageDataset
+    zp = tmp_path / "imgs.zip"
+    with zipfile.ZipFile(zp, "w") as zf:
+        for i in range(3):
+            img_path = tmp_path / f"i{i}.png"
+            Image.fromarray(np.random.randint(
+                0, 255, (8, 8, 3), dtype=np.uint8)).save(img_path)
+            zf.write(img_path, f"imgs/i{i}.png")
+    ds = ZipImageDataset(zp, [(f"imgs/i{i}.png", i) for i in range(3)])
+    x, y = ds[2]
+    assert x.shape == (3, 8, 8) and y == 2
+    assert len(ds) == 3
