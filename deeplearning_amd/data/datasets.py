@@ -104,6 +104,17 @@ class VOCDetectionDataset(Dataset):
                 torch.tensor(labels, dtype=torch.int64),
                 torch.tensor(iscrowd, dtype=torch.int64))
 
+    def get_height_and_width(self, i):
+        """Read (h, w) from the XML only — lets GroupedBatchSampler group by
+        aspect ratio without decoding images (ref fasterRcnn voc_dataset)."""
+        tree = ET.parse(self.root / "Annotations" / f"{self.ids[i]}.xml")
+        size = tree.find("size")
+        if size is not None:
+            return (int(size.find("height").text),
+                    int(size.find("width").text))
+        img = Image.open(self.root / "JPEGImages" / f"{self.ids[i]}.jpg")
+        return img.height, img.width
+
     def __getitem__(self, i):
         img_id = self.ids[i]
         img = Image.open(self.root / "JPEGImages" / f"{img_id}.jpg")

@@ -45,11 +45,13 @@ def test_voc_dataset_parsing():
         Image.fromarray(np.zeros((80, 100, 3), dtype=np.uint8)).save(
             root / "JPEGImages" / "im0.jpg")
         (root / "Annotations" / "im0.xml").write_text(
-            "<annotation><object><name>cat</name><bndbox>"
+            "<annotation><size><width>100</width><height>80</height></size>"
+            "<object><name>cat</name><bndbox>"
             "<xmin>10</xmin><ymin>20</ymin><xmax>50</xmax><ymax>60</ymax>"
             "</bndbox></object></annotation>")
         (root / "ImageSets" / "Main" / "train.txt").write_text("im0\n")
         ds = VOCDetectionDataset(root, "train")
+        assert ds.get_height_and_width(0) == (80, 100)
         img, target = ds[0]
         assert img.shape == (3, 80, 100)
         assert target["boxes"].tolist() == [[10.0, 20.0, 50.0, 60.0]]
