@@ -44,10 +44,22 @@ class BatchNorm2d(nn.BatchNorm2d):
                          track_running_stats=True)
         self.relu = relu
 
+    def _eaf(self) -> float:
+        """Exponential average factor; momentum=None means cumulative average
+        (torch semantics — what SWA's update_bn relies on)."""
+        if self.momentum is not None:
+            return self.momentum
+        if self.training and self.num_batches_tracked is not None:
+            return 1.0 / float(max(int(self.num_batches_tracked), 1))
+        return 0.0
+
     def forward(self, x):
         if not use_hip(x):
+            if self.training and self.momentum is None and \
+                    self.num_batches_tracked is not None:
+                self.num_batches_tracked.add_(1)
             y = F.batch_norm(x, self.running_mean, self.running_var, self.weight,
-                             self.bias, self.training, self.momentum, self.eps)
+                             self.bias, self.training, self._eaf(), self.eps)
             return torch.relu(y) if self.relu else y
         if self.running_mean.dtype != torch.float32:
             # .to(bf16) on the module converts buffers; the HIP kernel keeps
@@ -58,7 +70,8 @@ class BatchNorm2d(nn.BatchNorm2d):
             if self.num_batches_tracked is not None:
                 self.num_batches_tracked.add_(1)
             return _BNFn.apply(x, self.weight, self.bias, self.running_mean,
-                               self.running_var, self.momentum, self.eps, self.relu)
+                               self.running_var, self._eaf(), self.eps,
+                               self.relu)
         rstd = torch.rsqrt(self.running_var.float() + self.eps)
         scale = self.weight.float() * rstd
         shift = self.bias.float() - self.running_mean.float() * scale

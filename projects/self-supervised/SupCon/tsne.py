@@ -24,7 +24,7 @@ PALETTE = [(228, 26, 28), (55, 126, 184), (77, 175, 74), (152, 78, 163),
 
 
 def scatter_png(emb, labels, path, size=512):
-    emb = (emb - emb.min(0)) / (emb.ptp(0) + 1e-9)
+    emb = (emb - emb.min(0)) / (np.ptp(emb, axis=0) + 1e-9)
     img = Image.new("RGB", (size, size), "white")
     draw = ImageDraw.Draw(img)
     for (x, y), l in zip(emb, labels):
