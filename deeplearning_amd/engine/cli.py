@@ -72,6 +72,10 @@ def classification_argparser(default_model: str, **defaults):
     p.add_argument("--output", default="runs")
     p.add_argument("--name", default=defaults.get("name", "exp"))
     p.add_argument("--syncbn", action="store_true")
+    p.add_argument("--mixup", action="store_true",
+                   help="Mixup/CutMix soft-target training "
+                        "(swin dataLoader/build.py:90-96)")
+    p.add_argument("--label-smoothing", type=float, default=0.0)
     p.add_argument("--accumulate-steps", type=int, default=1)
     p.add_argument("--clip-grad", type=float, default=0.0)
     p.add_argument("--synthetic-size", type=int, default=256,
@@ -189,6 +193,13 @@ def classification_train_main(args) -> dict:
         best_acc = ckpt.get("max_accuracy", 0.0)
         logger.info(f"resumed from {resume} at epoch {start_epoch}")
 
+    mixup_fn = None
+    if getattr(args, "mixup", False):
+        from ..data import Mixup
+        mixup_fn = Mixup(num_classes=args.num_classes,
+                         label_smoothing=getattr(args, "label_smoothing", 0.1))
+    smoothing = getattr(args, "label_smoothing", 0.0)
+
     amp = args.amp and device.type == "cuda"
     amp_dtype = torch.float16 if getattr(args, "amp_dtype", "bf16") == "fp16" \
         else torch.bfloat16
@@ -206,14 +217,21 @@ def classification_train_main(args) -> dict:
         for it, (x, y) in enumerate(train_loader):
             x = x.to(device, non_blocking=True)
             y = y.to(device, non_blocking=True)
+            y_hard = y
+            if mixup_fn is not None:
+                x, y = mixup_fn(x, y)  # y becomes soft [B, C]
             with torch.autocast(device.type, dtype=amp_dtype, enabled=amp):
                 out = model(x)
                 logits = out[0] if isinstance(out, tuple) else out
-                loss = cross_entropy(logits, y)
+                if mixup_fn is not None:
+                    from ..ops import soft_target_cross_entropy
+                    loss = soft_target_cross_entropy(logits, y)
+                else:
+                    loss = cross_entropy(logits, y, smoothing=smoothing)
                 if isinstance(out, tuple):  # aux heads (GoogLeNet)
                     for aux in out[1:]:
                         if aux is not None:
-                            loss = loss + 0.3 * cross_entropy(aux, y)
+                            loss = loss + 0.3 * cross_entropy(aux, y_hard)
             scaler.scale(loss / args.accumulate_steps).backward()
             if (it + 1) % args.accumulate_steps == 0:
                 if args.clip_grad > 0:
@@ -227,7 +245,7 @@ def classification_train_main(args) -> dict:
                 optimizer.zero_grad(set_to_none=True)
                 scheduler.step()
             with torch.no_grad():
-                acc1 = accuracy(logits.float(), y)[0]
+                acc1 = accuracy(logits.float(), y_hard)[0]
             loss_m.update(float(loss.detach()), x.shape[0])
             acc_m.update(float(acc1), x.shape[0])
         logger.info(f"epoch {epoch}: loss {loss_m.avg:.4f} "
