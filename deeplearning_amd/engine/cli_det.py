@@ -95,6 +95,9 @@ def det_argparser(default_model: str, **defaults):
     p.add_argument("--synthetic-size", type=int, default=16)
     p.add_argument("--eval-every", type=int, default=0,
                    help="0 = eval only at the end")
+    p.add_argument("--aspect-ratio-group-factor", type=int, default=-1,
+                   help=">=0: group batches by aspect ratio (2k+1 buckets; "
+                        "ref fasterRcnn utils/group_by_aspect_ratio.py)")
     return p
 
 
@@ -135,10 +138,25 @@ def det_train_main(args, model_kwargs=None) -> dict:
 
     ds = build_det_dataset(args)
     sampler = DistributedSampler(ds) if get_world_size() > 1 else None
-    loader = DataLoader(ds, batch_size=args.batch_size,
-                        shuffle=sampler is None, sampler=sampler,
-                        num_workers=args.workers,
-                        collate_fn=getattr(type(ds), "collate_fn", None))
+    group_factor = getattr(args, "aspect_ratio_group_factor", -1)
+    if group_factor >= 0:
+        from torch.utils.data import RandomSampler
+
+        from ..data.samplers import (GroupedBatchSampler,
+                                     compute_aspect_ratios,
+                                     create_aspect_ratio_groups)
+        base = sampler if sampler is not None else RandomSampler(ds)
+        group_ids = create_aspect_ratio_groups(
+            compute_aspect_ratios(ds), k=group_factor)
+        batch_sampler = GroupedBatchSampler(base, group_ids, args.batch_size)
+        loader = DataLoader(ds, batch_sampler=batch_sampler,
+                            num_workers=args.workers,
+                            collate_fn=getattr(type(ds), "collate_fn", None))
+    else:
+        loader = DataLoader(ds, batch_size=args.batch_size,
+                            shuffle=sampler is None, sampler=sampler,
+                            num_workers=args.workers,
+                            collate_fn=getattr(type(ds), "collate_fn", None))
 
     optimizer = torch.optim.SGD(
         [p for p in model.parameters() if p.requires_grad], lr=args.lr,
