@@ -6,9 +6,11 @@ from pathlib import Path
 sys.path.insert(0, str(Path(__file__).resolve().parents[3]))
 
 import argparse
+import random
 import time
 
 import torch
+import torch.nn.functional as TF
 from torch.utils.data import DataLoader
 
 from deeplearning_amd.core.checkpoint import save_checkpoint, save_weights
@@ -34,6 +36,9 @@ if __name__ == "__main__":
     p.add_argument("--device", default="cuda")
     p.add_argument("--output", default="runs")
     p.add_argument("--synthetic-size", type=int, default=16)
+    p.add_argument("--multiscale", action="store_true",
+                   help="random input size every 10 iters, broadcast-synced "
+                        "across ranks (ref yolox_base.py:167-187)")
     args = p.parse_args()
 
     init_distributed()
@@ -56,6 +61,8 @@ if __name__ == "__main__":
     opt = torch.optim.SGD(model.parameters(), lr=args.lr, momentum=0.9,
                           weight_decay=5e-4, nesterov=True)
     amp = device.type == "cuda"
+    cur_size = args.img_size
+    it_count = 0
     for epoch in range(args.epochs):
         model.train()
         t0 = time.time()
@@ -64,6 +71,23 @@ if __name__ == "__main__":
             x = torch.stack(list(images)).to(device)
             targets = [{k: v.to(device) for k, v in t.items()}
                        for t in targets]
+            if args.multiscale:
+                if it_count % 10 == 0:
+                    s_t = torch.tensor(
+                        random.choice(range(args.img_size - 64,
+                                            args.img_size + 65, 32)),
+                        device=device)
+                    if get_world_size() > 1:
+                        import torch.distributed as dist
+                        dist.broadcast(s_t, src=0)
+                    cur_size = int(s_t)
+                it_count += 1
+                if cur_size != x.shape[-1]:
+                    scale = cur_size / x.shape[-1]
+                    x = TF.interpolate(x, size=(cur_size, cur_size),
+                                       mode="bilinear", align_corners=False)
+                    for t in targets:
+                        t["boxes"] = t["boxes"] * scale
             with torch.autocast(device.type, dtype=torch.bfloat16,
                                 enabled=amp):
                 losses = model(x, targets)

@@ -75,7 +75,11 @@ if __name__ == "__main__":
                         collate_fn=SyntheticDetection.collate_fn)
     opt = torch.optim.SGD(model.parameters(), lr=args.lr, momentum=0.937,
                           weight_decay=5e-4, nesterov=True)
+    # nominal batch 64: accumulate gradients to reach it (ref train.py:176)
+    accumulate = max(round(64 / args.batch_size), 1)
     amp = device.type == "cuda"
+    opt.zero_grad(set_to_none=True)
+    it = 0
     for epoch in range(args.epochs):
         model.train()
         t0 = time.time()
@@ -87,14 +91,16 @@ if __name__ == "__main__":
                                 enabled=amp):
                 preds = model(x)
                 loss, items = compute_loss(preds, t)
-            opt.zero_grad(set_to_none=True)
             loss.backward()
-            finalize = getattr(model, "finalize", None)
-            if finalize is not None:
-                finalize()
-            opt.step()
-            if ema:
-                ema.update(model)
+            it += 1
+            if it % accumulate == 0:
+                finalize = getattr(model, "finalize", None)
+                if finalize is not None:
+                    finalize()
+                opt.step()
+                opt.zero_grad(set_to_none=True)
+                if ema:
+                    ema.update(model)
             tot += float(loss.detach())
         logger.info(f"epoch {epoch}: loss {tot / len(loader):.4f} "
                     f"({time.time() - t0:.1f}s)")
