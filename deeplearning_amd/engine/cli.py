@@ -151,6 +151,12 @@ def classification_train_main(args) -> dict:
     logger = create_logger(str(run_dir) if is_main_process() else None,
                            dist_rank=get_rank())
     writer = SummaryWriter(str(run_dir)) if is_main_process() else None
+    if is_main_process():  # dump the resolved config (ref swin main.py:349-353)
+        import json
+        with open(run_dir / "config.json", "w") as f:
+            json.dump({k: v for k, v in vars(args).items()
+                       if isinstance(v, (int, float, str, bool, list,
+                                         type(None)))}, f, indent=2)
 
     try:  # size-dependent models (swin, vit, transfg) take img_size
         model = build_model(args.model, num_classes=args.num_classes,
