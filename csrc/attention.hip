@@ -71,7 +71,12 @@ __device__ void stage_rows(const __hip_bfloat16* src, int64_t row_stride,
 }
 
 // ---------------------------------------------------------------- forward --
-template <int D, bool HAS_BIAS, bool HAS_MASK, bool SAVE_P>
+// COSINE (Swin-v2): S = logit_scale[h] * cos(q, k) instead of scale * q.k —
+// per-row q/k inverse norms are computed in-kernel; `scale` is unused and
+// `lscale` carries the per-head clamped-exp logit scales. Inference-staged
+// (backward through the normalize is not implemented; the autograd wrapper
+// only routes no-grad calls here).
+template <int D, bool HAS_BIAS, bool HAS_MASK, bool SAVE_P, bool COSINE = false>
 __global__ __launch_bounds__(512)
 void attn_fwd_kernel(const __hip_bfloat16* __restrict__ qkv,  // [B,N,3,H,D]
                      const float* __restrict__ bias,          // [H,N,N]|null
@@ -79,6 +84,7 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ qkv,  // [B,N,3,H,D]
                      __hip_bfloat16* __restrict__ out,        // [B,N,H*D]
                      __hip_bfloat16* __restrict__ p_out,      // [B,H,N,N]|null
                      float* __restrict__ stats,               // [2,B,H,N]|null
+                     const float* __restrict__ lscale,        // [H]|null
                      int B, int N, int H, int n_win, float scale) {
   constexpr int KSLICES = D / 32;
   const int b = blockIdx.x / H;
@@ -95,12 +101,26 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ qkv,  // [B,N,3,H,D]
   const int VROW = Npad + 8;
   __hip_bfloat16* p_lds = vt_lds + D * VROW;              // [nwaves][PBUF]
 
+  float* kinv = (float*)(p_lds + nwaves * PBUF);  // [Npad] (COSINE only)
+
   const int64_t bh_stride = (int64_t)3 * H * D;
   stage_rows<D>(qkv + ((int64_t)b * N * 3 + 1) * H * D + (int64_t)h * D,
                 bh_stride, k_lds, nullptr, N, Npad, VROW);
   stage_rows<D>(qkv + ((int64_t)b * N * 3 + 2) * H * D + (int64_t)h * D,
                 bh_stride, nullptr, vt_lds, N, Npad, VROW);
   __syncthreads();
+  if (COSINE) {
+    for (int key = threadIdx.x; key < Npad; key += blockDim.x) {
+      float sum = 0.f;
+      for (int d = 0; d < D; ++d) {
+        const float v = to_f32(k_lds[key * KPAD + d]);
+        sum += v * v;
+      }
+      kinv[key] = rsqrtf(fmaxf(sum, 1e-12f));
+    }
+    __syncthreads();
+  }
+  const float ls = COSINE ? lscale[h] : scale;
 
   const __hip_bfloat16* q_src = qkv + (int64_t)b * N * 3 * H * D +
                                 (int64_t)h * D;
@@ -118,6 +138,28 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ qkv,  // [B,N,3,H,D]
         const int d0 = sl * 32 + (lane >> 4) * 8;
         q_frag[sl] = *(const bf16x8*)(q_src + srow * bh_stride + d0);
       }
+    }
+    float q_inv = 1.f;  // per A-fragment row (lane&15); COSINE only
+    if (COSINE) {
+      float qs = 0.f;
+#pragma unroll
+      for (int sl = 0; sl < KSLICES; ++sl)
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          __hip_bfloat16 hv;
+          short sv = q_frag[sl][j];
+          __builtin_memcpy(&hv, &sv, sizeof(hv));
+          const float v = to_f32(hv);
+          qs += v * v;
+        }
+      // row total lives across lanes {r, r+16, r+32, r+48}
+      qs += __shfl_xor(qs, 16, 64);
+      qs += __shfl_xor(qs, 32, 64);
+      q_inv = rsqrtf(fmaxf(qs, 1e-12f));
+    }
+    __shared__ float qn_sh[8][16];  // per wave: qinv by row-in-block
+    if (COSINE) {
+      qn_sh[wave][lane & 15] = q_inv;  // lanes of a row write the same value
     }
 
     // S = Q K^T over all key tiles (kept in VGPRs)
@@ -149,7 +191,10 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ qkv,  // [B,N,3,H,D]
       for (int kt = 0; kt < 16; ++kt) {
         if (kt < n_ktiles) {
           const int key = kt * 16 + col;
-          float sv = s_acc[kt][r] * scale;
+          float sv = COSINE
+              ? s_acc[kt][r] * ls * qn_sh[wave][(lane >> 4) * 4 + r] *
+                    kinv[key]
+              : s_acc[kt][r] * scale;
           if ((HAS_BIAS || HAS_MASK) && key < N && qrow < N) {
             if (HAS_BIAS) sv += bias[((int64_t)h * N + qrow) * N + key];
             if (HAS_MASK)
@@ -652,7 +697,7 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor qkv, int64_t num_heads,
                        mask_ptr, (__hip_bfloat16*)out.data_ptr(),
                        save_p ? (__hip_bfloat16*)p.data_ptr() : nullptr,
                        save_stats ? stats.data_ptr<float>() : nullptr,
-                       B, N, H, n_win, (float)scale);
+                       nullptr, B, N, H, n_win, (float)scale);
   };
   auto d3 = [&](auto dtag) {
     using T = std::true_type;
@@ -724,4 +769,66 @@ torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B) {
                      D.data_ptr<float>());
   HIP_CHECK_ERR();
   return {D};
+}
+
+
+// Swin-v2 cosine attention forward (inference; round-2 staging — see
+// ROADMAP.md). logit_scale: [H] fp32 = clamp(exp(param), max=100).
+torch::Tensor attn_fwd_cosine(torch::Tensor qkv, int64_t num_heads,
+                              torch::Tensor logit_scale,
+                              c10::optional<torch::Tensor> bias,
+                              c10::optional<torch::Tensor> mask) {
+  TORCH_CHECK(qkv.scalar_type() == at::kBFloat16 && qkv.is_contiguous());
+  const int B = (int)qkv.size(0);
+  const int N = (int)qkv.size(1);
+  const int H = (int)num_heads;
+  const int D = (int)(qkv.numel() / ((int64_t)B * N * 3 * H));
+  TORCH_CHECK(D == 32 || D == 64);
+  TORCH_CHECK(N <= 256);
+  auto ls = logit_scale.to(torch::kFloat).contiguous();
+  TORCH_CHECK(ls.numel() == H, "logit_scale must be [H]");
+  auto out = torch::empty({B, N, (int64_t)H * D}, qkv.options());
+
+  int n_win = 1;
+  const float* bias_ptr = nullptr;
+  const float* mask_ptr = nullptr;
+  torch::Tensor bias_f, mask_f;
+  if (bias.has_value()) {
+    bias_f = bias->to(torch::kFloat).contiguous();
+    bias_ptr = bias_f.data_ptr<float>();
+  }
+  if (mask.has_value()) {
+    mask_f = mask->to(torch::kFloat).contiguous();
+    n_win = (int)mask_f.size(0);
+    mask_ptr = mask_f.data_ptr<float>();
+  }
+  const int Npad = (N + 15) & ~15;
+  const int nwaves = N > 96 ? 8 : 4;
+  const int lds = (Npad * dla::KPAD + D * (Npad + 8) + nwaves * dla::PBUF) *
+                      (int)sizeof(__hip_bfloat16) +
+                  Npad * (int)sizeof(float);  // + kinv
+  dim3 grid(B * H), block(nwaves * 64);
+  auto run = [&](auto dtag, auto btag, auto mtag) {
+    constexpr int DD = decltype(dtag)::value;
+    constexpr bool BB = decltype(btag)::value;
+    constexpr bool MM = decltype(mtag)::value;
+    hipLaunchKernelGGL(
+        (dla::attn_fwd_kernel<DD, BB, MM, false, true>), grid, block, lds,
+        dla::stream(), (const __hip_bfloat16*)qkv.data_ptr(), bias_ptr,
+        mask_ptr, (__hip_bfloat16*)out.data_ptr(), nullptr, nullptr,
+        ls.data_ptr<float>(), B, N, H, n_win, 1.0f);
+  };
+  using T = std::true_type;
+  using F = std::false_type;
+  auto d2 = [&](auto dtag) {
+    const bool bb = bias_ptr != nullptr, mm = mask_ptr != nullptr;
+    if (bb && mm) run(dtag, T{}, T{});
+    else if (bb) run(dtag, T{}, F{});
+    else if (mm) run(dtag, F{}, T{});
+    else run(dtag, F{}, F{});
+  };
+  if (D == 64) d2(std::integral_constant<int, 64>{});
+  else d2(std::integral_constant<int, 32>{});
+  HIP_CHECK_ERR();
+  return out;
 }

@@ -95,8 +95,27 @@ class WindowAttention(nn.Module):
         return bias.permute(2, 0, 1).contiguous().unsqueeze(0)  # 1, nH, N, N
 
     def forward(self, x, mask=None):
+        import os
+
         from ...ops.attention import fused_attention
         B_, N, C = x.shape
+        if self.v2 and not self.training and not torch.is_grad_enabled() \
+                and os.environ.get("DLA_V2_FUSED") == "1":
+            # staged fused cosine path (inference; see ROADMAP.md)
+            from ...ops.attention import fused_attention_cosine
+            qkv_bias = None
+            if self.q_bias is not None:
+                qkv_bias = torch.cat((
+                    self.q_bias, torch.zeros_like(self.v_bias), self.v_bias))
+            qkv = torch.nn.functional.linear(x, self.qkv.weight, qkv_bias)
+            lscale = torch.clamp(
+                self.logit_scale,
+                max=torch.log(torch.tensor(100.0,
+                                           device=x.device))).exp().flatten()
+            out = fused_attention_cosine(qkv, self.num_heads, lscale,
+                                         bias=self._bias().squeeze(0),
+                                         mask=mask)
+            return self.proj_drop(self.proj(out))
         if not self.v2 and (self.attn_drop.p == 0.0 or not self.training):
             # one fused HIP MFMA kernel: scale*QK^T + rel-pos bias (+window
             # mask) + softmax + V, straight from the packed qkv projection

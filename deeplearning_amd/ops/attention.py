@@ -118,3 +118,33 @@ def fused_attention(qkv: torch.Tensor, num_heads: int, scale: float,
             return _AttnFusedBwdFn.apply(qkv, num_heads, scale)
         return _AttnFn.apply(qkv, num_heads, scale, bias, mask)
     return _eager_attention(qkv, num_heads, scale, bias, mask)
+
+
+def fused_attention_cosine(qkv: torch.Tensor, num_heads: int,
+                           logit_scale: torch.Tensor,
+                           bias: torch.Tensor | None = None,
+                           mask: torch.Tensor | None = None) -> torch.Tensor:
+    """Swin-v2 cosine attention, inference only (round-2 staging; enable the
+    fused path in the model with DLA_V2_FUSED=1). logit_scale: [H] already
+    clamp(exp(param), max=100)."""
+    d = qkv.shape[2] // (3 * num_heads)
+    if (use_hip(qkv) and qkv.dtype == torch.bfloat16 and d in (32, 64)
+            and qkv.shape[1] <= 256 and not torch.is_grad_enabled()):
+        return ext().attn_fwd_cosine(qkv.contiguous(), num_heads,
+                                     logit_scale, bias, mask)
+    # eager reference
+    B, N, _ = qkv.shape
+    q, k, v = qkv.reshape(B, N, 3, num_heads, -1).permute(
+        2, 0, 3, 1, 4).unbind(0)
+    attn = torch.nn.functional.normalize(q.float(), dim=-1) @ \
+        torch.nn.functional.normalize(k.float(), dim=-1).transpose(-2, -1)
+    attn = attn * logit_scale.view(1, -1, 1, 1)
+    if bias is not None:
+        attn = attn + bias.unsqueeze(0).float()
+    if mask is not None:
+        nW = mask.shape[0]
+        attn = attn.view(B // nW, nW, num_heads, N, N) + \
+            mask.float().unsqueeze(1).unsqueeze(0)
+        attn = attn.view(B, num_heads, N, N)
+    attn = attn.softmax(dim=-1).to(v.dtype)
+    return (attn @ v).transpose(1, 2).reshape(B, N, -1)
