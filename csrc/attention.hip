@@ -677,6 +677,8 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor qkv, int64_t num_heads,
     TORCH_CHECK(mask_f.dim() == 3 && mask_f.size(1) == N,
                 "mask must be [nW,N,N]");
     n_win = (int)mask_f.size(0);
+    TORCH_CHECK(B % n_win == 0, "batch (", B,
+                ") must be a multiple of the window count (", n_win, ")");
     mask_ptr = mask_f.data_ptr<float>();
   }
 
