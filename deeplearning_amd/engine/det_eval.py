@@ -142,10 +142,16 @@ class DetEvaluator:
         if not ap_per_class:
             return {"mAP": 0.0, "mAP50": 0.0, "mAP75": 0.0, "per_class": {}}
         all_aps = torch.tensor(list(ap_per_class.values()))  # C, T
+
+        def at_thr(thr):
+            try:
+                return float(all_aps[:, self.iou_thrs.index(thr)].mean())
+            except ValueError:  # threshold not in this evaluator's list
+                return 0.0
         return {
             "mAP": float(all_aps.mean()),
-            "mAP50": float(all_aps[:, 0].mean()),
-            "mAP75": float(all_aps[:, 5].mean()),
+            "mAP50": at_thr(0.5),
+            "mAP75": at_thr(0.75),
             "per_class": {c: float(torch.tensor(a).mean())
                           for c, a in ap_per_class.items()},
         }
