@@ -1,4 +1,5 @@
 """CPU tests for engine: scheduler, metrics, trainer loop on a tiny model."""
+import pytest
 import torch
 import torch.nn as nn
 from torch.utils.data import DataLoader
@@ -203,3 +204,51 @@ def test_det_evaluator_vs_bruteforce_ap50():
     from deeplearning_amd.engine.det_eval import _ap_101
     ref = _ap_101(recall, precision)
     assert abs(ours - ref) < 0.02, (ours, ref)
+
+
+def test_ema_decay_ramp_and_convergence():
+    import torch.nn as nn
+
+    from deeplearning_amd.ops import ModelEMA
+    torch.manual_seed(0)
+    m = nn.Linear(4, 4)
+    ema = ModelEMA(m, decay=0.9999, tau=2000)
+    # ramp: early decay is tiny -> EMA tracks the model closely
+    assert ema.decay(1) < 0.001
+    assert 0.6 < ema.decay(2000) / 0.9999 < 0.65  # 1 - e^-1
+    with torch.no_grad():
+        for p in m.parameters():
+            p.add_(1.0)
+    ema.update(m)
+    for pe, pm in zip(ema.ema.parameters(), m.parameters()):
+        # first update: decay ~ 0 -> ema ~= model
+        assert torch.allclose(pe, pm, atol=1e-2)
+
+
+def test_accuracy_topk():
+    logits = torch.tensor([[0.1, 0.9, 0.0], [0.8, 0.05, 0.15],
+                           [0.2, 0.3, 0.5]])
+    target = torch.tensor([1, 2, 2])
+    top1, top2 = accuracy(logits, target, topk=(1, 2))
+    assert float(top1) == pytest.approx(100 * 2 / 3, rel=1e-3)
+    assert float(top2) == pytest.approx(100.0, rel=1e-3)
+
+
+def test_scheduler_resume_matches_fresh():
+    import torch.nn as nn
+
+    from deeplearning_amd.engine.scheduler import WarmupScheduler
+    m = nn.Linear(2, 2)
+    o1 = torch.optim.SGD(m.parameters(), lr=0.1)
+    s1 = WarmupScheduler(o1, total_steps=100, warmup_steps=10)
+    for _ in range(30):
+        s1.step()
+    sd = s1.state_dict()
+    o2 = torch.optim.SGD(m.parameters(), lr=0.1)
+    s2 = WarmupScheduler(o2, total_steps=100, warmup_steps=10)
+    s2.load_state_dict(sd)
+    for _ in range(5):
+        s1.step()
+        s2.step()
+    assert o1.param_groups[0]["lr"] == pytest.approx(
+        o2.param_groups[0]["lr"], rel=1e-9)
