@@ -169,3 +169,55 @@ def test_simota_assigns_inside_gt():
         (centers[:, 1] > 8) & (centers[:, 1] < 40)
     assert bool((fg & ~inside).sum() == 0) or True  # fg ⊆ candidates
     assert (matched == 0).all()
+
+
+def test_fcos_target_assignment_semantics():
+    from deeplearning_amd.models.detection.fcos import (centerness_target,
+                                                        fcos_targets)
+    # one level, stride 8, 4x4 grid of points
+    xs = (torch.arange(4).float() + 0.5) * 8
+    pts = torch.stack(torch.meshgrid(xs, xs, indexing="ij"), -1)
+    pts = pts.flip(-1).reshape(-1, 2)  # (x, y)
+    # two nested boxes: points inside both must go to the SMALLER one
+    targets = [{"boxes": torch.tensor([[0.0, 0.0, 32.0, 32.0],
+                                       [8.0, 8.0, 24.0, 24.0]]),
+                "labels": torch.tensor([1, 2])}]
+    cls_t, reg_t = fcos_targets([pts], [8], targets, num_classes=3,
+                                ranges=((-1, 1e8),), center_radius=100.0)
+    # the center point (12.5 area) lies inside both -> label 2 (smaller box)
+    centre_idx = (pts[:, 0] == 12.0).logical_and(pts[:, 1] == 12.0)
+    assert cls_t[0][centre_idx].item() == 2
+    # ltrb offsets are consistent: l+r == box width for assigned points
+    pos = cls_t[0] == 2
+    if pos.any():
+        w = reg_t[0][pos][:, 0] + reg_t[0][pos][:, 2]
+        assert torch.allclose(w, torch.full_like(w, 16.0))
+    # centerness is 1 at the exact center of a box
+    c = centerness_target(torch.tensor([[8.0, 8.0, 8.0, 8.0]]))
+    assert c.item() == pytest.approx(1.0)
+    c2 = centerness_target(torch.tensor([[2.0, 8.0, 14.0, 8.0]]))
+    assert c2.item() < 1.0
+
+
+def test_yolov5_build_targets_neighbor_expansion():
+    from deeplearning_amd.models import build_model
+    from deeplearning_amd.models.detection import ComputeLoss
+    torch.manual_seed(0)
+    m = build_model("yolov5s", num_classes=5)
+    m.train()
+    preds = m(torch.rand(1, 3, 256, 256))
+    crit = ComputeLoss(m)
+    # a box sized like the first anchor (10x13 px) at an off-center cell pos
+    t = torch.tensor([[0, 1, 0.3003, 0.3003, 10 / 256, 13 / 256]])
+    tcls, tbox, indices, anch = crit.build_targets(preds, t)
+    # P3 (stride 8) must carry assignments incl. neighbor-cell expansion
+    n_p3 = indices[0][0].numel()
+    assert n_p3 >= 2  # center cell + at least one neighbor
+    # all assigned grid coords are in range
+    _, _, gj, gi = indices[0]
+    assert int(gj.max()) < preds[0].shape[2]
+    assert int(gi.max()) < preds[0].shape[3]
+    # anchor ratio filter: a huge box must NOT match the small P3 anchors
+    t2 = torch.tensor([[0, 1, 0.5, 0.5, 0.9, 0.9]])
+    _, _, idx2, _ = crit.build_targets(preds, t2)
+    assert idx2[0][0].numel() == 0  # no P3 assignment for a 230px box
