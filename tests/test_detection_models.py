@@ -221,3 +221,30 @@ def test_yolov5_build_targets_neighbor_expansion():
     t2 = torch.tensor([[0, 1, 0.5, 0.5, 0.9, 0.9]])
     _, _, idx2, _ = crit.build_targets(preds, t2)
     assert idx2[0][0].numel() == 0  # no P3 assignment for a 230px box
+
+
+def test_simota_conflict_resolution():
+    """An anchor claimed by two gts must go to the lower-cost one."""
+    from deeplearning_amd.models.detection import simota_assign
+    torch.manual_seed(0)
+    # 4 anchor points on a line, stride 10, centers at 5,15,25,35
+    grids = torch.tensor([[0.0, 0.0], [1.0, 0.0], [2.0, 0.0], [3.0, 0.0]])
+    stride = torch.full((4,), 10.0)
+    # two overlapping gts sharing the middle anchors
+    gt = torch.tensor([[0.0, -5.0, 22.0, 5.0], [12.0, -5.0, 40.0, 5.0]])
+    labels = torch.tensor([0, 1])
+    # predictions == their own gt-ish boxes, centered at the anchors
+    centers = grids * 10 + 5
+    pred_boxes = torch.cat([centers, torch.full((4, 2), 20.0)], 1)
+    pred_cls = torch.zeros(4, 3)
+    pred_obj = torch.zeros(4)
+    fg, matched, _ = simota_assign(pred_boxes, pred_cls, pred_obj, gt,
+                                   labels, grids, stride, 3,
+                                   center_radius=10.0)
+    # every foreground anchor is matched to exactly one gt
+    assert matched.numel() == int(fg.sum())
+    assert ((matched == 0) | (matched == 1)).all()
+    # left-most fg anchor belongs to gt0, right-most to gt1
+    fg_idx = torch.where(fg)[0]
+    if fg_idx.numel() >= 2:
+        assert matched[0].item() == 0 or matched[-1].item() == 1
