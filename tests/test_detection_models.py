@@ -248,3 +248,43 @@ def test_simota_conflict_resolution():
     fg_idx = torch.where(fg)[0]
     if fg_idx.numel() >= 2:
         assert matched[0].item() == 0 or matched[-1].item() == 1
+
+
+def test_roi_heads_sampling_always_has_positives():
+    """Appending gt boxes to proposals guarantees >=1 positive sample."""
+    from deeplearning_amd.models.detection.roi_heads import RoIHeads
+    torch.manual_seed(0)
+    rh = RoIHeads(num_classes=3)
+    # proposals nowhere near the gt
+    proposals = [torch.tensor([[200.0, 200.0, 220.0, 220.0]])]
+    targets = [{"boxes": torch.tensor([[10.0, 10.0, 50.0, 50.0]]),
+                "labels": torch.tensor([2])}]
+    props, labels, regs = rh.select_training_samples(proposals, targets)
+    assert (labels[0] == 2).sum() >= 1  # the appended gt is a positive
+    assert (labels[0] == 0).sum() >= 1  # the far proposal is background
+    pos_rows = labels[0] == 2
+    # regression targets for positives decode back onto the gt box
+    decoded = rh.box_coder.decode(regs[0][pos_rows], props[0][pos_rows])
+    assert torch.allclose(decoded, targets[0]["boxes"].expand_as(decoded),
+                          atol=1e-3)
+
+
+def test_multiscale_roialign_level_routing():
+    """Small RoIs must read the finest FPN level, large RoIs the coarsest
+    (canonical k = floor(4 + log2(sqrt(area)/224)))."""
+    from deeplearning_amd.ops import MultiScaleRoIAlign
+    pool = MultiScaleRoIAlign(["0", "1", "2", "3"], output_size=2,
+                              sampling_ratio=2)
+    # constant-valued levels so the source level is observable in the output
+    feats = {}
+    for i, (size, val) in enumerate(zip([64, 32, 16, 8],
+                                        [1.0, 2.0, 3.0, 4.0])):
+        feats[str(i)] = torch.full((1, 4, size, size), val)
+    image_shapes = [(256, 256)]
+    small = [torch.tensor([[10.0, 10.0, 42.0, 42.0]])]   # 32px -> P2 (idx 0)
+    large = [torch.tensor([[0.0, 0.0, 250.0, 250.0]])]   # 250px -> P4 (idx 2)
+    out_s = pool(feats, small, image_shapes)
+    out_l = pool(feats, large, image_shapes)
+    assert torch.allclose(out_s, torch.ones_like(out_s))
+    # canonical mapping: k = floor(4 + log2(250/224)) = 4 -> P4 = index 2
+    assert torch.allclose(out_l, torch.full_like(out_l, 3.0))
