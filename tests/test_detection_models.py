@@ -288,3 +288,31 @@ def test_multiscale_roialign_level_routing():
     assert torch.allclose(out_s, torch.ones_like(out_s))
     # canonical mapping: k = floor(4 + log2(250/224)) = 4 -> P4 = index 2
     assert torch.allclose(out_l, torch.full_like(out_l, 3.0))
+
+
+def test_anchor_cache_and_dtype():
+    gen = AnchorGenerator(sizes=((32,),), aspect_ratios=((1.0,),))
+
+    class IL:
+        tensors = torch.zeros(1, 3, 32, 32)
+    feats = [torch.zeros(1, 1, 4, 4)]
+    a1 = gen(IL(), feats)[0]
+    a2 = gen(IL(), feats)[0]
+    assert a1.data_ptr() == a2.data_ptr()  # cached, not regenerated
+    assert a1.dtype == torch.float32
+    # centered 32x32 anchor at stride 8: first anchor box
+    assert a1[0].tolist() == [-16.0, -16.0, 16.0, 16.0]
+
+
+def test_transform_postprocess_rescales_back():
+    from deeplearning_amd.models.detection import GeneralizedRCNNTransform
+    tr = GeneralizedRCNNTransform(min_size=100, max_size=200)
+    tr.eval()
+    img = torch.rand(3, 50, 80)
+    il, _ = tr([img])
+    h, w = il.image_sizes[0]
+    det = [{"boxes": torch.tensor([[0.0, 0.0, float(w), float(h)]]),
+            "scores": torch.ones(1), "labels": torch.ones(1)}]
+    out = tr.postprocess(det, il.image_sizes, [(50, 80)])
+    assert torch.allclose(out[0]["boxes"],
+                          torch.tensor([[0.0, 0.0, 80.0, 50.0]]), atol=0.5)
