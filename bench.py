@@ -1,9 +1,14 @@
 #!/usr/bin/env python3
 """Flagship training-step benchmark (BASELINE.json contract).
 
-Measures whole-job images/sec for ResNet-50 (default) or ViT-B/16 at 224^2,
-bf16, synthetic data, random-init weights, full train step (forward + loss +
-backward + DDP gradient all-reduce + SGD update) on N GPUs of one node.
+Measures whole-job images/sec at 224^2, bf16, synthetic data, random-init
+weights, full train step (forward + loss + backward + DDP gradient all-reduce
++ SGD update) on N GPUs of one node.
+
+BASELINE.json's headline metric names BOTH ResNet-50 and ViT-B/16: the default
+invocation (no --model flag) measures both back-to-back and emits ONE JSON
+line whose value is the ResNet-50 images/s, with the ViT-B/16 number carried
+in config.models. Pass --model to measure a single model.
 
 Launch (driver contract):
   python bench.py --gpus 1 --steps K --warmup W
@@ -14,6 +19,7 @@ from __future__ import annotations
 
 import argparse
 import json
+import sys
 import time
 
 import torch
@@ -31,11 +37,15 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=30)
     p.add_argument("--warmup", type=int, default=10)
-    p.add_argument("--model", default="resnet50", choices=["resnet50", "vit_b16"])
+    p.add_argument("--model", default="all",
+                   choices=["all", "resnet50", "vit_b16"],
+                   help="'all' (default) = both headline models, "
+                        "value = ResNet-50 img/s, ViT in config.models")
     p.add_argument("--batch-size", type=int, default=0,
                    help="per-GPU batch (0 = model default)")
     p.add_argument("--dp", default="bucketed", choices=["bucketed", "torch"])
-    p.add_argument("--bucket-mb", type=float, default=64.0)
+    p.add_argument("--bucket-mb", type=float, default=64.0,
+                   help="DDP gradient-bucket size sweep knob (16/32/64/128)")
     p.add_argument("--no-channels-last", action="store_true")
     p.add_argument("--lr", type=float, default=0.1)
     p.add_argument("--graph", default="off", choices=["auto", "on", "off"],
@@ -44,24 +54,17 @@ def parse_args():
     return p.parse_args()
 
 
-def main():
-    args = parse_args()
-    info = init_distributed()
-    world = get_world_size()
-    rank = get_rank()
-    has_gpu = torch.cuda.is_available()
-    device = torch.device("cuda", info["local_rank"]) if has_gpu else torch.device("cpu")
-    if has_gpu:
-        torch.cuda.set_device(device)
-        torch.backends.cudnn.benchmark = True
-
+def run_model(args, model_name, device, world, rank, has_gpu):
+    """Time one model's train loop; returns (images_per_sec, ms_per_step,
+    global_batch) with elapsed maxed over ranks."""
     per_gpu_batch = args.batch_size
     if per_gpu_batch == 0:
-        per_gpu_batch = (256 if args.model == "resnet50" else 256) if has_gpu else 8
+        per_gpu_batch = 256 if has_gpu else 8
 
     torch.manual_seed(1234)
-    model = build_model(args.model, num_classes=1000).to(device)
-    channels_last = (args.model == "resnet50") and not args.no_channels_last and has_gpu
+    model = build_model(model_name, num_classes=1000).to(device)
+    channels_last = (model_name == "resnet50") and not args.no_channels_last \
+        and has_gpu
     if channels_last:
         model = model.to(memory_format=torch.channels_last)
     model = wrap_data_parallel(model, style=args.dp, bucket_cap_mb=args.bucket_mb) \
@@ -114,8 +117,6 @@ def main():
             with torch.cuda.graph(graph):
                 one_step(0, set_to_none=False)
         except Exception as e:  # capture unsupported for this model: run eager
-            import sys
-
             print(f"[bench] graph capture failed ({e}); eager path",
                   file=sys.stderr)
             graph = None
@@ -141,36 +142,75 @@ def main():
     elapsed = time.perf_counter() - t0
 
     # take the max elapsed over ranks
-    t = torch.tensor([elapsed], dtype=torch.float64, device=device if has_gpu else "cpu")
+    t = torch.tensor([elapsed], dtype=torch.float64,
+                     device=device if has_gpu else "cpu")
     if world > 1:
         import torch.distributed as dist
 
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
     elapsed_max = float(t.item())
 
+    del graph, model, optimizer, loader, batches, static_x, static_y
+    if has_gpu:
+        torch.cuda.empty_cache()
+
     global_batch = per_gpu_batch * world
     images_per_sec = global_batch * args.steps / elapsed_max
+    ms_per_step = elapsed_max / args.steps * 1000
+    return images_per_sec, ms_per_step, global_batch
+
+
+def main():
+    args = parse_args()
+    info = init_distributed()
+    world = get_world_size()
+    rank = get_rank()
+    has_gpu = torch.cuda.is_available()
+    device = torch.device("cuda", info["local_rank"]) if has_gpu else torch.device("cpu")
+    if has_gpu:
+        torch.cuda.set_device(device)
+        torch.backends.cudnn.benchmark = True
+    # rank-count evidence for the driver's SCALE record: every rank reports in
+    print(f"[bench] rank {rank}/{world} local_rank {info['local_rank']} "
+          f"device {device} backend "
+          f"{'nccl(RCCL)' if has_gpu and world > 1 else ('gloo' if world > 1 else 'none')}",
+          file=sys.stderr, flush=True)
+
+    model_names = ["vit_b16", "resnet50"] if args.model == "all" else [args.model]
+    results = {}
+    for name in model_names:
+        img_s, ms, global_batch = run_model(args, name, device, world, rank,
+                                            has_gpu)
+        results[name] = {"images_per_sec": round(img_s, 2),
+                         "ms_per_step": round(ms, 3),
+                         "global_batch": global_batch}
+
+    primary = "resnet50" if "resnet50" in results else model_names[0]
     if rank == 0:
+        r = results[primary]
+        config = {
+            "model": primary if len(results) == 1 else "resnet50 & vit_b16",
+            "global_batch": r["global_batch"],
+            "seq_len": None,
+            "image_size": 224,
+            "parallelism": f"dp{world}",
+        }
+        if len(results) > 1:
+            config["models"] = results
         result = {
             "metric": "images/sec",
-            "value": round(images_per_sec, 2),
+            "value": r["images_per_sec"],
             "unit": "images/s",
             "n_gpus": world,
             "steps": args.steps,
             "warmup": args.warmup,
-            "ms_per_step": round(elapsed_max / args.steps * 1000, 3),
+            "ms_per_step": r["ms_per_step"],
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16" if amp_enabled else "fp32",
+            "dtype": "bf16" if has_gpu else "fp32",
             "data": "synthetic",
-            "config": {
-                "model": args.model,
-                "global_batch": global_batch,
-                "seq_len": None,
-                "image_size": 224,
-                "parallelism": f"dp{world}",
-            },
+            "config": config,
         }
         print(json.dumps(result), flush=True)
     cleanup()

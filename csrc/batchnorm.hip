@@ -410,6 +410,76 @@ std::vector<torch::Tensor> batchnorm_fwd(torch::Tensor x, torch::Tensor weight,
   return {y, mean, rstd};
 }
 
+// Train-mode BN forward with the stats pass ALREADY DONE by the producing
+// kernel (conv1x1_fwd's fused sum/sumsq epilogue): finalize + apply only —
+// saves one full read pass over x vs batchnorm_fwd. sums layout matches
+// bn_stats: [2C] = {sum[c], sumsq[C+c]}.
+std::vector<torch::Tensor> batchnorm_fwd_from_sums(
+    torch::Tensor x, torch::Tensor weight, torch::Tensor bias,
+    torch::Tensor sums, c10::optional<torch::Tensor> running_mean,
+    c10::optional<torch::Tensor> running_var, double momentum, double eps,
+    bool relu) {
+  DLA_CHECK_CUDA(x);
+  TORCH_CHECK(x.dim() == 4, "batchnorm_fwd_from_sums expects a 4D tensor");
+  const bool nhwc = dla::is_nhwc(x);
+  TORCH_CHECK(nhwc || x.is_contiguous(),
+              "batchnorm_fwd_from_sums: x must be contiguous");
+  const int N = (int)x.size(0), C = (int)x.size(1);
+  const int64_t HW = x.size(2) * x.size(3);
+  TORCH_CHECK(sums.numel() == 2 * C, "sums must be [2C]");
+  auto opts_f = x.options().dtype(torch::kFloat);
+  auto mean = torch::empty({C}, opts_f);
+  auto rstd = torch::empty({C}, opts_f);
+  auto scale = torch::empty({C}, opts_f);
+  auto shift = torch::empty({C}, opts_f);
+  auto y = torch::empty_like(x);
+  auto w32 = weight.to(torch::kFloat);
+  auto b32 = bias.to(torch::kFloat);
+  const int64_t rows = (int64_t)N * HW;
+  DLA_DISPATCH_FLOAT_TYPES(x.scalar_type(), "batchnorm_fwd_from_sums", [&] {
+    hipLaunchKernelGGL(dla::bn_finalize_kernel, dim3((C + 255) / 256), dim3(256),
+                       0, dla::stream(), sums.data_ptr<float>(),
+                       w32.data_ptr<float>(), b32.data_ptr<float>(),
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       running_mean.has_value() ? running_mean->data_ptr<float>() : nullptr,
+                       running_var.has_value() ? running_var->data_ptr<float>() : nullptr,
+                       scale.data_ptr<float>(), shift.data_ptr<float>(), C,
+                       (float)rows, (float)eps, (float)momentum);
+    const int64_t n_total = x.numel();
+    constexpr int VMAX = 16 / (int)sizeof(dev_t);
+    auto launch = [&](auto vtag, auto rtag) {
+      constexpr int V = decltype(vtag)::value;
+      constexpr bool R = decltype(rtag)::value;
+      if (nhwc) {
+        const int lanes = C / V;
+        const int64_t used = dla::nhwc_used_threads(lanes, rows, 524288);
+        hipLaunchKernelGGL((dla::bn_apply_nhwc_kernel<dev_t, V, R>),
+                           dim3((int)((used + 255) / 256)), dim3(256), 0,
+                           dla::stream(), (const dev_t*)x.data_ptr(),
+                           scale.data_ptr<float>(), shift.data_ptr<float>(),
+                           (dev_t*)y.data_ptr(), C, rows, used);
+        return;
+      }
+      const int grid = dla::grid_1d((n_total + V - 1) / V, 256);
+      hipLaunchKernelGGL((dla::bn_apply_kernel<dev_t, V, R>), dim3(grid),
+                         dim3(256), 0, dla::stream(),
+                         (const dev_t*)x.data_ptr(), scale.data_ptr<float>(),
+                         shift.data_ptr<float>(), (dev_t*)y.data_ptr(), C, HW,
+                         n_total);
+    };
+    const bool vec_ok = nhwc ? (C % VMAX == 0) : (HW % VMAX == 0);
+    if (vec_ok) {
+      if (relu) launch(std::integral_constant<int, VMAX>{}, std::true_type{});
+      else launch(std::integral_constant<int, VMAX>{}, std::false_type{});
+    } else {
+      if (relu) launch(std::integral_constant<int, 1>{}, std::true_type{});
+      else launch(std::integral_constant<int, 1>{}, std::false_type{});
+    }
+  });
+  HIP_CHECK_ERR();
+  return {y, mean, rstd};
+}
+
 // Frozen/eval BN apply (+optional ReLU): per-channel scale/shift precomputed host-side.
 torch::Tensor bn_apply(torch::Tensor x, torch::Tensor scale, torch::Tensor shift,
                        bool relu) {

@@ -26,6 +26,19 @@ std::vector<torch::Tensor> batchnorm_bwd(torch::Tensor dy, torch::Tensor x,
                                          c10::optional<torch::Tensor> y,
                                          torch::Tensor weight, torch::Tensor mean,
                                          torch::Tensor rstd, bool relu);
+std::vector<torch::Tensor> batchnorm_fwd_from_sums(
+    torch::Tensor x, torch::Tensor weight, torch::Tensor bias,
+    torch::Tensor sums, c10::optional<torch::Tensor> running_mean,
+    c10::optional<torch::Tensor> running_var, double momentum, double eps,
+    bool relu);
+// conv1x1.hip
+std::vector<torch::Tensor> conv1x1_fwd(torch::Tensor a, torch::Tensor b,
+                                       c10::optional<torch::Tensor> bias,
+                                       c10::optional<torch::Tensor> scale,
+                                       c10::optional<torch::Tensor> shift,
+                                       c10::optional<torch::Tensor> residual,
+                                       bool relu, bool want_stats);
+torch::Tensor conv1x1_wgrad(torch::Tensor dy, torch::Tensor x);
 // softmax_ce.hip
 std::vector<torch::Tensor> softmax_ce_fwd(torch::Tensor logits,
                                           c10::optional<torch::Tensor> target,
@@ -35,7 +48,8 @@ torch::Tensor softmax_ce_bwd(torch::Tensor logits,
                              c10::optional<torch::Tensor> target,
                              c10::optional<torch::Tensor> soft_target,
                              torch::Tensor lse, double smoothing,
-                             int64_t ignore_index, double grad_scale);
+                             int64_t ignore_index, torch::Tensor grad_out,
+                             torch::Tensor stats);
 // focal.hip
 torch::Tensor focal_loss_fwd(torch::Tensor logits, torch::Tensor targets,
                              double alpha, double gamma);
@@ -93,8 +107,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("add_relu_fwd", &add_relu_fwd);
   m.def("relu_mask_bwd", &relu_mask_bwd);
   m.def("batchnorm_fwd", &batchnorm_fwd);
+  m.def("batchnorm_fwd_from_sums", &batchnorm_fwd_from_sums);
   m.def("bn_apply", &bn_apply);
   m.def("batchnorm_bwd", &batchnorm_bwd);
+  m.def("conv1x1_fwd", &conv1x1_fwd, py::arg("a"), py::arg("b"),
+        py::arg("bias") = py::none(), py::arg("scale") = py::none(),
+        py::arg("shift") = py::none(), py::arg("residual") = py::none(),
+        py::arg("relu") = false, py::arg("want_stats") = false);
+  m.def("conv1x1_wgrad", &conv1x1_wgrad);
   m.def("softmax_ce_fwd", &softmax_ce_fwd);
   m.def("softmax_ce_bwd", &softmax_ce_bwd);
   m.def("focal_loss_fwd", &focal_loss_fwd);

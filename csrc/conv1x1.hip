@@ -1,0 +1,420 @@
+// 1x1 convolution as an implicit-GEMM MFMA kernel (gfx950 / CDNA4), the
+// north-star conv path of BASELINE.json: NHWC conv1x1 IS the GEMM
+//   C[M,N] = A[M,K] @ B[N,K]^T,  M = batch*H*W, K = Cin, N = Cout,
+// with the weight consumed in its native [Cout][Cin] layout (B^T), so the
+// SAME kernel serves forward (A=x, B=W) and data-grad (A=dy, B=W^T).
+//
+// Replaces the reference's nn.Conv2d 1x1 call sites
+// (classification/resnet/models/networks.py:27-35 Bottleneck conv1/conv3 and
+// downsample) with a hand-written CDNA4 kernel:
+//  - 128xBN tile (BN 64/128), BK=64, 4 waves, mfma_f32_16x16x32_bf16
+//  - global->LDS staging via __builtin_amdgcn_global_load_lds (16B pieces),
+//    double-buffered, one barrier per K-step (guide §5 step-3 structure)
+//  - XOR-swizzled LDS image (guide T2 / G4: byte ^= (row&7)<<4 applied on the
+//    glds SOURCE address and again on the ds_read_b128 fragment reads)
+//  - fused epilogue: optional per-channel bias / scale+shift (eval-mode BN) /
+//    residual add / ReLU, and optional per-channel sum+sumsq accumulation so
+//    train-mode BatchNorm needs NO separate stats pass over the output
+//    (sums layout matches batchnorm.hip: sums[c], sums[C+c])
+//  - epilogue stores go through an LDS f32 transpose so global writes are
+//    row-major 8B/lane coalesced (fragment layout is column-per-lane)
+//
+// wgrad: dW[N][K] = sum_m dy[m][n] * x[m][k] — separate kernel contracting
+// over M with transposed LDS images (scatter ds_write staging, m-contiguous
+// fragment reads), fp32 atomicAdd accumulation over M-split blocks.
+#include "common.h"
+#include "vec.h"
+
+namespace dla {
+
+using bf16x8 = __attribute__((ext_vector_type(8))) short;
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+
+using bf16 = __hip_bfloat16;
+
+// global->LDS direct copy, 16 bytes per lane (guide §5: the 16B width is the
+// fast variant; dest must be wave-uniform base + lane*16).
+__device__ __forceinline__ void glds16(const bf16* gsrc, bf16* lds_dst) {
+  __builtin_amdgcn_global_load_lds(
+      (const __attribute__((address_space(1))) unsigned int*)gsrc,
+      (__attribute__((address_space(3))) unsigned int*)lds_dst, 16, 0, 0);
+}
+
+// ---------------------------------------------------------------------------
+// conv1x1 forward / dgrad kernel: C[M,N] = A[M,K] @ B[N,K]^T (bf16 in/out,
+// fp32 accumulate). K % 64 == 0, N % BN == 0.
+// ---------------------------------------------------------------------------
+template <int BN, bool STATS>
+__global__ __launch_bounds__(256) void conv1x1_nt_kernel(
+    const bf16* __restrict__ A, const bf16* __restrict__ B,
+    bf16* __restrict__ C, const bf16* __restrict__ residual,  // [M,N] | null
+    const float* __restrict__ bias,                           // [N] | null
+    const float* __restrict__ scale,                          // [N] | null
+    const float* __restrict__ shift,                          // [N] | null
+    float* __restrict__ sums,  // [2N] | null (pre-zeroed)
+    int64_t M, int K, int N, bool relu) {
+  constexpr int BM = 128, BK = 64;
+  constexpr int WAVES_M = (BN == 128) ? 2 : 4;
+  constexpr int WAVES_N = (BN == 128) ? 2 : 1;
+  constexpr int WM = BM / WAVES_M;  // 64 or 32
+  constexpr int WN = BN / WAVES_N;  // 64
+  constexpr int MFR = WM / 16, NFR = WN / 16;
+  constexpr int ROWB = BK * 2;  // LDS row bytes (128)
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int64_t m0 = (int64_t)blockIdx.x * BM;
+  const int n0 = blockIdx.y * BN;
+
+  extern __shared__ __attribute__((aligned(16))) char lds_raw[];
+  bf16* const lds16 = (bf16*)lds_raw;
+  auto a_lds = [&](int buf) { return lds16 + buf * (BM * BK); };
+  auto b_lds = [&](int buf) { return lds16 + 2 * BM * BK + buf * (BN * BK); };
+
+  const int nk = K / BK;
+
+  // stage one K-step of A[BM][BK] and B[BN][BK] into buffer `buf`.
+  // LDS image is row-major with the read-side XOR swizzle baked into the
+  // SOURCE address (rule 21: linear dest + inverse-swizzled source).
+  auto stage = [&](int buf, int kt) {
+    const int k0b = kt * BK * 2;  // byte offset into a row of A/B
+    // A: 16 pieces of 1KB (8 rows x 128B); 4 per wave
+#pragma unroll
+    for (int pp = 0; pp < 4; ++pp) {
+      const int p = wave + pp * 4;
+      const int row = p * 8 + (lane >> 3);
+      const int b = (lane & 7) * 16;
+      const int64_t rg = m0 + row < M ? m0 + row : M - 1;
+      const char* src = (const char*)A + rg * (int64_t)K * 2 + k0b +
+                        (b ^ ((row & 7) << 4));
+      glds16((const bf16*)src, a_lds(buf) + p * 512);
+    }
+    // B: BN/8 pieces
+#pragma unroll
+    for (int pp = 0; pp < BN / 32; ++pp) {
+      const int p = wave + pp * 4;
+      const int row = p * 8 + (lane >> 3);
+      const int b = (lane & 7) * 16;
+      const char* src = (const char*)B + (int64_t)(n0 + row) * K * 2 + k0b +
+                        (b ^ ((row & 7) << 4));
+      glds16((const bf16*)src, b_lds(buf) + p * 512);
+    }
+  };
+
+  const int wm_off = (wave / WAVES_N) * WM;
+  const int wn_off = (wave % WAVES_N) * WN;
+
+  f32x4 acc[MFR][NFR];
+#pragma unroll
+  for (int mi = 0; mi < MFR; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < NFR; ++ni) acc[mi][ni] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  stage(0, 0);
+  __syncthreads();  // drains the glds (vmcnt0 inside the barrier)
+  int cur = 0;
+  for (int kt = 0; kt < nk; ++kt) {
+    if (kt + 1 < nk) stage(cur ^ 1, kt + 1);
+    const char* ab = (const char*)a_lds(cur);
+    const char* bb = (const char*)b_lds(cur);
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      const int kbyte = (ks * 32 + (lane >> 4) * 8) * 2;
+      bf16x8 af[MFR], bfr[NFR];
+#pragma unroll
+      for (int mi = 0; mi < MFR; ++mi) {
+        const int row = wm_off + mi * 16 + (lane & 15);
+        af[mi] = *(const bf16x8*)(ab + row * ROWB + (kbyte ^ ((row & 7) << 4)));
+      }
+#pragma unroll
+      for (int ni = 0; ni < NFR; ++ni) {
+        const int row = wn_off + ni * 16 + (lane & 15);
+        bfr[ni] = *(const bf16x8*)(bb + row * ROWB + (kbyte ^ ((row & 7) << 4)));
+      }
+#pragma unroll
+      for (int mi = 0; mi < MFR; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < NFR; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  // ---- stats from registers (fp32-exact, before the bf16 rounding) -------
+  // fragment layout: lane's channel for n-tile ni is col = ni*16 + (lane&15);
+  // its 4*MFR values per ni are rows of that channel. Padded rows (gm >= M)
+  // must not contribute.
+  if (STATS) {
+#pragma unroll
+    for (int ni = 0; ni < NFR; ++ni) {
+      const int ng = n0 + wn_off + ni * 16 + (lane & 15);
+      const float b = bias != nullptr ? bias[ng] : 0.f;
+      float s = 0.f, q = 0.f;
+#pragma unroll
+      for (int mi = 0; mi < MFR; ++mi)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int64_t gm = m0 + wm_off + mi * 16 + (lane >> 4) * 4 + r;
+          if (gm < M) {
+            const float raw = acc[mi][ni][r] + b;
+            s += raw;
+            q += raw * raw;
+          }
+        }
+      // lanes l, l+16, l+32, l+48 hold the same channel (different rows)
+      s += __shfl_xor(s, 16, 64);
+      s += __shfl_xor(s, 32, 64);
+      q += __shfl_xor(q, 16, 64);
+      q += __shfl_xor(q, 32, 64);
+      if (lane < 16) {
+        atomicAdd(&sums[ng], s);
+        atomicAdd(&sums[N + ng], q);
+      }
+    }
+  }
+
+  // ---- epilogue: LDS bf16 transpose -> row-major 16B stores --------------
+  constexpr int EROW = WN + 8;  // +8 bf16 pad keeps b128 reads 16B-aligned
+  bf16* ep = (bf16*)lds_raw + wave * WM * EROW;
+  {
+    const float bcol[4] = {
+        bias != nullptr ? bias[n0 + wn_off + 0 * 16 + (lane & 15)] : 0.f,
+        bias != nullptr ? bias[n0 + wn_off + 1 * 16 + (lane & 15)] : 0.f,
+        NFR > 2 && bias != nullptr ? bias[n0 + wn_off + 2 * 16 + (lane & 15)] : 0.f,
+        NFR > 3 && bias != nullptr ? bias[n0 + wn_off + 3 * 16 + (lane & 15)] : 0.f};
+#pragma unroll
+    for (int mi = 0; mi < MFR; ++mi)
+#pragma unroll
+      for (int ni = 0; ni < NFR; ++ni)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int row = mi * 16 + (lane >> 4) * 4 + r;
+          const int col = ni * 16 + (lane & 15);
+          ep[row * EROW + col] = from_f32<bf16>(acc[mi][ni][r] + bcol[ni]);
+        }
+  }
+  __syncthreads();
+
+  // lane -> fixed 8-channel chunk (c0), rows strided by 8
+  const int c0 = (lane & 7) * 8;           // channel chunk in wave tile
+  const int ng0 = n0 + wn_off + c0;        // global channel of chunk start
+  const bool has_ss = scale != nullptr;
+  float sc[8], sh[8];
+  if (has_ss) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) { sc[j] = scale[ng0 + j]; sh[j] = shift[ng0 + j]; }
+  }
+#pragma unroll
+  for (int i = 0; i < WM / 8; ++i) {
+    const int r = (lane >> 3) + 8 * i;
+    const int64_t gm = m0 + wm_off + r;
+    if (gm < M) {
+      Vec<bf16, 8> v = vload<bf16, 8>(ep + r * EROW + c0);
+      Vec<bf16, 8> res;
+      if (residual != nullptr)
+        res = vload<bf16, 8>(residual + gm * N + ng0);
+      Vec<bf16, 8> out;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float y = to_f32(v.v[j]);
+        if (has_ss) y = y * sc[j] + sh[j];
+        if (residual != nullptr) y += to_f32(res.v[j]);
+        if (relu) y = fmaxf(y, 0.f);
+        out.v[j] = from_f32<bf16>(y);
+      }
+      vstore<bf16, 8>(C + gm * N + ng0, out);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// wgrad kernel: dW[N][K] += sum_m dy[m][n] * x[m][k] (fp32 atomics).
+// Both operands staged TRANSPOSED in LDS ([ch][m+pad], scatter b16 writes
+// with an m-XOR so the channel-group writes spread over banks), fragments
+// then read m-contiguous b128. 64x64 output tile, 4 waves (32x32 each).
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void conv1x1_wgrad_kernel(
+    const bf16* __restrict__ dy,  // [M,N]
+    const bf16* __restrict__ x,   // [M,K]
+    float* __restrict__ dW,       // [N,K] pre-zeroed
+    int64_t M, int K, int N, int64_t chunk) {
+  constexpr int KM = 64;   // m per step
+  constexpr int TR = 72;   // LDS row length (elements) for [64ch][KM] image
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int n0 = blockIdx.x * 64;  // co tile
+  const int k0 = blockIdx.y * 64;  // ci tile
+  const int64_t m_begin = (int64_t)blockIdx.z * chunk;
+  const int64_t m_end = m_begin + chunk < M ? m_begin + chunk : M;
+
+  extern __shared__ __attribute__((aligned(16))) char lds_raw[];
+  bf16* const lds16 = (bf16*)lds_raw;
+  auto dyt = [&](int buf) { return lds16 + buf * (64 * TR); };
+  auto xt = [&](int buf) { return lds16 + (2 + buf) * (64 * TR); };
+
+  // transposed scatter-stage of a [KM m][64 ch] global chunk into
+  // lds[ch][m ^ mswz(ch)] (mswz spreads the 8 channel-groups over banks)
+  auto stage_t = [&](bf16* lds, const bf16* g, int stride_elems, int ch0g,
+                     int64_t mt) {
+#pragma unroll
+    for (int it = 0; it < 2; ++it) {
+      const int idx = (int)threadIdx.x + it * 256;
+      const int m = idx >> 3;              // 0..63
+      const int ch = (idx & 7) * 8;        // 0..56
+      Vec<bf16, 8> v;
+      if (mt + m < m_end) {
+        v = vload<bf16, 8>(g + (mt + m) * (int64_t)stride_elems + ch0g + ch);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) v.v[j] = from_f32<bf16>(0.f);
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int c = ch + j;
+        lds[c * TR + (m ^ (((c >> 3) & 7) << 3))] = v.v[j];
+      }
+    }
+  };
+
+  f32x4 acc[2][2];
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni) acc[mi][ni] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int wco = (wave >> 1) * 32;
+  const int wci = (wave & 1) * 32;
+
+  stage_t(dyt(0), dy, N, n0, m_begin);
+  stage_t(xt(0), x, K, k0, m_begin);
+  __syncthreads();
+  int cur = 0;
+  for (int64_t mt = m_begin; mt < m_end; mt += KM) {
+    if (mt + KM < m_end) {
+      stage_t(dyt(cur ^ 1), dy, N, n0, mt + KM);
+      stage_t(xt(cur ^ 1), x, K, k0, mt + KM);
+    }
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      const int mm = ks * 32 + (lane >> 4) * 8;
+      bf16x8 af[2], bfr[2];
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi) {
+        const int c = wco + mi * 16 + (lane & 15);
+        af[mi] = *(const bf16x8*)(dyt(cur) + c * TR +
+                                  (mm ^ (((c >> 3) & 7) << 3)));
+      }
+#pragma unroll
+      for (int ni = 0; ni < 2; ++ni) {
+        const int c = wci + ni * 16 + (lane & 15);
+        bfr[ni] = *(const bf16x8*)(xt(cur) + c * TR +
+                                   (mm ^ (((c >> 3) & 7) << 3)));
+      }
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+
+#pragma unroll
+  for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+    for (int ni = 0; ni < 2; ++ni)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int co = n0 + wco + mi * 16 + (lane >> 4) * 4 + r;
+        const int ci = k0 + wci + ni * 16 + (lane & 15);
+        atomicAdd(&dW[(int64_t)co * K + ci], acc[mi][ni][r]);
+      }
+}
+
+}  // namespace dla
+
+// A [M,K] bf16, B [N,K] bf16 -> C = A @ B^T [M,N] bf16, with fused epilogue.
+// Returns {C, sums[2N] fp32} (sums undefined unless want_stats).
+std::vector<torch::Tensor> conv1x1_fwd(torch::Tensor a, torch::Tensor b,
+                                       c10::optional<torch::Tensor> bias,
+                                       c10::optional<torch::Tensor> scale,
+                                       c10::optional<torch::Tensor> shift,
+                                       c10::optional<torch::Tensor> residual,
+                                       bool relu, bool want_stats) {
+  DLA_CHECK_INPUT(a);
+  DLA_CHECK_INPUT(b);
+  TORCH_CHECK(a.scalar_type() == torch::kBFloat16 &&
+                  b.scalar_type() == torch::kBFloat16,
+              "conv1x1_fwd: bf16 only");
+  const int64_t M = a.size(0);
+  const int K = (int)a.size(1), N = (int)b.size(0);
+  TORCH_CHECK((int)b.size(1) == K, "conv1x1_fwd: K mismatch");
+  TORCH_CHECK(K % 64 == 0 && N % 64 == 0, "conv1x1_fwd: K,N must be %64");
+  auto C = torch::empty({M, N}, a.options());
+  auto opts_f = a.options().dtype(torch::kFloat);
+  auto sums = want_stats ? torch::zeros({2 * N}, opts_f)
+                         : torch::empty({0}, opts_f);
+  const dla::bf16* res_p =
+      residual.has_value() ? (const dla::bf16*)residual->data_ptr() : nullptr;
+  const float* bias_p = bias.has_value() ? bias->data_ptr<float>() : nullptr;
+  const float* scale_p = scale.has_value() ? scale->data_ptr<float>() : nullptr;
+  const float* shift_p = shift.has_value() ? shift->data_ptr<float>() : nullptr;
+  float* sums_p = want_stats ? sums.data_ptr<float>() : nullptr;
+
+  const int gx = (int)((M + 127) / 128);
+  auto launch = [&](auto bntag, auto stag) {
+    constexpr int BN = decltype(bntag)::value;
+    constexpr bool ST = decltype(stag)::value;
+    constexpr int WM = (BN == 128) ? 64 : 32;
+    const int lds_stage = (128 * 64 + BN * 64) * 2 * 2;
+    const int lds_ep = 4 * WM * (64 + 8) * 2;  // bf16 transpose scratch
+    const int lds = std::max(lds_stage, lds_ep);
+    hipLaunchKernelGGL((dla::conv1x1_nt_kernel<BN, ST>), dim3(gx, N / BN),
+                       dim3(256), lds, dla::stream(),
+                       (const dla::bf16*)a.data_ptr(),
+                       (const dla::bf16*)b.data_ptr(),
+                       (dla::bf16*)C.data_ptr(), res_p, bias_p, scale_p,
+                       shift_p, sums_p, M, K, N, relu);
+  };
+  if (N % 128 == 0) {
+    if (want_stats) launch(std::integral_constant<int, 128>{}, std::true_type{});
+    else launch(std::integral_constant<int, 128>{}, std::false_type{});
+  } else {
+    if (want_stats) launch(std::integral_constant<int, 64>{}, std::true_type{});
+    else launch(std::integral_constant<int, 64>{}, std::false_type{});
+  }
+  HIP_CHECK_ERR();
+  return {C, sums};
+}
+
+// dW[N,K] fp32 = dy[M,N]^T @ x[M,K]
+torch::Tensor conv1x1_wgrad(torch::Tensor dy, torch::Tensor x) {
+  DLA_CHECK_INPUT(dy);
+  DLA_CHECK_INPUT(x);
+  TORCH_CHECK(dy.scalar_type() == torch::kBFloat16 &&
+                  x.scalar_type() == torch::kBFloat16,
+              "conv1x1_wgrad: bf16 only");
+  const int64_t M = x.size(0);
+  const int K = (int)x.size(1), N = (int)dy.size(1);
+  TORCH_CHECK(dy.size(0) == M, "conv1x1_wgrad: M mismatch");
+  TORCH_CHECK(K % 64 == 0 && N % 64 == 0, "conv1x1_wgrad: K,N must be %64");
+  auto dW = torch::zeros({N, K}, x.options().dtype(torch::kFloat));
+  const int tiles = (N / 64) * (K / 64);
+  int splits = (int)std::min<int64_t>(
+      std::max<int64_t>(1, 1024 / tiles),
+      (M + 4095) / 4096);
+  const int64_t chunk0 = (M + splits - 1) / splits;
+  const int64_t chunk = ((chunk0 + 63) / 64) * 64;  // multiple of KM
+  splits = (int)((M + chunk - 1) / chunk);
+  const int lds = 4 * 64 * 72 * 2;
+  hipLaunchKernelGGL(
+      dla::conv1x1_wgrad_kernel, dim3(N / 64, K / 64, splits), dim3(256),
+      lds, dla::stream(), (const dla::bf16*)dy.data_ptr(),
+      (const dla::bf16*)x.data_ptr(), dW.data_ptr<float>(), M, K, N, chunk);
+  HIP_CHECK_ERR();
+  return dW;
+}

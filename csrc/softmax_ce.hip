@@ -2,6 +2,12 @@
 // Covers the reference's nn.CrossEntropyLoss / timm LabelSmoothingCrossEntropy /
 // SoftTargetCrossEntropy call sites (swin main.py:111-117, everywhere else).
 // Hard-label path (int64 targets) and soft-target path (B,C probabilities).
+//
+// The loss MEAN is folded into the forward kernel: each row's loss is
+// atomicAdd-ed into stats[0] and valid rows counted into stats[1], so the
+// wrapper does one division instead of a sum + mask + clamp chain, and the
+// backward kernel reads its grad scale from device memory (no host sync,
+// hipGraph-capturable).
 #include <cstdlib>
 
 #include "common.h"
@@ -9,22 +15,21 @@
 
 namespace dla {
 
-// The wave-per-row CE variants are gated behind DLA_CE_WAVE=1 until they get
-// hardware numerics validation (ROADMAP item; __expf/__logf are fast-math).
-inline bool ce_wave_enabled() {
-  static const bool v = []{
-    const char* e = std::getenv("DLA_CE_WAVE");
-    return e != nullptr && e[0] == '1';
-  }();
-  return v;
+// stats = {loss_sum, valid_count} (pre-zeroed by the wrapper)
+__device__ __forceinline__ void ce_accumulate(float* stats, float row_loss,
+                                              bool valid) {
+  if (valid) {
+    atomicAdd(stats + 0, row_loss);
+    atomicAdd(stats + 1, 1.f);
+  }
 }
 
-// one block per row; returns per-row loss and saves lse for backward
+// one block per row; accumulates loss into stats and saves lse for backward
 template <typename dev_t>
 __global__ void ce_fwd_kernel(const dev_t* __restrict__ logits,
                               const int64_t* __restrict__ target,
                               const dev_t* __restrict__ soft_target,
-                              float* __restrict__ loss, float* __restrict__ lse_out,
+                              float* __restrict__ stats, float* __restrict__ lse_out,
                               int B, int C, float smoothing, int64_t ignore_index) {
   __shared__ float smem[16];
   for (int row = blockIdx.x; row < B; row += gridDim.x) {
@@ -47,20 +52,20 @@ __global__ void ce_fwd_kernel(const dev_t* __restrict__ logits,
     if (soft_target != nullptr) {
       ssum = block_reduce_sum(ssum, smem);
       if (threadIdx.x == 0) {
-        loss[row] = lse - ssum;
+        ce_accumulate(stats, lse - ssum, true);
         lse_out[row] = lse;
       }
     } else {
       const int64_t t = target[row];
       if (smoothing > 0.f) ssum = block_reduce_sum(ssum, smem);
       if (threadIdx.x == 0) {
-        if (t == ignore_index) {
-          loss[row] = 0.f;
-        } else {
+        if (t != ignore_index) {
           const float zt = to_f32(lr[t]);
           // (1-eps)*(lse - z_t) + eps * (lse - mean(z))
-          loss[row] = (1.f - smoothing) * (lse - zt) +
-                      smoothing * (lse - ssum / C);
+          ce_accumulate(stats,
+                        (1.f - smoothing) * (lse - zt) +
+                            smoothing * (lse - ssum / C),
+                        true);
         }
         lse_out[row] = lse;
       }
@@ -68,7 +73,9 @@ __global__ void ce_fwd_kernel(const dev_t* __restrict__ logits,
   }
 }
 
-// dlogits = (softmax - q) * dloss_row ; q = one-hot smoothed or soft target
+// dlogits = (softmax - q) * grad_out/valid ; q = one-hot smoothed or soft
+// target. grad_out (upstream grad of the scalar mean) and stats live on
+// device so backward never syncs the host.
 template <typename dev_t>
 __global__ void ce_bwd_kernel(const dev_t* __restrict__ logits,
                               const int64_t* __restrict__ target,
@@ -76,7 +83,9 @@ __global__ void ce_bwd_kernel(const dev_t* __restrict__ logits,
                               const float* __restrict__ lse,
                               dev_t* __restrict__ dlogits, int B, int C,
                               float smoothing, int64_t ignore_index,
-                              float grad_scale) {
+                              const float* __restrict__ grad_out,
+                              const float* __restrict__ stats) {
+  const float grad_scale = grad_out[0] / fmaxf(stats[1], 1.f);
   for (int row = blockIdx.x; row < B; row += gridDim.x) {
     const dev_t* lr = logits + (int64_t)row * C;
     dev_t* dr = dlogits + (int64_t)row * C;
@@ -107,7 +116,7 @@ __global__ __launch_bounds__(256)
 void ce_fwd_wave_kernel(const dev_t* __restrict__ logits,
                         const int64_t* __restrict__ target,
                         const dev_t* __restrict__ soft_target,
-                        float* __restrict__ loss, float* __restrict__ lse_out,
+                        float* __restrict__ stats, float* __restrict__ lse_out,
                         int B, int C, float smoothing, int64_t ignore_index) {
   constexpr int MAX_PL = 8;
   const int lane = threadIdx.x & 63;
@@ -150,19 +159,19 @@ void ce_fwd_wave_kernel(const dev_t* __restrict__ logits,
     if (soft_target != nullptr) {
       ssum = wave_reduce_sum(ssum);
       if (lane == 0) {
-        loss[row] = lse - ssum;
+        ce_accumulate(stats, lse - ssum, true);
         lse_out[row] = lse;
       }
     } else {
       if (smoothing > 0.f) ssum = wave_reduce_sum(ssum);
       if (lane == 0) {
         const int64_t t = target[row];
-        if (t == ignore_index) {
-          loss[row] = 0.f;
-        } else {
+        if (t != ignore_index) {
           const float zt = to_f32(lr[t]);
-          loss[row] = (1.f - smoothing) * (lse - zt) +
-                      smoothing * (lse - ssum / C);
+          ce_accumulate(stats,
+                        (1.f - smoothing) * (lse - zt) +
+                            smoothing * (lse - ssum / C),
+                        true);
         }
         lse_out[row] = lse;
       }
@@ -178,8 +187,10 @@ void ce_bwd_wave_kernel(const dev_t* __restrict__ logits,
                         const float* __restrict__ lse,
                         dev_t* __restrict__ dlogits, int B, int C,
                         float smoothing, int64_t ignore_index,
-                        float grad_scale) {
+                        const float* __restrict__ grad_out,
+                        const float* __restrict__ stats) {
   constexpr int MAX_PL = 8;
+  const float grad_scale = grad_out[0] / fmaxf(stats[1], 1.f);
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int nwaves = blockDim.x >> 6;
@@ -220,19 +231,19 @@ void ce_bwd_wave_kernel(const dev_t* __restrict__ logits,
 
 }  // namespace dla
 
-// Returns {loss[B] fp32, lse[B] fp32}
+// Returns {stats[2] fp32 = {loss_sum, valid_count}, lse[B] fp32}
 std::vector<torch::Tensor> softmax_ce_fwd(torch::Tensor logits,
                                           c10::optional<torch::Tensor> target,
                                           c10::optional<torch::Tensor> soft_target,
                                           double smoothing, int64_t ignore_index) {
   DLA_CHECK_INPUT(logits);
   const int B = (int)logits.size(0), C = (int)logits.size(1);
-  auto loss = torch::empty({B}, logits.options().dtype(torch::kFloat));
+  auto stats = torch::zeros({2}, logits.options().dtype(torch::kFloat));
   auto lse = torch::empty({B}, logits.options().dtype(torch::kFloat));
   const int grid = (int)std::min<int64_t>(B, dla::kMaxGrid);
   DLA_DISPATCH_FLOAT_TYPES(logits.scalar_type(), "softmax_ce_fwd", [&] {
     constexpr int VMAX = 16 / (int)sizeof(dev_t);
-    if (dla::ce_wave_enabled() && C % VMAX == 0 && C <= 64 * VMAX * 8) {
+    if (C % VMAX == 0 && C <= 64 * VMAX * 8) {
       const int g = (int)std::min<int64_t>((B + 3) / 4, dla::kMaxGrid);
       hipLaunchKernelGGL(
           (dla::ce_fwd_wave_kernel<dev_t, VMAX>), dim3(g), dim3(256), 0,
@@ -240,7 +251,7 @@ std::vector<torch::Tensor> softmax_ce_fwd(torch::Tensor logits,
           target.has_value() ? target->data_ptr<int64_t>() : nullptr,
           soft_target.has_value() ? (const dev_t*)soft_target->data_ptr()
                                   : nullptr,
-          loss.data_ptr<float>(), lse.data_ptr<float>(), B, C,
+          stats.data_ptr<float>(), lse.data_ptr<float>(), B, C,
           (float)smoothing, ignore_index);
       return;
     }
@@ -249,25 +260,26 @@ std::vector<torch::Tensor> softmax_ce_fwd(torch::Tensor logits,
         (const dev_t*)logits.data_ptr(),
         target.has_value() ? target->data_ptr<int64_t>() : nullptr,
         soft_target.has_value() ? (const dev_t*)soft_target->data_ptr() : nullptr,
-        loss.data_ptr<float>(), lse.data_ptr<float>(), B, C, (float)smoothing,
+        stats.data_ptr<float>(), lse.data_ptr<float>(), B, C, (float)smoothing,
         ignore_index);
   });
   HIP_CHECK_ERR();
-  return {loss, lse};
+  return {stats, lse};
 }
 
 torch::Tensor softmax_ce_bwd(torch::Tensor logits,
                              c10::optional<torch::Tensor> target,
                              c10::optional<torch::Tensor> soft_target,
                              torch::Tensor lse, double smoothing,
-                             int64_t ignore_index, double grad_scale) {
+                             int64_t ignore_index, torch::Tensor grad_out,
+                             torch::Tensor stats) {
   DLA_CHECK_INPUT(logits);
   const int B = (int)logits.size(0), C = (int)logits.size(1);
   auto dlogits = torch::empty_like(logits);
   const int grid = (int)std::min<int64_t>(B, dla::kMaxGrid);
   DLA_DISPATCH_FLOAT_TYPES(logits.scalar_type(), "softmax_ce_bwd", [&] {
     constexpr int VMAX = 16 / (int)sizeof(dev_t);
-    if (dla::ce_wave_enabled() && C % VMAX == 0 && C <= 64 * VMAX * 8) {
+    if (C % VMAX == 0 && C <= 64 * VMAX * 8) {
       const int g = (int)std::min<int64_t>((B + 3) / 4, dla::kMaxGrid);
       hipLaunchKernelGGL(
           (dla::ce_bwd_wave_kernel<dev_t, VMAX>), dim3(g), dim3(256), 0,
@@ -276,7 +288,8 @@ torch::Tensor softmax_ce_bwd(torch::Tensor logits,
           soft_target.has_value() ? (const dev_t*)soft_target->data_ptr()
                                   : nullptr,
           lse.data_ptr<float>(), (dev_t*)dlogits.data_ptr(), B, C,
-          (float)smoothing, ignore_index, (float)grad_scale);
+          (float)smoothing, ignore_index, grad_out.data_ptr<float>(),
+          stats.data_ptr<float>());
       return;
     }
     hipLaunchKernelGGL(
@@ -285,7 +298,7 @@ torch::Tensor softmax_ce_bwd(torch::Tensor logits,
         target.has_value() ? target->data_ptr<int64_t>() : nullptr,
         soft_target.has_value() ? (const dev_t*)soft_target->data_ptr() : nullptr,
         lse.data_ptr<float>(), (dev_t*)dlogits.data_ptr(), B, C, (float)smoothing,
-        ignore_index, (float)grad_scale);
+        ignore_index, grad_out.data_ptr<float>(), stats.data_ptr<float>());
   });
   HIP_CHECK_ERR();
   return dlogits;

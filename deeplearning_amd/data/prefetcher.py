@@ -38,14 +38,28 @@ class DataPrefetcher:
             self.next_batch = self._to_device(batch)
             self.event.record(self.stream)
 
+    def _record_stream(self, x, stream):
+        # Every copied tensor was allocated on the side stream but is consumed
+        # on the compute stream; without record_stream the caching allocator
+        # may hand its memory to the next preload's H2D copy while compute is
+        # still reading it (mirror _to_device's recursion over nested batches).
+        if torch.is_tensor(x):
+            if x.is_cuda:
+                x.record_stream(stream)
+        elif isinstance(x, (list, tuple)):
+            for v in x:
+                self._record_stream(v, stream)
+        elif isinstance(x, dict):
+            for v in x.values():
+                self._record_stream(v, stream)
+
     def next(self):
         if self.next_batch is None:
             return None
         torch.cuda.current_stream().wait_event(self.event)
         batch = self.next_batch
         # keep the copied tensors alive until the compute stream used them
-        if torch.is_tensor(batch):
-            batch.record_stream(torch.cuda.current_stream())
+        self._record_stream(batch, torch.cuda.current_stream())
         self.preload()
         return batch
 

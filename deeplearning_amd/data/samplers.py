@@ -112,7 +112,15 @@ class GroupedBatchSampler(BatchSampler):
                 yield buf
 
     def __len__(self):
-        return (len(self.sampler) + self.batch_size - 1) // self.batch_size
+        # __iter__ yields full batches per group plus one padded flush batch
+        # for every group with a remainder, so count from the group-id
+        # distribution (len(sampler)//batch_size under-counts and schedulers
+        # sized from len(loader) would step past their total).
+        counts = defaultdict(int)
+        for idx in self.sampler:
+            counts[self.group_ids[idx]] += 1
+        return sum((n + self.batch_size - 1) // self.batch_size
+                   for n in counts.values())
 
 
 def create_aspect_ratio_groups(aspect_ratios, k=3):

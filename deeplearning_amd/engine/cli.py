@@ -263,14 +263,18 @@ def classification_train_main(args) -> dict:
             writer.add_scalar("train/loss", loss_m.avg, epoch)
             writer.add_scalar("val/acc1", val_acc, epoch)
             save_weights(model, weights_dir / f"model_{epoch}.pth")
-            save_checkpoint(weights_dir / f"ckpt_epoch_{epoch}.pth", model,
-                            optimizer, scheduler, epoch,
-                            max_accuracy=best_acc)
+            # update best BEFORE writing the checkpoint so a resumed run
+            # sees this epoch's accuracy as max_accuracy
             if val_acc >= best_acc:
                 best_acc = val_acc
                 save_weights(model, weights_dir / "best_model.pth")
+            save_checkpoint(weights_dir / f"ckpt_epoch_{epoch}.pth", model,
+                            optimizer, scheduler, epoch,
+                            max_accuracy=best_acc)
+        else:
+            best_acc = max(best_acc, val_acc)
         logger.info(f"epoch {epoch}: val acc1 {val_acc:.2f} "
-                    f"(best {max(best_acc, val_acc):.2f})")
+                    f"(best {best_acc:.2f})")
     if writer:
         writer.close()
     cleanup()

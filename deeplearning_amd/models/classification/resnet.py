@@ -13,7 +13,17 @@ import torch
 import torch.nn as nn
 
 from ...ops import BatchNorm2d, add_relu
+from ...ops.conv1x1 import conv_bn
 from ..registry import register_model
+
+
+def _downsample_fwd(downsample, x):
+    """Route a (conv1x1, bn) downsample Sequential through the fused
+    implicit-GEMM conv+BN path when it applies."""
+    if (isinstance(downsample, nn.Sequential) and len(downsample) == 2 and
+            isinstance(downsample[0], nn.Conv2d)):
+        return conv_bn(x, downsample[0], downsample[1])
+    return downsample(x)
 
 
 class _NormRelu(nn.Sequential):
@@ -86,11 +96,14 @@ class Bottleneck(nn.Module):
 
     def forward(self, x):
         identity = x
-        out = self.bn1(self.conv1(x))
+        # conv1/conv3 (1x1) ride the hand-written MFMA implicit-GEMM kernel
+        # with the BN stats pass fused into the conv epilogue (conv1x1.hip);
+        # conv2 (3x3) stays on the library conv.
+        out = conv_bn(x, self.conv1, self.bn1)
         out = self.bn2(self.conv2(out))
-        out = self.bn3(self.conv3(out))
+        out = conv_bn(out, self.conv3, self.bn3)
         if self.downsample is not None:
-            identity = self.downsample(x)
+            identity = _downsample_fwd(self.downsample, x)
         return add_relu(out, identity)
 
 

@@ -68,6 +68,10 @@ class GeneralizedRCNNTransform(nn.Module):
         max_h = int(math.ceil(max_h / s) * s)
         max_w = int(math.ceil(max_w / s) * s)
         batch = images[0].new_zeros(len(images), 3, max_h, max_w)
+        if images[0].is_cuda:
+            # NHWC layout for the HIP BN/conv path (the module docstring's
+            # stated MI355X layout); CPU stays NCHW-contiguous.
+            batch = batch.to(memory_format=torch.channels_last)
         for img, pad in zip(images, batch):
             pad[:, :img.shape[-2], :img.shape[-1]].copy_(img)
         return batch

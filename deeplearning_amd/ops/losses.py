@@ -16,32 +16,37 @@ from ._ext import ext, use_hip
 
 
 class _SoftmaxCEFn(torch.autograd.Function):
+    """Mean CE in 2 kernels fwd / 1 kernel bwd: the forward kernel folds the
+    loss sum + valid-row count into a stats[2] tensor (one division follows),
+    and the backward kernel reads its grad scale from device memory — no host
+    sync anywhere, so the loss is hipGraph-capturable."""
+
     @staticmethod
     def forward(ctx, logits, target, soft_target, smoothing, ignore_index):
         logits = logits.contiguous()
-        loss, lse = ext().softmax_ce_fwd(logits, target, soft_target, smoothing,
-                                         ignore_index)
-        if target is not None:
-            valid = (target != ignore_index).sum().clamp(min=1)
-        else:
-            valid = torch.tensor(float(logits.shape[0]), device=logits.device)
+        stats, lse = ext().softmax_ce_fwd(logits, target, soft_target,
+                                          smoothing, ignore_index)
         ctx.save_for_backward(logits, lse,
                               target if target is not None else torch.empty(0),
                               soft_target if soft_target is not None else torch.empty(0),
-                              valid)
+                              stats)
         ctx.smoothing = smoothing
         ctx.ignore_index = ignore_index
         ctx.has_target = target is not None
-        return loss.sum() / valid.to(loss.dtype)
+        return stats[0] / stats[1].clamp(min=1)
 
     @staticmethod
     def backward(ctx, dloss):
-        logits, lse, target, soft_target, valid = ctx.saved_tensors
-        grad_scale = float(dloss) / float(valid)
+        logits, lse, target, soft_target, stats = ctx.saved_tensors
+        grad_out = dloss.detach().reshape(1)
+        if grad_out.dtype != torch.float32:
+            grad_out = grad_out.float()
+        if not grad_out.is_cuda:  # backward of a CPU-held scalar grad
+            grad_out = grad_out.to(logits.device)
         dlogits = ext().softmax_ce_bwd(
             logits, target if ctx.has_target else None,
             soft_target if not ctx.has_target else None, lse, ctx.smoothing,
-            ctx.ignore_index, grad_scale)
+            ctx.ignore_index, grad_out.contiguous(), stats)
         return dlogits, None, None, None, None
 
 

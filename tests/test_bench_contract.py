@@ -32,7 +32,12 @@ def test_bench_single_process_cpu():
     assert REQUIRED_KEYS <= set(res)
     assert res["n_gpus"] == 1
     assert res["metric"] == "images/sec"
-    assert res["config"]["model"] == "resnet50"
+    # default invocation measures BOTH headline models (BASELINE.json names
+    # "ResNet-50 & ViT-B/16"); value is the ResNet-50 number
+    assert res["config"]["model"] == "resnet50 & vit_b16"
+    assert set(res["config"]["models"]) == {"resnet50", "vit_b16"}
+    assert res["config"]["models"]["vit_b16"]["images_per_sec"] > 0
+    assert res["value"] == res["config"]["models"]["resnet50"]["images_per_sec"]
     assert res["value"] > 0
 
 

@@ -1,0 +1,219 @@
+"""GPU parity tests for the implicit-GEMM MFMA conv1x1 kernel
+(csrc/conv1x1.hip) against plain PyTorch fp32 references.
+
+Covers: raw GEMM fwd (incl. M-edge), epilogue fusions (bias, scale/shift,
+residual, relu, channel stats), wgrad, the autograd Function end-to-end, the
+fused conv+BN module path vs the eager conv+BN chain, and a full Bottleneck
+A/B against the non-fused route.
+"""
+import os
+
+import pytest
+import torch
+
+from deeplearning_amd.ops._ext import ext
+
+pytestmark = pytest.mark.gpu
+
+
+def _rand2d(m, k, seed=0):
+    g = torch.Generator(device="cuda").manual_seed(seed)
+    return torch.randn(m, k, device="cuda", generator=g,
+                       dtype=torch.float32).to(torch.bfloat16)
+
+
+def _assert_close(got, ref, rtol=2e-2, atol_scale=None, msg=""):
+    got = got.float()
+    ref = ref.float()
+    denom = ref.abs().max().clamp(min=1.0)
+    err = (got - ref).abs().max() / denom
+    assert err < rtol, f"{msg}: rel err {err:.4g} (max|ref|={denom:.3g})"
+
+
+@pytest.mark.parametrize("m,k,n", [
+    (256, 64, 64),
+    (300, 64, 256),       # M not a multiple of the 128 tile
+    (1024, 128, 128),
+    (512, 2048, 512),     # deepest ResNet-50 1x1 shape class
+    (129, 192, 320),      # odd-ish N (uses BN=64 path), M edge
+])
+def test_gemm_fwd_parity(m, k, n):
+    a = _rand2d(m, k, 1)
+    b = _rand2d(n, k, 2)
+    y, _ = ext().conv1x1_fwd(a, b, None, None, None, None, False, False)
+    ref = a.float() @ b.float().t()
+    _assert_close(y, ref, msg=f"fwd {m}x{k}x{n}")
+
+
+def test_gemm_fwd_asymmetric_transpose_check():
+    # guide G9: asymmetric B catches a transposed C-write
+    a = torch.zeros(128, 64, device="cuda", dtype=torch.bfloat16)
+    for i in range(16):
+        a[i, i] = 1.0  # A = [I16 | 0] in the top block
+    b = _rand2d(128, 64, 3)
+    y, _ = ext().conv1x1_fwd(a, b, None, None, None, None, False, False)
+    ref = a.float() @ b.float().t()
+    _assert_close(y[:16], ref[:16], msg="identity-A asymmetric-B")
+    assert torch.allclose(y[:16].float(), b.float()[:, :16].t(), atol=1e-2)
+
+
+def test_epilogue_bias_scale_shift_relu_residual():
+    m, k, n = 640, 128, 256
+    a = _rand2d(m, k, 4)
+    b = _rand2d(n, k, 5)
+    bias = torch.randn(n, device="cuda")
+    scale = torch.randn(n, device="cuda")
+    shift = torch.randn(n, device="cuda")
+    res = _rand2d(m, n, 6)
+    y, _ = ext().conv1x1_fwd(a, b, bias, scale, shift, res, True, False)
+    raw = a.float() @ b.float().t() + bias
+    # kernel rounds raw to bf16 in the LDS transpose before the epilogue ops
+    raw = raw.to(torch.bfloat16).float()
+    ref = torch.relu(raw * scale + shift + res.float())
+    _assert_close(y, ref, rtol=4e-2, msg="fused epilogue")
+
+
+def test_stats_epilogue_matches_column_sums():
+    m, k, n = 999, 256, 128   # M edge: padded rows must not pollute stats
+    a = _rand2d(m, k, 7)
+    b = _rand2d(n, k, 8)
+    y, sums = ext().conv1x1_fwd(a, b, None, None, None, None, False, True)
+    ref = (a.float() @ b.float().t())
+    ref_sum = ref.sum(0)
+    ref_sq = (ref * ref).sum(0)
+    s, q = sums[:n], sums[n:]
+    assert torch.allclose(s, ref_sum, rtol=2e-2, atol=2e-2 * m ** 0.5), \
+        (s - ref_sum).abs().max()
+    assert torch.allclose(q, ref_sq, rtol=2e-2, atol=2e-2 * m), \
+        (q - ref_sq).abs().max()
+
+
+@pytest.mark.parametrize("m,k,n", [
+    (512, 64, 256),
+    (1000, 512, 128),     # M edge inside the KM=64 step
+    (12544, 2048, 512),
+])
+def test_wgrad_parity(m, k, n):
+    dy = _rand2d(m, n, 9)
+    x = _rand2d(m, k, 10)
+    dw = ext().conv1x1_wgrad(dy, x)
+    ref = dy.float().t() @ x.float()
+    err = (dw - ref).abs().max() / ref.abs().max().clamp(min=1)
+    assert err < 2e-2, f"wgrad rel err {err:.4g}"
+
+
+def test_autograd_function_end_to_end():
+    from deeplearning_amd.ops.conv1x1 import _Conv1x1Fn
+
+    m, k, n = 512, 128, 256
+    a = _rand2d(m, k, 11).requires_grad_(True)
+    w = torch.randn(n, k, device="cuda", dtype=torch.float32,
+                    requires_grad=True)
+    y = _Conv1x1Fn.apply(a, w, None, None, None, None, False, False)
+    loss = (y.float() ** 2).mean()
+    loss.backward()
+
+    a32 = a.detach().float().requires_grad_(True)
+    w32 = w.detach().clone().requires_grad_(True)
+    # reference follows the kernel's compute path: bf16 operands, fp32 acc
+    yr = a32 @ w32.t()
+    (yr ** 2).mean().backward()
+    _assert_close(y, yr.detach(), msg="fn fwd")
+    _assert_close(a.grad, a32.grad, rtol=4e-2, msg="dgrad")
+    _assert_close(w.grad, w32.grad, rtol=4e-2, msg="wgrad")
+    assert w.grad.dtype == torch.float32
+
+
+@pytest.mark.parametrize("stride", [1, 2])
+def test_conv1x1_bn_train_parity(stride):
+    """Fused conv+BN(+ReLU) vs eager F.conv2d + F.batch_norm in fp32."""
+    from deeplearning_amd.ops.batchnorm import BatchNorm2d
+    from deeplearning_amd.ops.conv1x1 import conv_bn
+
+    torch.manual_seed(0)
+    B, C, H, W, N = 8, 64, 28, 28, 128
+    conv = torch.nn.Conv2d(C, N, 1, stride=stride, bias=False).cuda()
+    bn = BatchNorm2d(N, relu=True).cuda()
+    x = torch.randn(B, C, H, W, device="cuda").to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last).requires_grad_(True)
+
+    y = conv_bn(x, conv, bn)
+    assert y.shape == (B, N, H // stride, W // stride)
+    loss = y.float().square().mean()
+    loss.backward()
+
+    # fp32 eager reference
+    xr = x.detach().float().requires_grad_(True)
+    convr = torch.nn.Conv2d(C, N, 1, stride=stride, bias=False).cuda()
+    convr.weight.data.copy_(conv.weight.data)
+    bnr = torch.nn.BatchNorm2d(N).cuda()
+    bnr.weight.data.copy_(bn.weight.data)
+    bnr.bias.data.copy_(bn.bias.data)
+    yr = torch.relu(bnr(convr(xr)))
+    yr.square().mean().backward()
+
+    _assert_close(y, yr.detach(), rtol=5e-2, msg="fused fwd")
+    _assert_close(x.grad, xr.grad, rtol=8e-2, msg="fused dx")
+    _assert_close(conv.weight.grad, convr.weight.grad, rtol=8e-2,
+                  msg="fused dw")
+    _assert_close(bn.weight.grad, bnr.weight.grad, rtol=8e-2, msg="bn dw")
+    _assert_close(bn.bias.grad, bnr.bias.grad, rtol=8e-2, msg="bn db")
+    # running stats updated like eager BN
+    _assert_close(bn.running_mean, bnr.running_mean, rtol=5e-2, msg="rmean")
+    _assert_close(bn.running_var, bnr.running_var, rtol=5e-2, msg="rvar")
+    assert int(bn.num_batches_tracked) == 1
+
+
+def test_conv1x1_bn_eval_single_kernel():
+    from deeplearning_amd.ops.batchnorm import BatchNorm2d
+    from deeplearning_amd.ops.conv1x1 import conv_bn
+
+    torch.manual_seed(1)
+    B, C, H, W, N = 4, 128, 14, 14, 256
+    conv = torch.nn.Conv2d(C, N, 1, bias=False).cuda()
+    bn = BatchNorm2d(N, relu=True).cuda().eval()
+    bn.running_mean.uniform_(-0.5, 0.5)
+    bn.running_var.uniform_(0.5, 1.5)
+    conv.eval()
+    x = torch.randn(B, C, H, W, device="cuda").to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    with torch.no_grad():
+        y = conv_bn(x, conv, bn)
+        xr = x.float()
+        yr = torch.relu(torch.nn.functional.batch_norm(
+            torch.nn.functional.conv2d(xr, conv.weight.float()),
+            bn.running_mean, bn.running_var, bn.weight.float(),
+            bn.bias.float(), False, 0.0, bn.eps))
+    _assert_close(y, yr, rtol=5e-2, msg="eval fused")
+
+
+def test_bottleneck_ab_vs_unfused():
+    """resnet50 layer3 block fwd+bwd: fused path vs DLA_NO_CONV1X1=1."""
+    from deeplearning_amd.models import build_model
+
+    def run(fused: bool):
+        os.environ["DLA_NO_CONV1X1"] = "0" if fused else "1"
+        try:
+            torch.manual_seed(7)
+            model = build_model("resnet50", num_classes=10).cuda() \
+                .to(memory_format=torch.channels_last)
+            block = model.layer3[0]
+            x = torch.randn(4, 512, 28, 28, device="cuda") \
+                .to(torch.bfloat16) \
+                .contiguous(memory_format=torch.channels_last) \
+                .requires_grad_(True)
+            y = block(x)
+            y.float().square().mean().backward()
+            return (y.detach().float(), x.grad.detach().float(),
+                    block.conv1.weight.grad.detach().float(),
+                    block.conv3.weight.grad.detach().float())
+        finally:
+            os.environ["DLA_NO_CONV1X1"] = "0"
+
+    yf, dxf, dw1f, dw3f = run(True)
+    ye, dxe, dw1e, dw3e = run(False)
+    for got, ref, name in [(yf, ye, "y"), (dxf, dxe, "dx"),
+                           (dw1f, dw1e, "dw1"), (dw3f, dw3e, "dw3")]:
+        denom = ref.abs().max().clamp(min=1e-3)
+        err = (got - ref).abs().max() / denom
+        assert err < 8e-2, f"bottleneck {name} rel err {err:.4g}"
