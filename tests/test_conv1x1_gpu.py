@@ -202,7 +202,9 @@ def test_bottleneck_ab_vs_unfused():
                 .to(torch.bfloat16) \
                 .contiguous(memory_format=torch.channels_last) \
                 .requires_grad_(True)
-            y = block(x)
+            # autocast so the unfused route casts conv weights like training
+            with torch.autocast("cuda", dtype=torch.bfloat16):
+                y = block(x)
             y.float().square().mean().backward()
             return (y.detach().float(), x.grad.detach().float(),
                     block.conv1.weight.grad.detach().float(),
