@@ -45,17 +45,18 @@ __device__ __forceinline__ void glds16(const bf16* gsrc, bf16* lds_dst) {
 // fp32 accumulate). K % 64 == 0, N % BN == 0.
 // ---------------------------------------------------------------------------
 template <int BN, bool STATS>
-__global__ __launch_bounds__(256) void conv1x1_nt_kernel(
+__global__ __launch_bounds__((BN == 256) ? 512 : 256) void conv1x1_nt_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
     bf16* __restrict__ C, const bf16* __restrict__ residual,  // [M,N] | null
     const float* __restrict__ bias,                           // [N] | null
     const float* __restrict__ scale,                          // [N] | null
     const float* __restrict__ shift,                          // [N] | null
-    float* __restrict__ sums,  // [2N] | null (pre-zeroed)
+    float* __restrict__ sums,  // [gridDim.x, 2N] per-block partials | null
     int64_t M, int K, int N, bool relu) {
   constexpr int BM = 128, BK = 64;
-  constexpr int WAVES_M = (BN == 128) ? 2 : 4;
-  constexpr int WAVES_N = (BN == 128) ? 2 : 1;
+  constexpr int NWAVES = (BN == 256) ? 8 : 4;
+  constexpr int WAVES_M = (BN == 64) ? 4 : 2;
+  constexpr int WAVES_N = NWAVES / WAVES_M;  // 4 / 2 / 1
   constexpr int WM = BM / WAVES_M;  // 64 or 32
   constexpr int WN = BN / WAVES_N;  // 64
   constexpr int MFR = WM / 16, NFR = WN / 16;
@@ -68,20 +69,20 @@ __global__ __launch_bounds__(256) void conv1x1_nt_kernel(
 
   extern __shared__ __attribute__((aligned(16))) char lds_raw[];
   bf16* const lds16 = (bf16*)lds_raw;
-  auto a_lds = [&](int buf) { return lds16 + buf * (BM * BK); };
-  auto b_lds = [&](int buf) { return lds16 + 2 * BM * BK + buf * (BN * BK); };
-
   const int nk = K / BK;
+  const int nbuf = nk > 1 ? 2 : 1;  // single staging buffer when one K-step
+  auto a_lds = [&](int buf) { return lds16 + buf * (BM * BK); };
+  auto b_lds = [&](int buf) { return lds16 + nbuf * BM * BK + buf * (BN * BK); };
 
   // stage one K-step of A[BM][BK] and B[BN][BK] into buffer `buf`.
   // LDS image is row-major with the read-side XOR swizzle baked into the
   // SOURCE address (rule 21: linear dest + inverse-swizzled source).
   auto stage = [&](int buf, int kt) {
     const int k0b = kt * BK * 2;  // byte offset into a row of A/B
-    // A: 16 pieces of 1KB (8 rows x 128B); 4 per wave
+    // A: 16 pieces of 1KB (8 rows x 128B)
 #pragma unroll
-    for (int pp = 0; pp < 4; ++pp) {
-      const int p = wave + pp * 4;
+    for (int pp = 0; pp < 16 / NWAVES; ++pp) {
+      const int p = wave + pp * NWAVES;
       const int row = p * 8 + (lane >> 3);
       const int b = (lane & 7) * 16;
       const int64_t rg = m0 + row < M ? m0 + row : M - 1;
@@ -91,8 +92,8 @@ __global__ __launch_bounds__(256) void conv1x1_nt_kernel(
     }
     // B: BN/8 pieces
 #pragma unroll
-    for (int pp = 0; pp < BN / 32; ++pp) {
-      const int p = wave + pp * 4;
+    for (int pp = 0; pp < (BN / 8) / NWAVES; ++pp) {
+      const int p = wave + pp * NWAVES;
       const int row = p * 8 + (lane >> 3);
       const int b = (lane & 7) * 16;
       const char* src = (const char*)B + (int64_t)(n0 + row) * K * 2 + k0b +
@@ -147,10 +148,15 @@ __global__ __launch_bounds__(256) void conv1x1_nt_kernel(
   // its 4*MFR values per ni are rows of that channel. Padded rows (gm >= M)
   // must not contribute.
   if (STATS) {
+    // per-block partial sums -> slab row sums[blockIdx.x * 2N + ...], NO
+    // global atomics (a 6k-deep atomic chain per channel measured 6x the
+    // whole GEMM); the [2C] reduction happens in one torch sum downstream.
+    float* sbuf = (float*)lds_raw;  // [WAVES_M][2*BN] (staging LDS is free)
+    const int wm = wave / WAVES_N;
 #pragma unroll
     for (int ni = 0; ni < NFR; ++ni) {
-      const int ng = n0 + wn_off + ni * 16 + (lane & 15);
-      const float b = bias != nullptr ? bias[ng] : 0.f;
+      const int c = wn_off + ni * 16 + (lane & 15);  // channel within block
+      const float b = bias != nullptr ? bias[n0 + c] : 0.f;
       float s = 0.f, q = 0.f;
 #pragma unroll
       for (int mi = 0; mi < MFR; ++mi)
@@ -169,10 +175,20 @@ __global__ __launch_bounds__(256) void conv1x1_nt_kernel(
       q += __shfl_xor(q, 16, 64);
       q += __shfl_xor(q, 32, 64);
       if (lane < 16) {
-        atomicAdd(&sums[ng], s);
-        atomicAdd(&sums[N + ng], q);
+        sbuf[wm * (2 * BN) + c] = s;
+        sbuf[wm * (2 * BN) + BN + c] = q;
       }
     }
+    __syncthreads();
+    float* slab = sums + (int64_t)blockIdx.x * (2 * N);
+    for (int t = threadIdx.x; t < 2 * BN; t += NWAVES * 64) {
+      float tot = 0.f;
+#pragma unroll
+      for (int wmi = 0; wmi < WAVES_M; ++wmi) tot += sbuf[wmi * (2 * BN) + t];
+      const int c = t < BN ? t : t - BN;
+      slab[(t < BN ? 0 : N) + n0 + c] = tot;
+    }
+    __syncthreads();  // sbuf is about to be reused by the store transpose
   }
 
   // ---- epilogue: LDS bf16 transpose -> row-major 16B stores --------------
@@ -356,7 +372,10 @@ std::vector<torch::Tensor> conv1x1_fwd(torch::Tensor a, torch::Tensor b,
   TORCH_CHECK(K % 64 == 0 && N % 64 == 0, "conv1x1_fwd: K,N must be %64");
   auto C = torch::empty({M, N}, a.options());
   auto opts_f = a.options().dtype(torch::kFloat);
-  auto sums = want_stats ? torch::zeros({2 * N}, opts_f)
+  const int gx = (int)((M + 127) / 128);
+  // per-block stats partials (fully written by the kernel; reduced to [2N]
+  // by the caller with one torch sum — no global atomics)
+  auto sums = want_stats ? torch::empty({gx, 2 * N}, opts_f)
                          : torch::empty({0}, opts_f);
   const dla::bf16* res_p =
       residual.has_value() ? (const dla::bf16*)residual->data_ptr() : nullptr;
@@ -365,28 +384,41 @@ std::vector<torch::Tensor> conv1x1_fwd(torch::Tensor a, torch::Tensor b,
   const float* shift_p = shift.has_value() ? shift->data_ptr<float>() : nullptr;
   float* sums_p = want_stats ? sums.data_ptr<float>() : nullptr;
 
-  const int gx = (int)((M + 127) / 128);
+  const int nk = K / 64;
   auto launch = [&](auto bntag, auto stag) {
     constexpr int BN = decltype(bntag)::value;
     constexpr bool ST = decltype(stag)::value;
-    constexpr int WM = (BN == 128) ? 64 : 32;
-    const int lds_stage = (128 * 64 + BN * 64) * 2 * 2;
-    const int lds_ep = 4 * WM * (64 + 8) * 2;  // bf16 transpose scratch
-    const int lds = std::max(lds_stage, lds_ep);
+    constexpr int NW = (BN == 256) ? 8 : 4;
+    constexpr int WM = (BN == 64) ? 32 : 64;
+    // single staging buffer suffices when there is only one K-step
+    const int nbuf = nk > 1 ? 2 : 1;
+    const int lds_stage = (128 * 64 + BN * 64) * 2 * nbuf;
+    const int lds_ep = NW * WM * (64 + 8) * 2;  // bf16 transpose scratch
+    const int lds_st = ST ? ((BN == 64) ? 4 : 2) * 2 * BN * 4 : 0;
+    const int lds = std::max(std::max(lds_stage, lds_ep), lds_st);
+    if (lds > 65536) {
+      static bool done[2] = {false, false};
+      if (!done[ST]) {
+        (void)hipFuncSetAttribute(
+            (const void*)&dla::conv1x1_nt_kernel<BN, ST>,
+            hipFuncAttributeMaxDynamicSharedMemorySize, 163840);
+        done[ST] = true;
+      }
+    }
     hipLaunchKernelGGL((dla::conv1x1_nt_kernel<BN, ST>), dim3(gx, N / BN),
-                       dim3(256), lds, dla::stream(),
+                       dim3(NW * 64), lds, dla::stream(),
                        (const dla::bf16*)a.data_ptr(),
                        (const dla::bf16*)b.data_ptr(),
                        (dla::bf16*)C.data_ptr(), res_p, bias_p, scale_p,
                        shift_p, sums_p, M, K, N, relu);
   };
-  if (N % 128 == 0) {
-    if (want_stats) launch(std::integral_constant<int, 128>{}, std::true_type{});
-    else launch(std::integral_constant<int, 128>{}, std::false_type{});
-  } else {
-    if (want_stats) launch(std::integral_constant<int, 64>{}, std::true_type{});
-    else launch(std::integral_constant<int, 64>{}, std::false_type{});
-  }
+  auto pick = [&](auto bntag) {
+    if (want_stats) launch(bntag, std::true_type{});
+    else launch(bntag, std::false_type{});
+  };
+  if (N % 256 == 0) pick(std::integral_constant<int, 256>{});
+  else if (N % 128 == 0) pick(std::integral_constant<int, 128>{});
+  else pick(std::integral_constant<int, 64>{});
   HIP_CHECK_ERR();
   return {C, sums};
 }

@@ -179,7 +179,9 @@ def conv1x1_bn(x: torch.Tensor, conv: nn.Conv2d, bn) -> torch.Tensor:
     if bn.num_batches_tracked is not None:
         bn.num_batches_tracked.add_(1)
     momentum = bn._eaf() if hasattr(bn, "_eaf") else (bn.momentum or 0.0)
-    y2d, sums = _Conv1x1Fn.apply(x2d, wb, bias, None, None, None, False, True)
+    y2d, partials = _Conv1x1Fn.apply(x2d, wb, bias, None, None, None, False,
+                                     True)
+    sums = partials.sum(dim=0)  # [n_blocks, 2C] slab -> [2C]
     y_raw = _unflatten_nhwc(y2d, B, H, W)
     return _BNFromStatsFn.apply(y_raw, bn.weight, bn.bias, sums,
                                 bn.running_mean, bn.running_var, momentum,

@@ -77,7 +77,9 @@ def test_stats_epilogue_matches_column_sums():
     m, k, n = 999, 256, 128   # M edge: padded rows must not pollute stats
     a = _rand2d(m, k, 7)
     b = _rand2d(n, k, 8)
-    y, sums = ext().conv1x1_fwd(a, b, None, None, None, None, False, True)
+    y, partials = ext().conv1x1_fwd(a, b, None, None, None, None, False, True)
+    assert partials.shape == ((m + 127) // 128, 2 * n)
+    sums = partials.sum(0)
     ref = (a.float() @ b.float().t())
     ref_sum = ref.sum(0)
     ref_sq = (ref * ref).sum(0)
