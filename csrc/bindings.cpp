@@ -39,6 +39,10 @@ std::vector<torch::Tensor> conv1x1_fwd(torch::Tensor a, torch::Tensor b,
                                        c10::optional<torch::Tensor> residual,
                                        bool relu, bool want_stats);
 torch::Tensor conv1x1_wgrad(torch::Tensor dy, torch::Tensor x);
+// reorder.hip
+torch::Tensor transpose2d(torch::Tensor src);
+torch::Tensor stride2_gather(torch::Tensor x);
+torch::Tensor stride2_scatter(torch::Tensor dy, int64_t IH, int64_t IW);
 // softmax_ce.hip
 std::vector<torch::Tensor> softmax_ce_fwd(torch::Tensor logits,
                                           c10::optional<torch::Tensor> target,
@@ -115,6 +119,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("shift") = py::none(), py::arg("residual") = py::none(),
         py::arg("relu") = false, py::arg("want_stats") = false);
   m.def("conv1x1_wgrad", &conv1x1_wgrad);
+  m.def("transpose2d", &transpose2d);
+  m.def("stride2_gather", &stride2_gather);
+  m.def("stride2_scatter", &stride2_scatter);
   m.def("softmax_ce_fwd", &softmax_ce_fwd);
   m.def("softmax_ce_bwd", &softmax_ce_bwd);
   m.def("focal_loss_fwd", &focal_loss_fwd);

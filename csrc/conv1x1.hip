@@ -70,9 +70,15 @@ __global__ __launch_bounds__((BN == 256) ? 512 : 256) void conv1x1_nt_kernel(
   extern __shared__ __attribute__((aligned(16))) char lds_raw[];
   bf16* const lds16 = (bf16*)lds_raw;
   const int nk = K / BK;
+  // one K-step: A fragments load straight from global (whole rows are
+  // consumed, L2-line-efficient) and only B goes through LDS — saves the
+  // A round-trip and one barrier per tile
+  const bool direct_a = (nk == 1);
   const int nbuf = nk > 1 ? 2 : 1;  // single staging buffer when one K-step
   auto a_lds = [&](int buf) { return lds16 + buf * (BM * BK); };
-  auto b_lds = [&](int buf) { return lds16 + nbuf * BM * BK + buf * (BN * BK); };
+  auto b_lds = [&](int buf) {
+    return (direct_a ? lds16 : lds16 + nbuf * BM * BK) + buf * (BN * BK);
+  };
 
   // stage one K-step of A[BM][BK] and B[BN][BK] into buffer `buf`.
   // LDS image is row-major with the read-side XOR swizzle baked into the
@@ -80,15 +86,17 @@ __global__ __launch_bounds__((BN == 256) ? 512 : 256) void conv1x1_nt_kernel(
   auto stage = [&](int buf, int kt) {
     const int k0b = kt * BK * 2;  // byte offset into a row of A/B
     // A: 16 pieces of 1KB (8 rows x 128B)
+    if (!direct_a) {
 #pragma unroll
-    for (int pp = 0; pp < 16 / NWAVES; ++pp) {
-      const int p = wave + pp * NWAVES;
-      const int row = p * 8 + (lane >> 3);
-      const int b = (lane & 7) * 16;
-      const int64_t rg = m0 + row < M ? m0 + row : M - 1;
-      const char* src = (const char*)A + rg * (int64_t)K * 2 + k0b +
-                        (b ^ ((row & 7) << 4));
-      glds16((const bf16*)src, a_lds(buf) + p * 512);
+      for (int pp = 0; pp < 16 / NWAVES; ++pp) {
+        const int p = wave + pp * NWAVES;
+        const int row = p * 8 + (lane >> 3);
+        const int b = (lane & 7) * 16;
+        const int64_t rg = m0 + row < M ? m0 + row : M - 1;
+        const char* src = (const char*)A + rg * (int64_t)K * 2 + k0b +
+                          (b ^ ((row & 7) << 4));
+        glds16((const bf16*)src, a_lds(buf) + p * 512);
+      }
     }
     // B: BN/8 pieces
 #pragma unroll
@@ -125,7 +133,14 @@ __global__ __launch_bounds__((BN == 256) ? 512 : 256) void conv1x1_nt_kernel(
 #pragma unroll
       for (int mi = 0; mi < MFR; ++mi) {
         const int row = wm_off + mi * 16 + (lane & 15);
-        af[mi] = *(const bf16x8*)(ab + row * ROWB + (kbyte ^ ((row & 7) << 4)));
+        if (direct_a) {
+          const int64_t rg = m0 + row < M ? m0 + row : M - 1;
+          af[mi] = *(const bf16x8*)((const char*)A + rg * (int64_t)K * 2 +
+                                    kbyte);
+        } else {
+          af[mi] =
+              *(const bf16x8*)(ab + row * ROWB + (kbyte ^ ((row & 7) << 4)));
+        }
       }
 #pragma unroll
       for (int ni = 0; ni < NFR; ++ni) {
@@ -251,34 +266,42 @@ __global__ __launch_bounds__((BN == 256) ? 512 : 256) void conv1x1_nt_kernel(
 // with an m-XOR so the channel-group writes spread over banks), fragments
 // then read m-contiguous b128. 64x64 output tile, 4 waves (32x32 each).
 // ---------------------------------------------------------------------------
-__global__ __launch_bounds__(256) void conv1x1_wgrad_kernel(
+template <int BCO, int BCI>
+__global__ __launch_bounds__((BCO == 128) ? 512 : 256) void conv1x1_wgrad_kernel(
     const bf16* __restrict__ dy,  // [M,N]
     const bf16* __restrict__ x,   // [M,K]
     float* __restrict__ dW,       // [N,K] pre-zeroed
     int64_t M, int K, int N, int64_t chunk) {
   constexpr int KM = 64;   // m per step
-  constexpr int TR = 72;   // LDS row length (elements) for [64ch][KM] image
+  constexpr int TR = 72;   // LDS row length (elements) for [ch][KM] image
+  constexpr int NW = (BCO == 128) ? 8 : 4;
+  constexpr int WGC = 2;               // wave grid over co
+  constexpr int WGI = NW / WGC;        // wave grid over ci (4 or 2)
+  constexpr int WCO = BCO / WGC;       // 64 or 32
+  constexpr int WCI = BCI / WGI;       // 32
+  constexpr int FCO = WCO / 16, FCI = WCI / 16;
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  const int n0 = blockIdx.x * 64;  // co tile
-  const int k0 = blockIdx.y * 64;  // ci tile
+  const int n0 = blockIdx.x * BCO;  // co tile
+  const int k0 = blockIdx.y * BCI;  // ci tile
   const int64_t m_begin = (int64_t)blockIdx.z * chunk;
   const int64_t m_end = m_begin + chunk < M ? m_begin + chunk : M;
 
   extern __shared__ __attribute__((aligned(16))) char lds_raw[];
   bf16* const lds16 = (bf16*)lds_raw;
-  auto dyt = [&](int buf) { return lds16 + buf * (64 * TR); };
-  auto xt = [&](int buf) { return lds16 + (2 + buf) * (64 * TR); };
+  auto dyt = [&](int buf) { return lds16 + buf * (BCO * TR); };
+  auto xt = [&](int buf) { return lds16 + 2 * BCO * TR + buf * (BCI * TR); };
 
-  // transposed scatter-stage of a [KM m][64 ch] global chunk into
+  // transposed scatter-stage of a [KM m][BCH ch] global chunk into
   // lds[ch][m ^ mswz(ch)] (mswz spreads the 8 channel-groups over banks)
   auto stage_t = [&](bf16* lds, const bf16* g, int stride_elems, int ch0g,
-                     int64_t mt) {
+                     int64_t mt, auto bch_tag) {
+    constexpr int BCH = decltype(bch_tag)::value;
 #pragma unroll
-    for (int it = 0; it < 2; ++it) {
-      const int idx = (int)threadIdx.x + it * 256;
-      const int m = idx >> 3;              // 0..63
-      const int ch = (idx & 7) * 8;        // 0..56
+    for (int it = 0; it < (KM * BCH / 8) / (NW * 64); ++it) {
+      const int idx = (int)threadIdx.x + it * (NW * 64);
+      const int m = idx / (BCH / 8);            // 0..63
+      const int ch = (idx % (BCH / 8)) * 8;
       Vec<bf16, 8> v;
       if (mt + m < m_end) {
         v = vload<bf16, 8>(g + (mt + m) * (int64_t)stride_elems + ch0g + ch);
@@ -294,44 +317,46 @@ __global__ __launch_bounds__(256) void conv1x1_wgrad_kernel(
     }
   };
 
-  f32x4 acc[2][2];
+  f32x4 acc[FCO][FCI];
 #pragma unroll
-  for (int mi = 0; mi < 2; ++mi)
+  for (int mi = 0; mi < FCO; ++mi)
 #pragma unroll
-    for (int ni = 0; ni < 2; ++ni) acc[mi][ni] = f32x4{0.f, 0.f, 0.f, 0.f};
+    for (int ni = 0; ni < FCI; ++ni) acc[mi][ni] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  const int wco = (wave >> 1) * 32;
-  const int wci = (wave & 1) * 32;
+  const int wco = (wave / WGI) * WCO;
+  const int wci = (wave % WGI) * WCI;
 
-  stage_t(dyt(0), dy, N, n0, m_begin);
-  stage_t(xt(0), x, K, k0, m_begin);
+  stage_t(dyt(0), dy, N, n0, m_begin, std::integral_constant<int, BCO>{});
+  stage_t(xt(0), x, K, k0, m_begin, std::integral_constant<int, BCI>{});
   __syncthreads();
   int cur = 0;
   for (int64_t mt = m_begin; mt < m_end; mt += KM) {
     if (mt + KM < m_end) {
-      stage_t(dyt(cur ^ 1), dy, N, n0, mt + KM);
-      stage_t(xt(cur ^ 1), x, K, k0, mt + KM);
+      stage_t(dyt(cur ^ 1), dy, N, n0, mt + KM,
+              std::integral_constant<int, BCO>{});
+      stage_t(xt(cur ^ 1), x, K, k0, mt + KM,
+              std::integral_constant<int, BCI>{});
     }
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
       const int mm = ks * 32 + (lane >> 4) * 8;
-      bf16x8 af[2], bfr[2];
+      bf16x8 af[FCO], bfr[FCI];
 #pragma unroll
-      for (int mi = 0; mi < 2; ++mi) {
+      for (int mi = 0; mi < FCO; ++mi) {
         const int c = wco + mi * 16 + (lane & 15);
         af[mi] = *(const bf16x8*)(dyt(cur) + c * TR +
                                   (mm ^ (((c >> 3) & 7) << 3)));
       }
 #pragma unroll
-      for (int ni = 0; ni < 2; ++ni) {
+      for (int ni = 0; ni < FCI; ++ni) {
         const int c = wci + ni * 16 + (lane & 15);
         bfr[ni] = *(const bf16x8*)(xt(cur) + c * TR +
                                    (mm ^ (((c >> 3) & 7) << 3)));
       }
 #pragma unroll
-      for (int mi = 0; mi < 2; ++mi)
+      for (int mi = 0; mi < FCO; ++mi)
 #pragma unroll
-        for (int ni = 0; ni < 2; ++ni)
+        for (int ni = 0; ni < FCI; ++ni)
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
     }
@@ -340,9 +365,9 @@ __global__ __launch_bounds__(256) void conv1x1_wgrad_kernel(
   }
 
 #pragma unroll
-  for (int mi = 0; mi < 2; ++mi)
+  for (int mi = 0; mi < FCO; ++mi)
 #pragma unroll
-    for (int ni = 0; ni < 2; ++ni)
+    for (int ni = 0; ni < FCI; ++ni)
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int co = n0 + wco + mi * 16 + (lane >> 4) * 4 + r;
@@ -390,9 +415,11 @@ std::vector<torch::Tensor> conv1x1_fwd(torch::Tensor a, torch::Tensor b,
     constexpr bool ST = decltype(stag)::value;
     constexpr int NW = (BN == 256) ? 8 : 4;
     constexpr int WM = (BN == 64) ? 32 : 64;
-    // single staging buffer suffices when there is only one K-step
+    // single staging buffer suffices when there is only one K-step; with one
+    // K-step A skips LDS entirely (direct-global fragments)
     const int nbuf = nk > 1 ? 2 : 1;
-    const int lds_stage = (128 * 64 + BN * 64) * 2 * nbuf;
+    const int lds_stage = nk == 1 ? BN * 64 * 2
+                                  : (128 * 64 + BN * 64) * 2 * nbuf;
     const int lds_ep = NW * WM * (64 + 8) * 2;  // bf16 transpose scratch
     const int lds_st = ST ? ((BN == 64) ? 4 : 2) * 2 * BN * 4 : 0;
     const int lds = std::max(std::max(lds_stage, lds_ep), lds_st);
@@ -435,18 +462,34 @@ torch::Tensor conv1x1_wgrad(torch::Tensor dy, torch::Tensor x) {
   TORCH_CHECK(dy.size(0) == M, "conv1x1_wgrad: M mismatch");
   TORCH_CHECK(K % 64 == 0 && N % 64 == 0, "conv1x1_wgrad: K,N must be %64");
   auto dW = torch::zeros({N, K}, x.options().dtype(torch::kFloat));
-  const int tiles = (N / 64) * (K / 64);
-  int splits = (int)std::min<int64_t>(
-      std::max<int64_t>(1, 1024 / tiles),
-      (M + 4095) / 4096);
-  const int64_t chunk0 = (M + splits - 1) / splits;
-  const int64_t chunk = ((chunk0 + 63) / 64) * 64;  // multiple of KM
-  splits = (int)((M + chunk - 1) / chunk);
-  const int lds = 4 * 64 * 72 * 2;
-  hipLaunchKernelGGL(
-      dla::conv1x1_wgrad_kernel, dim3(N / 64, K / 64, splits), dim3(256),
-      lds, dla::stream(), (const dla::bf16*)dy.data_ptr(),
-      (const dla::bf16*)x.data_ptr(), dW.data_ptr<float>(), M, K, N, chunk);
+  auto launch = [&](auto tag) {
+    constexpr int BT = decltype(tag)::value;  // BCO == BCI == BT
+    const int tiles = (N / BT) * (K / BT);
+    int splits = (int)std::min<int64_t>(
+        std::max<int64_t>(1, 1024 / tiles),
+        (M + 4095) / 4096);
+    const int64_t chunk0 = (M + splits - 1) / splits;
+    const int64_t chunk = ((chunk0 + 63) / 64) * 64;  // multiple of KM
+    splits = (int)((M + chunk - 1) / chunk);
+    const int lds = 2 * (2 * BT) * 72 * 2;
+    if (lds > 65536) {
+      static bool done = false;
+      if (!done) {
+        (void)hipFuncSetAttribute(
+            (const void*)&dla::conv1x1_wgrad_kernel<BT, BT>,
+            hipFuncAttributeMaxDynamicSharedMemorySize, 163840);
+        done = true;
+      }
+    }
+    hipLaunchKernelGGL((dla::conv1x1_wgrad_kernel<BT, BT>),
+                       dim3(N / BT, K / BT, splits),
+                       dim3(BT == 128 ? 512 : 256), lds, dla::stream(),
+                       (const dla::bf16*)dy.data_ptr(),
+                       (const dla::bf16*)x.data_ptr(), dW.data_ptr<float>(),
+                       M, K, N, chunk);
+  };
+  if (N % 128 == 0 && K % 128 == 0) launch(std::integral_constant<int, 128>{});
+  else launch(std::integral_constant<int, 64>{});
   HIP_CHECK_ERR();
   return dW;
 }

@@ -64,7 +64,8 @@ class _Conv1x1Fn(torch.autograd.Function):
             dy2d = dy2d.to(torch.bfloat16)
         dx = dw = None
         if ctx.needs_input_grad[0]:
-            wt = wb.t().contiguous()  # [K,N] bf16
+            # LDS-tiled transpose kernel (torch's strided copy is ~15x slower)
+            wt = ext().transpose2d(wb)  # [K,N] bf16
             dx, _ = ext().conv1x1_fwd(dy2d, wt, None, None, None, None,
                                       False, False)
             if dx.dtype != ctx.x_dtype:
@@ -99,6 +100,23 @@ class _BNFromStatsFn(torch.autograd.Function):
         return dx, dw, db, None, None, None, None, None, None
 
 
+class _Stride2Fn(torch.autograd.Function):
+    """Stride-2 NHWC subsample via coalesced row-copy kernels; backward is a
+    single-pass scatter that zero-fills unsampled rows (torch's strided
+    slicing assign ran ~700us per call on these shapes)."""
+
+    @staticmethod
+    def forward(ctx, x):
+        ctx.ihw = (x.shape[2], x.shape[3])
+        return ext().stride2_gather(x)
+
+    @staticmethod
+    def backward(ctx, dy):
+        if not dy.is_contiguous(memory_format=torch.channels_last):
+            dy = dy.contiguous(memory_format=torch.channels_last)
+        return ext().stride2_scatter(dy, *ctx.ihw)
+
+
 def _conv1x1_disabled() -> bool:
     return os.environ.get("DLA_NO_CONV1X1", "0") == "1"
 
@@ -127,7 +145,7 @@ def _prep(x: torch.Tensor, conv: nn.Conv2d):
     leaves get their fp32 wgrad directly."""
     stride = conv.stride[0]
     if stride == 2:
-        x = x[:, :, ::2, ::2].contiguous(memory_format=torch.channels_last)
+        x = _Stride2Fn.apply(x)
     B, C, H, W = x.shape
     x2d = _flatten_nhwc(x)
     w2d = conv.weight.reshape(conv.weight.shape[0], conv.weight.shape[1])

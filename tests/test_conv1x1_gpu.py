@@ -221,3 +221,42 @@ def test_bottleneck_ab_vs_unfused():
         denom = ref.abs().max().clamp(min=1e-3)
         err = (got - ref).abs().max() / denom
         assert err < 8e-2, f"bottleneck {name} rel err {err:.4g}"
+
+
+def test_transpose2d_parity():
+    for r, c in [(64, 64), (512, 2048), (100, 96)]:
+        src = _rand2d(r, c, 20 + r)
+        dst = ext().transpose2d(src)
+        assert torch.equal(dst, src.t().contiguous())
+
+
+def test_stride2_gather_scatter_parity():
+    torch.manual_seed(5)
+    for B, C, H, W in [(4, 64, 28, 28), (2, 256, 56, 56), (1, 64, 7, 7)]:
+        x = torch.randn(B, C, H, W, device="cuda").to(torch.bfloat16) \
+            .contiguous(memory_format=torch.channels_last)
+        y = ext().stride2_gather(x)
+        ref = x[:, :, ::2, ::2]
+        assert y.shape == ref.shape
+        assert torch.equal(y.float(), ref.float())
+        # scatter: zero everywhere except even positions
+        dy = torch.randn_like(y).contiguous(
+            memory_format=torch.channels_last)
+        dx = ext().stride2_scatter(dy, H, W)
+        ref_dx = torch.zeros(B, C, H, W, device="cuda",
+                             dtype=torch.bfloat16)
+        ref_dx[:, :, ::2, ::2] = dy
+        assert torch.equal(dx.float(), ref_dx.float())
+
+
+def test_stride2_autograd_roundtrip():
+    from deeplearning_amd.ops.conv1x1 import _Stride2Fn
+
+    x = torch.randn(2, 64, 14, 14, device="cuda").to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last).requires_grad_(True)
+    y = _Stride2Fn.apply(x)
+    y.float().square().sum().backward()
+    xr = x.detach().float().requires_grad_(True)
+    yr = xr[:, :, ::2, ::2]
+    yr.square().sum().backward()
+    assert torch.allclose(x.grad.float(), xr.grad, atol=1e-2)
