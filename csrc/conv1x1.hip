@@ -70,10 +70,9 @@ __global__ __launch_bounds__((BN == 256) ? 512 : 256) void conv1x1_nt_kernel(
   extern __shared__ __attribute__((aligned(16))) char lds_raw[];
   bf16* const lds16 = (bf16*)lds_raw;
   const int nk = K / BK;
-  // one K-step: A fragments load straight from global (whole rows are
-  // consumed, L2-line-efficient) and only B goes through LDS — saves the
-  // A round-trip and one barrier per tile
-  const bool direct_a = (nk == 1);
+  // direct-global A fragments measured SLOWER than LDS staging at nk==1
+  // (latency-exposed loads, no glds prefetch): keep the staged path.
+  const bool direct_a = false;
   const int nbuf = nk > 1 ? 2 : 1;  // single staging buffer when one K-step
   auto a_lds = [&](int buf) { return lds16 + buf * (BM * BK); };
   auto b_lds = [&](int buf) {
@@ -207,28 +206,12 @@ __global__ __launch_bounds__((BN == 256) ? 512 : 256) void conv1x1_nt_kernel(
   }
 
   // ---- epilogue: LDS bf16 transpose -> row-major 16B stores --------------
+  // chunked over m-halves so the scratch stays at half a wave-tile (keeps
+  // the K=64 single-buffer path at 3 blocks/CU)
   constexpr int EROW = WN + 8;  // +8 bf16 pad keeps b128 reads 16B-aligned
-  bf16* ep = (bf16*)lds_raw + wave * WM * EROW;
-  {
-    const float bcol[4] = {
-        bias != nullptr ? bias[n0 + wn_off + 0 * 16 + (lane & 15)] : 0.f,
-        bias != nullptr ? bias[n0 + wn_off + 1 * 16 + (lane & 15)] : 0.f,
-        NFR > 2 && bias != nullptr ? bias[n0 + wn_off + 2 * 16 + (lane & 15)] : 0.f,
-        NFR > 3 && bias != nullptr ? bias[n0 + wn_off + 3 * 16 + (lane & 15)] : 0.f};
-#pragma unroll
-    for (int mi = 0; mi < MFR; ++mi)
-#pragma unroll
-      for (int ni = 0; ni < NFR; ++ni)
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int row = mi * 16 + (lane >> 4) * 4 + r;
-          const int col = ni * 16 + (lane & 15);
-          ep[row * EROW + col] = from_f32<bf16>(acc[mi][ni][r] + bcol[ni]);
-        }
-  }
-  __syncthreads();
-
-  // lane -> fixed 8-channel chunk (c0), rows strided by 8
+  constexpr int EPC = (MFR >= 4) ? 2 : 1;   // epilogue chunks
+  constexpr int MFC = MFR / EPC;            // m-frags per chunk
+  bf16* ep = (bf16*)lds_raw + wave * (MFC * 16) * EROW;
   const int c0 = (lane & 7) * 8;           // channel chunk in wave tile
   const int ng0 = n0 + wn_off + c0;        // global channel of chunk start
   const bool has_ss = scale != nullptr;
@@ -237,26 +220,48 @@ __global__ __launch_bounds__((BN == 256) ? 512 : 256) void conv1x1_nt_kernel(
 #pragma unroll
     for (int j = 0; j < 8; ++j) { sc[j] = scale[ng0 + j]; sh[j] = shift[ng0 + j]; }
   }
+  float bcol[NFR];
 #pragma unroll
-  for (int i = 0; i < WM / 8; ++i) {
-    const int r = (lane >> 3) + 8 * i;
-    const int64_t gm = m0 + wm_off + r;
-    if (gm < M) {
-      Vec<bf16, 8> v = vload<bf16, 8>(ep + r * EROW + c0);
-      Vec<bf16, 8> res;
-      if (residual != nullptr)
-        res = vload<bf16, 8>(residual + gm * N + ng0);
-      Vec<bf16, 8> out;
+  for (int ni = 0; ni < NFR; ++ni)
+    bcol[ni] = bias != nullptr ? bias[n0 + wn_off + ni * 16 + (lane & 15)]
+                               : 0.f;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float y = to_f32(v.v[j]);
-        if (has_ss) y = y * sc[j] + sh[j];
-        if (residual != nullptr) y += to_f32(res.v[j]);
-        if (relu) y = fmaxf(y, 0.f);
-        out.v[j] = from_f32<bf16>(y);
-      }
-      vstore<bf16, 8>(C + gm * N + ng0, out);
+  for (int h = 0; h < EPC; ++h) {
+#pragma unroll
+    for (int mc = 0; mc < MFC; ++mc) {
+      const int mi = h * MFC + mc;
+#pragma unroll
+      for (int ni = 0; ni < NFR; ++ni)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int row = mc * 16 + (lane >> 4) * 4 + r;
+          const int col = ni * 16 + (lane & 15);
+          ep[row * EROW + col] = from_f32<bf16>(acc[mi][ni][r] + bcol[ni]);
+        }
     }
+    __syncthreads();
+#pragma unroll
+    for (int i = 0; i < (MFC * 16) / 8; ++i) {
+      const int r = (lane >> 3) + 8 * i;
+      const int64_t gm = m0 + wm_off + h * (MFC * 16) + r;
+      if (gm < M) {
+        Vec<bf16, 8> v = vload<bf16, 8>(ep + r * EROW + c0);
+        Vec<bf16, 8> res;
+        if (residual != nullptr)
+          res = vload<bf16, 8>(residual + gm * N + ng0);
+        Vec<bf16, 8> out;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float y = to_f32(v.v[j]);
+          if (has_ss) y = y * sc[j] + sh[j];
+          if (residual != nullptr) y += to_f32(res.v[j]);
+          if (relu) y = fmaxf(y, 0.f);
+          out.v[j] = from_f32<bf16>(y);
+        }
+        vstore<bf16, 8>(C + gm * N + ng0, out);
+      }
+    }
+    if (h + 1 < EPC) __syncthreads();
   }
 }
 
@@ -418,9 +423,9 @@ std::vector<torch::Tensor> conv1x1_fwd(torch::Tensor a, torch::Tensor b,
     // single staging buffer suffices when there is only one K-step; with one
     // K-step A skips LDS entirely (direct-global fragments)
     const int nbuf = nk > 1 ? 2 : 1;
-    const int lds_stage = nk == 1 ? BN * 64 * 2
-                                  : (128 * 64 + BN * 64) * 2 * nbuf;
-    const int lds_ep = NW * WM * (64 + 8) * 2;  // bf16 transpose scratch
+    const int lds_stage = (128 * 64 + BN * 64) * 2 * nbuf;
+    // chunked transpose scratch: half a wave-tile when MFR >= 4
+    const int lds_ep = NW * (WM >= 64 ? WM / 2 : WM) * (64 + 8) * 2;
     const int lds_st = ST ? ((BN == 64) ? 4 : 2) * 2 * BN * 4 : 0;
     const int lds = std::max(std::max(lds_stage, lds_ep), lds_st);
     if (lds > 65536) {
