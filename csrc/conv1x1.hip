@@ -272,16 +272,16 @@ __global__ __launch_bounds__((BN == 256) ? 512 : 256) void conv1x1_nt_kernel(
 // then read m-contiguous b128. 64x64 output tile, 4 waves (32x32 each).
 // ---------------------------------------------------------------------------
 template <int BCO, int BCI>
-__global__ __launch_bounds__((BCO == 128) ? 512 : 256) void conv1x1_wgrad_kernel(
+__global__ __launch_bounds__((BCO * BCI >= 8192) ? 512 : 256) void conv1x1_wgrad_kernel(
     const bf16* __restrict__ dy,  // [M,N]
     const bf16* __restrict__ x,   // [M,K]
     float* __restrict__ dW,       // [N,K] pre-zeroed
     int64_t M, int K, int N, int64_t chunk) {
   constexpr int KM = 64;   // m per step
   constexpr int TR = 72;   // LDS row length (elements) for [ch][KM] image
-  constexpr int NW = (BCO == 128) ? 8 : 4;
-  constexpr int WGC = 2;               // wave grid over co
-  constexpr int WGI = NW / WGC;        // wave grid over ci (4 or 2)
+  constexpr int NW = (BCO * BCI >= 8192) ? 8 : 4;
+  constexpr int WGC = (BCO >= 128 && BCI == 64) ? 4 : 2;  // wave grid over co
+  constexpr int WGI = NW / WGC;        // wave grid over ci
   constexpr int WCO = BCO / WGC;       // 64 or 32
   constexpr int WCI = BCI / WGI;       // 32
   constexpr int FCO = WCO / 16, FCI = WCI / 16;
@@ -467,34 +467,45 @@ torch::Tensor conv1x1_wgrad(torch::Tensor dy, torch::Tensor x) {
   TORCH_CHECK(dy.size(0) == M, "conv1x1_wgrad: M mismatch");
   TORCH_CHECK(K % 64 == 0 && N % 64 == 0, "conv1x1_wgrad: K,N must be %64");
   auto dW = torch::zeros({N, K}, x.options().dtype(torch::kFloat));
-  auto launch = [&](auto tag) {
-    constexpr int BT = decltype(tag)::value;  // BCO == BCI == BT
-    const int tiles = (N / BT) * (K / BT);
+  auto launch = [&](auto cotag, auto citag) {
+    constexpr int BCO = decltype(cotag)::value;
+    constexpr int BCI = decltype(citag)::value;
+    const int tiles = (N / BCO) * (K / BCI);
     int splits = (int)std::min<int64_t>(
         std::max<int64_t>(1, 1024 / tiles),
-        (M + 4095) / 4096);
+        (M + 2047) / 2048);
     const int64_t chunk0 = (M + splits - 1) / splits;
     const int64_t chunk = ((chunk0 + 63) / 64) * 64;  // multiple of KM
     splits = (int)((M + chunk - 1) / chunk);
-    const int lds = 2 * (2 * BT) * 72 * 2;
+    const int lds = 2 * (BCO + BCI) * 72 * 2;
     if (lds > 65536) {
       static bool done = false;
       if (!done) {
         (void)hipFuncSetAttribute(
-            (const void*)&dla::conv1x1_wgrad_kernel<BT, BT>,
+            (const void*)&dla::conv1x1_wgrad_kernel<BCO, BCI>,
             hipFuncAttributeMaxDynamicSharedMemorySize, 163840);
         done = true;
       }
     }
-    hipLaunchKernelGGL((dla::conv1x1_wgrad_kernel<BT, BT>),
-                       dim3(N / BT, K / BT, splits),
-                       dim3(BT == 128 ? 512 : 256), lds, dla::stream(),
-                       (const dla::bf16*)dy.data_ptr(),
+    hipLaunchKernelGGL((dla::conv1x1_wgrad_kernel<BCO, BCI>),
+                       dim3(N / BCO, K / BCI, splits),
+                       dim3(BCO * BCI >= 8192 ? 512 : 256), lds,
+                       dla::stream(), (const dla::bf16*)dy.data_ptr(),
                        (const dla::bf16*)x.data_ptr(), dW.data_ptr<float>(),
                        M, K, N, chunk);
   };
-  if (N % 128 == 0 && K % 128 == 0) launch(std::integral_constant<int, 128>{});
-  else launch(std::integral_constant<int, 64>{});
+  // pick the instantiated tile minimizing re-read traffic (1/BCO + 1/BCI)
+  const bool n256 = N % 256 == 0, n128 = N % 128 == 0, k128 = K % 128 == 0;
+  if (n128 && k128)
+    launch(std::integral_constant<int, 128>{}, std::integral_constant<int, 128>{});
+  else if (n256)
+    launch(std::integral_constant<int, 256>{}, std::integral_constant<int, 64>{});
+  else if (k128)
+    launch(std::integral_constant<int, 64>{}, std::integral_constant<int, 128>{});
+  else if (n128)
+    launch(std::integral_constant<int, 128>{}, std::integral_constant<int, 64>{});
+  else
+    launch(std::integral_constant<int, 64>{}, std::integral_constant<int, 64>{});
   HIP_CHECK_ERR();
   return dW;
 }
