@@ -448,7 +448,10 @@ std::vector<torch::Tensor> conv1x1_fwd(torch::Tensor a, torch::Tensor b,
     if (want_stats) launch(bntag, std::true_type{});
     else launch(bntag, std::false_type{});
   };
-  if (N % 256 == 0) pick(std::integral_constant<int, 256>{});
+  // BN=256 (512 threads, single A pass) wins only at nk==1 where the LDS
+  // fits 3 blocks/CU; at nk>=2 its 96KB double-buffer drops occupancy to
+  // 1 block/CU and BN=128 (2 blocks/CU) measures faster.
+  if (N % 256 == 0 && nk == 1) pick(std::integral_constant<int, 256>{});
   else if (N % 128 == 0) pick(std::integral_constant<int, 128>{});
   else pick(std::integral_constant<int, 64>{});
   HIP_CHECK_ERR();
@@ -494,12 +497,11 @@ torch::Tensor conv1x1_wgrad(torch::Tensor dy, torch::Tensor x) {
                        (const dla::bf16*)x.data_ptr(), dW.data_ptr<float>(),
                        M, K, N, chunk);
   };
-  // pick the instantiated tile minimizing re-read traffic (1/BCO + 1/BCI)
-  const bool n256 = N % 256 == 0, n128 = N % 128 == 0, k128 = K % 128 == 0;
+  // pick the instantiated tile minimizing re-read traffic; (256,64) measured
+  // slower than (128,64) (1 block/CU, staging-bound) so 128 caps BCO
+  const bool n128 = N % 128 == 0, k128 = K % 128 == 0;
   if (n128 && k128)
     launch(std::integral_constant<int, 128>{}, std::integral_constant<int, 128>{});
-  else if (n256)
-    launch(std::integral_constant<int, 256>{}, std::integral_constant<int, 64>{});
   else if (k128)
     launch(std::integral_constant<int, 64>{}, std::integral_constant<int, 128>{});
   else if (n128)
