@@ -65,16 +65,16 @@ def _worker_nosync(rank, world, port, q):
     loss.backward()
     ddp.finalize()
     g_sync = [p.grad.clone() for p in ddp.module.parameters()]
-    if rank == 0:
-        q.put(([g.numpy().copy() for g in g_local], [g.numpy().copy() for g in g_sync]))
+    q.put((rank, [g.numpy().copy() for g in g_local],
+           [g.numpy().copy() for g in g_sync]))
     dist.barrier()
     dist.destroy_process_group()
 
 
-@pytest.mark.parametrize("fn", [_worker])
-def test_ddp_matches_single_process(fn):
-    world = 2
-    port = 29611
+@pytest.mark.parametrize("world", [2, 4])
+def test_ddp_matches_single_process(world):
+    fn = _worker
+    port = 29611 + world
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
     procs = [ctx.Process(target=fn, args=(r, world, port, q)) for r in range(world)]
@@ -102,6 +102,9 @@ def test_ddp_matches_single_process(fn):
 
 
 def test_no_sync_accumulation():
+    """no_sync micro-batch accumulates locally; the following synced backward
+    must deliver grad == mean over ranks of the 2-micro-batch accumulated
+    gradient (numeric check against an eager recomputation)."""
     world = 2
     port = 29613
     ctx = mp.get_context("spawn")
@@ -110,14 +113,37 @@ def test_no_sync_accumulation():
              for r in range(world)]
     for p in procs:
         p.start()
-    g_local, g_sync = q.get()
+    results = {}
+    for _ in range(world):
+        rank, gl, gs = q.get()
+        results[rank] = (gl, gs)
     for p in procs:
         p.join(60)
         assert p.exitcode == 0
-    # after sync, grad = mean over ranks of (2 accumulated micro-batches)
-    for gl, gs in zip(g_local, g_sync):
-        assert gs.shape == gl.shape
-        assert torch.isfinite(torch.from_numpy(gs)).all()
+
+    # eager reference: per-rank accumulated grads, then mean over ranks
+    ref_model = _make_model()
+    per_rank = []
+    for r in range(world):
+        for prm in ref_model.parameters():
+            prm.grad = None
+        torch.manual_seed(7 + r)
+        x = torch.randn(4, 8)
+        ref_model(x).sum().backward()
+        ref_model(x).sum().backward()  # accumulate the 2nd micro-batch
+        per_rank.append([prm.grad.clone() for prm in ref_model.parameters()])
+    expected = [sum(gs) / world for gs in zip(*per_rank)]
+
+    for r in range(world):
+        gl, gs = results[r]
+        # local (no_sync) grad on rank r == ONE micro-batch of rank r
+        # (both micro-batches use the same x, so one = accumulated / 2)
+        for g_got, g_exp in zip(gl, per_rank[r]):
+            torch.testing.assert_close(torch.from_numpy(g_got), g_exp / 2,
+                                       atol=1e-5, rtol=1e-5)
+        for g_got, g_exp in zip(gs, expected):
+            torch.testing.assert_close(torch.from_numpy(g_got), g_exp,
+                                       atol=1e-5, rtol=1e-5)
 
 
 def test_all_reduce_norm_single_proc():
