@@ -160,3 +160,49 @@ def shared_random_seed() -> int:
         t = t.cuda()
     dist.broadcast(t, src=0)
     return int(t.item())
+
+
+def find_free_port() -> int:
+    """Bind port 0 to let the OS pick a free port (reference YOLOX
+    yolox/core/launch.py:24-36)."""
+    import socket
+
+    sock = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
+    sock.bind(("", 0))
+    port = sock.getsockname()[1]
+    sock.close()
+    return port
+
+
+def _launch_worker(local_rank, main_func, world_size, num_gpus, machine_rank,
+                   dist_url, args):
+    host_port = dist_url.split("//")[-1]
+    host, port = host_port.rsplit(":", 1)
+    os.environ.update(MASTER_ADDR=host, MASTER_PORT=port,
+                      RANK=str(machine_rank * num_gpus + local_rank),
+                      WORLD_SIZE=str(world_size), LOCAL_RANK=str(local_rank))
+    main_func(*args)
+
+
+def launch(main_func, num_gpus_per_machine, num_machines=1, machine_rank=0,
+           dist_url="auto", args=(), start_method="spawn"):
+    """Self-spawning multi-process launcher (reference YOLOX
+    yolox/core/launch.py:39-147): free-port autodetect + mp.start_processes.
+
+    Each worker exports the torchrun-style env (MASTER_*/RANK/LOCAL_RANK/
+    WORLD_SIZE) and calls main_func, whose init_distributed() then joins the
+    group over RCCL (or gloo on CPU). world_size==1 calls main_func inline.
+    """
+    import torch.multiprocessing as mp
+
+    world_size = num_machines * num_gpus_per_machine
+    if world_size <= 1:
+        return main_func(*args)
+    if dist_url == "auto":
+        assert num_machines == 1, "dist_url=auto needs a single machine"
+        dist_url = f"tcp://127.0.0.1:{find_free_port()}"
+    mp.start_processes(
+        _launch_worker, nprocs=num_gpus_per_machine,
+        args=(main_func, world_size, num_gpus_per_machine, machine_rank,
+              dist_url, args),
+        daemon=False, start_method=start_method)

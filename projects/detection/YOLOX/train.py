@@ -6,6 +6,7 @@ from pathlib import Path
 sys.path.insert(0, str(Path(__file__).resolve().parents[3]))
 
 import argparse
+import os
 import random
 import time
 
@@ -24,7 +25,9 @@ from deeplearning_amd.models import build_model
 from deeplearning_amd.ops import ModelEMA
 from deeplearning_amd.parallel.syncbn import all_reduce_norm
 
-if __name__ == "__main__":
+
+
+def parse_args():
     p = argparse.ArgumentParser()
     p.add_argument("--model", default="yolox_s",
                    choices=["yolox_s", "yolox_m", "yolox_l", "yolox_x"])
@@ -39,7 +42,13 @@ if __name__ == "__main__":
     p.add_argument("--multiscale", action="store_true",
                    help="random input size every 10 iters, broadcast-synced "
                         "across ranks (ref yolox_base.py:167-187)")
-    args = p.parse_args()
+    p.add_argument("--devices", type=int, default=1,
+                   help="self-spawn N processes via launch() "
+                        "(ref yolox/core/launch.py:39-147); 1 = inline")
+    return p.parse_args()
+
+
+def main(args):
 
     init_distributed()
     seed_everything(0, rank=get_rank())
@@ -109,3 +118,12 @@ if __name__ == "__main__":
             save_checkpoint(run_dir / "weights" / "latest_ckpt.pth", model,
                             opt, epoch=epoch)
     cleanup()
+
+
+if __name__ == "__main__":
+    args = parse_args()
+    if args.devices > 1 and "RANK" not in os.environ:
+        from deeplearning_amd.core.dist import launch
+        launch(main, args.devices, args=(args,))
+    else:
+        main(args)

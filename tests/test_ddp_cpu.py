@@ -169,3 +169,24 @@ def test_syncbn_conversion_preserves_fused_relu():
     assert isinstance(conv[1][1], nn.ReLU)
     assert isinstance(conv[2], nn.SyncBatchNorm)
     assert float(conv[1][0].running_mean[0]) == 0.5
+
+
+def _launch_probe(tag):
+    import torch.distributed as dist
+
+    from deeplearning_amd.core.dist import cleanup, init_distributed
+    info = init_distributed()
+    assert info["world_size"] == 2, info
+    t = torch.tensor([dist.get_rank() + 1.0])
+    dist.all_reduce(t)
+    assert float(t) == 3.0  # 1 + 2
+    assert tag == "hello"
+    cleanup()
+
+
+def test_yolox_style_launch_spawns_world2():
+    """launch() (ref YOLOX yolox/core/launch.py:39-147): free-port autodetect
+    + mp.start_processes; workers join the group via init_distributed()."""
+    from deeplearning_amd.core.dist import launch
+
+    launch(_launch_probe, num_gpus_per_machine=2, args=("hello",))
