@@ -14,7 +14,9 @@
 
 namespace {
 
-inline float iou_xyxy(const float* a, const float* b) {
+// pycocotools protocol (maskUtils.iou): for iscrowd gt the denominator is
+// the DETECTION's area, not the union.
+inline float iou_xyxy(const float* a, const float* b, bool crowd) {
   const float ix1 = std::max(a[0], b[0]);
   const float iy1 = std::max(a[1], b[1]);
   const float ix2 = std::min(a[2], b[2]);
@@ -24,7 +26,7 @@ inline float iou_xyxy(const float* a, const float* b) {
   const float inter = iw * ih;
   const float area_a = std::max(0.f, a[2] - a[0]) * std::max(0.f, a[3] - a[1]);
   const float area_b = std::max(0.f, b[2] - b[0]) * std::max(0.f, b[3] - b[1]);
-  const float uni = area_a + area_b - inter;
+  const float uni = crowd ? area_a : area_a + area_b - inter;
   return uni > 0.f ? inter / uni : 0.f;
 }
 
@@ -60,14 +62,14 @@ std::vector<torch::Tensor> cocoeval_match_image(torch::Tensor det_boxes,
     return {matched, ignored, scores, torch::tensor(n_gt)};
 
   // IoU matrix once
+  const bool* cp = crowd.data_ptr<bool>();
   std::vector<float> ious((size_t)D * G);
   const float* bp = boxes.data_ptr<float>();
   const float* gp = gt.data_ptr<float>();
   for (int64_t d = 0; d < D; ++d)
     for (int64_t g = 0; g < G; ++g)
-      ious[d * G + g] = iou_xyxy(bp + d * 4, gp + g * 4);
+      ious[d * G + g] = iou_xyxy(bp + d * 4, gp + g * 4, cp[g]);
 
-  const bool* cp = crowd.data_ptr<bool>();
   const float* tp = thrs.data_ptr<float>();
   auto m_acc = matched.accessor<bool, 2>();
   auto i_acc = ignored.accessor<bool, 2>();
@@ -75,24 +77,25 @@ std::vector<torch::Tensor> cocoeval_match_image(torch::Tensor det_boxes,
   for (int64_t t = 0; t < T; ++t) {
     std::fill(taken.begin(), taken.end(), 0);
     for (int64_t d = 0; d < D; ++d) {
-      float best_iou = tp[t];
-      int64_t best_g = -1;
+      // pycocotools semantics (gts sorted ignore-last + break): ANY
+      // non-crowd match beats any crowd match regardless of IoU; ties
+      // within a class of gts keep the later one.
+      float best_nc = tp[t], best_c = tp[t];
+      int64_t g_nc = -1, g_c = -1;
       for (int64_t g = 0; g < G; ++g) {
-        if (taken[g] && !cp[g]) continue;
-        // prefer non-crowd matches; crowd only if nothing else found
-        if (best_g >= 0 && !cp[best_g] && cp[g]) continue;
-        if (ious[d * G + g] >= best_iou) {
-          best_iou = ious[d * G + g];
-          best_g = g;
+        const float v = ious[d * G + g];
+        if (cp[g]) {
+          if (v >= best_c) { best_c = v; g_c = g; }
+        } else {
+          if (taken[g]) continue;
+          if (v >= best_nc) { best_nc = v; g_nc = g; }
         }
       }
-      if (best_g >= 0) {
-        if (cp[best_g]) {
-          i_acc[t][d] = true;
-        } else {
-          m_acc[t][d] = true;
-          taken[best_g] = 1;
-        }
+      if (g_nc >= 0) {
+        m_acc[t][d] = true;
+        taken[g_nc] = 1;
+      } else if (g_c >= 0) {
+        i_acc[t][d] = true;
       }
     }
   }

@@ -49,26 +49,37 @@ def match_image(det_boxes, det_scores, gt_boxes, gt_crowd, iou_thrs,
     if D == 0 or G == 0:
         return matched, ignored, det_scores, n_gt
     ious = box_iou(det_boxes, gt_boxes)  # D, G
+    if gt_crowd.any():
+        # pycocotools protocol: crowd IoU = intersection / det area
+        lt = torch.max(det_boxes[:, None, :2], gt_boxes[None, :, :2])
+        rb = torch.min(det_boxes[:, None, 2:], gt_boxes[None, :, 2:])
+        inter = (rb - lt).clamp(min=0).prod(-1)
+        det_area = ((det_boxes[:, 2] - det_boxes[:, 0]) *
+                    (det_boxes[:, 3] - det_boxes[:, 1])).clamp(min=1e-9)
+        crowd_iou = inter / det_area[:, None]
+        ious = torch.where(gt_crowd[None, :], crowd_iou, ious)
     for t, thr in enumerate(iou_thrs):
         taken = torch.zeros(G, dtype=torch.bool)
         for d in range(D):
-            best_iou = thr
-            best_g = -1
+            # pycocotools semantics (gts sorted ignore-last + break): ANY
+            # non-crowd match beats any crowd match regardless of IoU
+            best_nc = best_c = thr
+            g_nc = g_c = -1
             for g in range(G):
-                if taken[g] and not gt_crowd[g]:
-                    continue
-                # prefer non-crowd matches; crowd only if nothing else
-                if best_g >= 0 and not gt_crowd[best_g] and gt_crowd[g]:
-                    continue
-                if ious[d, g] >= best_iou:
-                    best_iou = float(ious[d, g])
-                    best_g = g
-            if best_g >= 0:
-                if gt_crowd[best_g]:
-                    ignored[t, d] = True
+                v = float(ious[d, g])
+                if gt_crowd[g]:
+                    if v >= best_c:
+                        best_c, g_c = v, g
                 else:
-                    matched[t, d] = True
-                    taken[best_g] = True
+                    if taken[g]:
+                        continue
+                    if v >= best_nc:
+                        best_nc, g_nc = v, g
+            if g_nc >= 0:
+                matched[t, d] = True
+                taken[g_nc] = True
+            elif g_c >= 0:
+                ignored[t, d] = True
     return matched, ignored, det_scores, n_gt
 
 
