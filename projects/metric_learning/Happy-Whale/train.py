@@ -19,6 +19,12 @@ from deeplearning_amd.models.metric import TripletLoss
 if __name__ == "__main__":
     p = argparse.ArgumentParser()
     p.add_argument("--num-ids", type=int, default=8)
+    p.add_argument("--backbone", default="supcon_resnet50",
+                   choices=["supcon_resnet50", "dpn68", "dpn92",
+                            "inception_v4", "xception", "senet154",
+                            "polynet", "nasnet_a_mobile", "nasnet_a_large"],
+                   help="retrieval backbone (reference modelZoo: dpn/"
+                        "inceptionV4/xception/senet/ployNet/nasnet)")
     p.add_argument("--epochs", type=int, default=2)
     p.add_argument("--batch-size", type=int, default=8)
     p.add_argument("--lr", type=float, default=1e-3)
@@ -27,7 +33,25 @@ if __name__ == "__main__":
 
     seed_everything(0)
     device = select_device(args.device)
-    model = build_model("supcon_resnet50", feat_dim=256).to(device)
+    if args.backbone == "supcon_resnet50":
+        model = build_model(args.backbone, feat_dim=256).to(device)
+    else:
+        # zoo backbone -> pooled features -> 256-d embedding (ref
+        # retrieval/train.py feature head)
+        bb = build_model(args.backbone, num_classes=1)
+        embed = torch.nn.Linear(bb.num_features, 256)
+
+        class _Retrieval(torch.nn.Module):
+            def __init__(self, bb, embed):
+                super().__init__()
+                self.bb, self.embed = bb, embed
+
+            def forward(self, x):
+                f = self.bb.forward_features(x)
+                f = torch.nn.functional.adaptive_avg_pool2d(f, 1).flatten(1)
+                return torch.nn.functional.normalize(self.embed(f), dim=1)
+
+        model = _Retrieval(bb, embed).to(device)
     head = torch.nn.Linear(256, args.num_ids).to(device)
     triplet = TripletLoss(margin=None)  # soft-margin (ref :38-189)
     opt = torch.optim.AdamW(list(model.parameters()) +

@@ -93,3 +93,32 @@ def test_zoo_extra_forward_backward(name, size):
     out = m(torch.randn(2, 3, size, size))
     out.sum().backward()
     assert out.shape == (2, 10)
+
+
+@pytest.mark.parametrize("name,size,feat", [
+    ("xception", 96, 2048),
+    ("senet154", 96, 2048),
+    ("polynet", 128, 2048),
+    ("nasnet_a_mobile", 96, 1056),
+])
+def test_zoo_tail_models(name, size, feat):
+    """Happy-Whale model-zoo tail (ref modelZoo/{xception,senet,ployNet,
+    nasnet}.py): forward shape + feature dim + one backward step."""
+    from deeplearning_amd.models import build_model
+
+    m = build_model(name, num_classes=5)
+    assert m.num_features == feat
+    x = torch.randn(2, 3, size, size)
+    y = m(x)
+    assert y.shape == (2, 5)
+    y.square().mean().backward()
+    g = next(p.grad for p in m.parameters() if p.grad is not None)
+    assert torch.isfinite(g).all()
+
+
+def test_nasnet_large_constructs():
+    from deeplearning_amd.models import build_model
+
+    m = build_model("nasnet_a_large", num_classes=3)
+    n = sum(p.numel() for p in m.parameters())
+    assert n > 50e6  # the @large config (published 88.7M)
