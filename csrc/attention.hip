@@ -774,12 +774,17 @@ torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B) {
 }
 
 
-// Swin-v2 cosine attention forward (inference; round-2 staging — see
-// ROADMAP.md). logit_scale: [H] fp32 = clamp(exp(param), max=100).
-torch::Tensor attn_fwd_cosine(torch::Tensor qkv, int64_t num_heads,
+// Swin-v2 cosine attention forward. logit_scale: [H] fp32 =
+// clamp(exp(param), max=100). save_p additionally returns the softmax
+// probabilities P [B,H,N,N] bf16 for the training backward (the Python
+// wrapper runs the cosine chain rule from P — same split as the v1
+// bias path).
+std::vector<torch::Tensor> attn_fwd_cosine(torch::Tensor qkv,
+                              int64_t num_heads,
                               torch::Tensor logit_scale,
                               c10::optional<torch::Tensor> bias,
-                              c10::optional<torch::Tensor> mask) {
+                              c10::optional<torch::Tensor> mask,
+                              bool save_p) {
   TORCH_CHECK(qkv.scalar_type() == at::kBFloat16 && qkv.is_contiguous());
   const int B = (int)qkv.size(0);
   const int N = (int)qkv.size(1);
@@ -790,6 +795,8 @@ torch::Tensor attn_fwd_cosine(torch::Tensor qkv, int64_t num_heads,
   auto ls = logit_scale.to(torch::kFloat).contiguous();
   TORCH_CHECK(ls.numel() == H, "logit_scale must be [H]");
   auto out = torch::empty({B, N, (int64_t)H * D}, qkv.options());
+  torch::Tensor pten;
+  if (save_p) pten = torch::empty({B, H, N, N}, qkv.options());
 
   int n_win = 1;
   const float* bias_ptr = nullptr;
@@ -810,27 +817,31 @@ torch::Tensor attn_fwd_cosine(torch::Tensor qkv, int64_t num_heads,
                       (int)sizeof(__hip_bfloat16) +
                   Npad * (int)sizeof(float);  // + kinv
   dim3 grid(B * H), block(nwaves * 64);
-  auto run = [&](auto dtag, auto btag, auto mtag) {
+  auto run = [&](auto dtag, auto btag, auto mtag, auto ptag) {
     constexpr int DD = decltype(dtag)::value;
     constexpr bool BB = decltype(btag)::value;
     constexpr bool MM = decltype(mtag)::value;
+    constexpr bool PP = decltype(ptag)::value;
     hipLaunchKernelGGL(
-        (dla::attn_fwd_kernel<DD, BB, MM, false, true>), grid, block, lds,
+        (dla::attn_fwd_kernel<DD, BB, MM, PP, true>), grid, block, lds,
         dla::stream(), (const __hip_bfloat16*)qkv.data_ptr(), bias_ptr,
-        mask_ptr, (__hip_bfloat16*)out.data_ptr(), nullptr, nullptr,
+        mask_ptr, (__hip_bfloat16*)out.data_ptr(),
+        save_p ? (__hip_bfloat16*)pten.data_ptr() : nullptr, nullptr,
         ls.data_ptr<float>(), B, N, H, n_win, 1.0f);
   };
   using T = std::true_type;
   using F = std::false_type;
   auto d2 = [&](auto dtag) {
     const bool bb = bias_ptr != nullptr, mm = mask_ptr != nullptr;
-    if (bb && mm) run(dtag, T{}, T{});
-    else if (bb) run(dtag, T{}, F{});
-    else if (mm) run(dtag, F{}, T{});
-    else run(dtag, F{}, F{});
+    if (bb && mm) save_p ? run(dtag, T{}, T{}, T{}) : run(dtag, T{}, T{}, F{});
+    else if (bb) save_p ? run(dtag, T{}, F{}, T{}) : run(dtag, T{}, F{}, F{});
+    else if (mm) save_p ? run(dtag, F{}, T{}, T{}) : run(dtag, F{}, T{}, F{});
+    else save_p ? run(dtag, F{}, F{}, T{}) : run(dtag, F{}, F{}, F{});
   };
   if (D == 64) d2(std::integral_constant<int, 64>{});
   else d2(std::integral_constant<int, 32>{});
   HIP_CHECK_ERR();
-  return out;
+  std::vector<torch::Tensor> res = {out};
+  if (save_p) res.push_back(pten);
+  return res;
 }

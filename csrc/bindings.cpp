@@ -78,10 +78,10 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor qkv, int64_t num_heads,
 std::vector<torch::Tensor> attn_bwd(torch::Tensor qkv, torch::Tensor dout,
                                     torch::Tensor stats, torch::Tensor drow,
                                     int64_t num_heads, double scale);
-torch::Tensor attn_fwd_cosine(torch::Tensor qkv, int64_t num_heads,
+std::vector<torch::Tensor> attn_fwd_cosine(torch::Tensor qkv, int64_t num_heads,
                               torch::Tensor logit_scale,
                               c10::optional<torch::Tensor> bias,
-                              c10::optional<torch::Tensor> mask);
+                              c10::optional<torch::Tensor> mask, bool save_p);
 torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B);
 // my_add.cpp (custom-op export tutorial)
 torch::Tensor my_add(torch::Tensor a, torch::Tensor b);
@@ -137,7 +137,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_bwd", &attn_bwd);
   m.def("attn_fwd_cosine", &attn_fwd_cosine, py::arg("qkv"),
         py::arg("num_heads"), py::arg("logit_scale"),
-        py::arg("bias") = py::none(), py::arg("mask") = py::none());
+        py::arg("bias") = py::none(), py::arg("mask") = py::none(),
+        py::arg("save_p") = false);
   m.def("mfma_probe", &mfma_probe);
   m.def("cocoeval_match_image", &cocoeval_match_image);
   m.def("my_add", &my_add);

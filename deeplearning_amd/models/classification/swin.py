@@ -95,13 +95,12 @@ class WindowAttention(nn.Module):
         return bias.permute(2, 0, 1).contiguous().unsqueeze(0)  # 1, nH, N, N
 
     def forward(self, x, mask=None):
-        import os
-
         from ...ops.attention import fused_attention
         B_, N, C = x.shape
-        if self.v2 and not self.training and not torch.is_grad_enabled() \
-                and os.environ.get("DLA_V2_FUSED") == "1":
-            # staged fused cosine path (inference; see ROADMAP.md)
+        if self.v2 and (self.attn_drop.p == 0.0 or not self.training):
+            # fused cosine path, train + eval (in-kernel q/k norms +
+            # per-head logit scale; backward from kernel-saved P with
+            # grads for logit_scale and the CPB bias)
             from ...ops.attention import fused_attention_cosine
             qkv_bias = None
             if self.q_bias is not None:

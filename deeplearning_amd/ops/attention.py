@@ -120,18 +120,66 @@ def fused_attention(qkv: torch.Tensor, num_heads: int, scale: float,
     return _eager_attention(qkv, num_heads, scale, bias, mask)
 
 
+class _AttnCosineFn(torch.autograd.Function):
+    """Swin-v2 cosine attention: fused HIP forward (in-kernel q/k row norms
+    + per-head logit scale), backward = cosine chain rule from the
+    kernel-saved P (same forward-fused/backward-from-P split as the v1
+    bias path). Gradients flow to qkv, logit_scale and bias."""
+
+    @staticmethod
+    def forward(ctx, qkv, num_heads, lscale, bias, mask):
+        needs = torch.is_grad_enabled() and (
+            qkv.requires_grad or lscale.requires_grad or
+            (bias is not None and bias.requires_grad))
+        res = ext().attn_fwd_cosine(qkv.contiguous(), num_heads,
+                                    lscale.detach(), bias, mask, needs)
+        ctx.num_heads = num_heads
+        ctx.has_bias = bias is not None
+        if needs:
+            ctx.save_for_backward(qkv, lscale, res[1])
+        return res[0]
+
+    @staticmethod
+    def backward(ctx, dout):
+        import torch.nn.functional as F
+
+        qkv, lscale, p = ctx.saved_tensors
+        H = ctx.num_heads
+        B, N, _ = qkv.shape
+        d = qkv.shape[2] // (3 * H)
+        q, k, v = qkv.reshape(B, N, 3, H, d).permute(2, 0, 3, 1, 4).unbind(0)
+        dout = dout.reshape(B, N, H, d).permute(0, 2, 1, 3)
+        qh = F.normalize(q.float(), dim=-1)
+        kh = F.normalize(k.float(), dim=-1)
+        pf = p.float()
+        dv = p.transpose(-2, -1) @ dout
+        dp = (dout @ v.transpose(-2, -1)).float()
+        ds = pf * (dp - (dp * pf).sum(dim=-1, keepdim=True))
+        c = qh @ kh.transpose(-2, -1)
+        d_ls = (ds * c).sum(dim=(0, 2, 3))           # [H]
+        dc = ds * lscale.view(1, -1, 1, 1)
+        dqh = dc @ kh
+        dkh = dc.transpose(-2, -1) @ qh
+        qn = q.float().norm(dim=-1, keepdim=True).clamp_min(1e-12)
+        kn = k.float().norm(dim=-1, keepdim=True).clamp_min(1e-12)
+        dq = (dqh - (dqh * qh).sum(-1, keepdim=True) * qh) / qn
+        dk = (dkh - (dkh * kh).sum(-1, keepdim=True) * kh) / kn
+        dqkv = torch.stack([dq.to(qkv.dtype), dk.to(qkv.dtype), dv], dim=0)
+        dqkv = dqkv.permute(1, 3, 0, 2, 4).reshape(B, N, 3 * H * d)
+        dbias = ds.sum(dim=0) if ctx.has_bias else None
+        return dqkv, None, d_ls.to(lscale.dtype), dbias, None
+
+
 def fused_attention_cosine(qkv: torch.Tensor, num_heads: int,
                            logit_scale: torch.Tensor,
                            bias: torch.Tensor | None = None,
                            mask: torch.Tensor | None = None) -> torch.Tensor:
-    """Swin-v2 cosine attention, inference only (round-2 staging; enable the
-    fused path in the model with DLA_V2_FUSED=1). logit_scale: [H] already
-    clamp(exp(param), max=100)."""
+    """Swin-v2 cosine attention on the fused HIP kernel (train + eval).
+    logit_scale: [H] already clamp(exp(param), max=100)."""
     d = qkv.shape[2] // (3 * num_heads)
     if (use_hip(qkv) and qkv.dtype == torch.bfloat16 and d in (32, 64)
-            and qkv.shape[1] <= 256 and not torch.is_grad_enabled()):
-        return ext().attn_fwd_cosine(qkv.contiguous(), num_heads,
-                                     logit_scale, bias, mask)
+            and qkv.shape[1] <= 256):
+        return _AttnCosineFn.apply(qkv, num_heads, logit_scale, bias, mask)
     # eager reference
     B, N, _ = qkv.shape
     q, k, v = qkv.reshape(B, N, 3, num_heads, -1).permute(

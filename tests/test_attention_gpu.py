@@ -128,3 +128,62 @@ def test_swin_block_fused_vs_eager_path():
         del os.environ["DLA_FORCE_EAGER"]
     err = (y_fused.float() - y_eager.float()).abs().max().item()
     assert err < 0.1, f"swin fused-vs-eager {err}"
+
+
+def test_cosine_attention_train_parity():
+    """SwinV2 fused cosine attention (default-on): fwd + bwd vs fp32 eager,
+    including logit_scale and bias grads (VERDICT round-1 item 4)."""
+    import torch.nn.functional as F
+
+    torch.manual_seed(0)
+    B, N, H, d = 16, 49, 4, 32
+    qkv = torch.randn(B, N, 3 * H * d, device="cuda").to(torch.bfloat16) \
+        .requires_grad_(True)
+    ls_param = torch.randn(H, device="cuda").mul(0.1).requires_grad_(True)
+    bias = torch.randn(H, N, N, device="cuda").mul(0.5).requires_grad_(True)
+    mask = torch.zeros(4, N, N, device="cuda")
+    mask[1, :10, 10:] = -100.0
+
+    from deeplearning_amd.ops.attention import fused_attention_cosine
+    lscale = torch.clamp(ls_param, max=torch.log(
+        torch.tensor(100.0, device="cuda"))).exp()
+    out = fused_attention_cosine(qkv, H, lscale, bias=bias, mask=mask)
+    out.float().square().mean().backward()
+
+    # fp32 eager reference
+    qkv2 = qkv.detach().float().requires_grad_(True)
+    ls2 = ls_param.detach().clone().requires_grad_(True)
+    bias2 = bias.detach().float().requires_grad_(True)
+    q, k, v = qkv2.reshape(B, N, 3, H, d).permute(2, 0, 3, 1, 4).unbind(0)
+    attn = F.normalize(q, dim=-1) @ F.normalize(k, dim=-1).transpose(-2, -1)
+    attn = attn * torch.clamp(ls2, max=torch.log(
+        torch.tensor(100.0, device="cuda"))).exp().view(1, -1, 1, 1)
+    attn = attn + bias2.unsqueeze(0)
+    attn = attn.view(B // 4, 4, H, N, N) + mask.unsqueeze(1).unsqueeze(0)
+    attn = attn.view(B, H, N, N).softmax(-1)
+    ref = (attn @ v).transpose(1, 2).reshape(B, N, H * d)
+    ref.square().mean().backward()
+
+    def relerr(a, b):
+        return (a.float() - b.float()).abs().max() / \
+            b.float().abs().max().clamp(min=1e-6)
+
+    assert relerr(out, ref.detach()) < 4e-2, relerr(out, ref.detach())
+    assert relerr(qkv.grad, qkv2.grad) < 8e-2, relerr(qkv.grad, qkv2.grad)
+    assert relerr(ls_param.grad, ls2.grad) < 8e-2
+    assert relerr(bias.grad, bias2.grad) < 8e-2
+
+
+def test_swinv2_block_trains_through_fused_path():
+    """swinv2_t forward+backward on GPU exercises the fused cosine kernel
+    (no env gate) and produces finite grads incl. logit_scale."""
+    from deeplearning_amd.models import build_model
+
+    torch.manual_seed(1)
+    m = build_model("swinv2_t", num_classes=10).cuda()
+    x = torch.randn(2, 3, 224, 224, device="cuda")
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        y = m(x)
+    y.float().square().mean().backward()
+    ls = next(p for n, p in m.named_parameters() if "logit_scale" in n)
+    assert ls.grad is not None and torch.isfinite(ls.grad).all()
