@@ -128,9 +128,10 @@ class _AttnCosineFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, qkv, num_heads, lscale, bias, mask):
-        needs = torch.is_grad_enabled() and (
-            qkv.requires_grad or lscale.requires_grad or
-            (bias is not None and bias.requires_grad))
+        # grad mode is off inside Function.forward; the no-grad case was
+        # already routed around this Function by the wrapper
+        needs = (qkv.requires_grad or lscale.requires_grad or
+                 (bias is not None and bias.requires_grad))
         res = ext().attn_fwd_cosine(qkv.contiguous(), num_heads,
                                     lscale.detach(), bias, mask, needs)
         ctx.num_heads = num_heads
@@ -179,6 +180,9 @@ def fused_attention_cosine(qkv: torch.Tensor, num_heads: int,
     d = qkv.shape[2] // (3 * num_heads)
     if (use_hip(qkv) and qkv.dtype == torch.bfloat16 and d in (32, 64)
             and qkv.shape[1] <= 256):
+        if not torch.is_grad_enabled():
+            return ext().attn_fwd_cosine(qkv.contiguous(), num_heads,
+                                         logit_scale, bias, mask, False)[0]
         return _AttnCosineFn.apply(qkv, num_heads, logit_scale, bias, mask)
     # eager reference
     B, N, _ = qkv.shape

@@ -260,3 +260,48 @@ def test_stride2_autograd_roundtrip():
     yr = xr[:, :, ::2, ::2]
     yr.square().sum().backward()
     assert torch.allclose(x.grad.float(), xr.grad, atol=1e-2)
+
+
+def test_bucketed_ddp_single_rank_rccl():
+    """BucketedDataParallel over a REAL RCCL process group (world=1 on one
+    GPU): broadcast-at-wrap, bucketed async all-reduce and finalize() all
+    execute on the nccl(RCCL) backend rather than gloo."""
+    import os
+
+    import torch.distributed as dist
+
+    from deeplearning_amd.parallel import BucketedDataParallel
+
+    if dist.is_initialized():
+        pytest.skip("process group already initialized")
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29741")
+    dist.init_process_group("nccl", rank=0, world_size=1)
+    try:
+        torch.manual_seed(0)
+        model = torch.nn.Sequential(
+            torch.nn.Linear(64, 128), torch.nn.ReLU(),
+            torch.nn.Linear(128, 32)).cuda()
+        ddp = BucketedDataParallel(model, bucket_cap_mb=0.001)
+        opt = torch.optim.SGD(ddp.module.parameters(), lr=0.1)
+        x = torch.randn(16, 64, device="cuda")
+        ref = torch.nn.Sequential(
+            torch.nn.Linear(64, 128), torch.nn.ReLU(),
+            torch.nn.Linear(128, 32)).cuda()
+        ref.load_state_dict(model.state_dict())
+        for _ in range(3):
+            loss = ddp(x).square().mean()
+            opt.zero_grad(set_to_none=True)
+            loss.backward()
+            ddp.finalize()
+            opt.step()
+        ref_opt = torch.optim.SGD(ref.parameters(), lr=0.1)
+        for _ in range(3):
+            loss = ref(x).square().mean()
+            ref_opt.zero_grad(set_to_none=True)
+            loss.backward()
+            ref_opt.step()
+        for p, q in zip(ddp.module.parameters(), ref.parameters()):
+            torch.testing.assert_close(p, q, atol=1e-5, rtol=1e-5)
+    finally:
+        dist.destroy_process_group()
