@@ -181,7 +181,7 @@ def test_swinv2_block_trains_through_fused_path():
 
     torch.manual_seed(1)
     m = build_model("swinv2_t", num_classes=10).cuda()
-    x = torch.randn(2, 3, 224, 224, device="cuda")
+    x = torch.randn(2, 3, 256, 256, device="cuda")  # v2 default img 256
     with torch.autocast("cuda", dtype=torch.bfloat16):
         y = m(x)
     y.float().square().mean().backward()
