@@ -85,7 +85,8 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ qkv,  // [B,N,3,H,D]
                      __hip_bfloat16* __restrict__ p_out,      // [B,H,N,N]|null
                      float* __restrict__ stats,               // [2,B,H,N]|null
                      const float* __restrict__ lscale,        // [H]|null
-                     int B, int N, int H, int n_win, float scale) {
+                     int B, int N, int H, int n_win, float scale,
+                     int ablate = 0) {  // phase-ablation probe (0 = full)
   constexpr int KSLICES = D / 32;
   const int b = blockIdx.x / H;
   const int h = blockIdx.x % H;
@@ -120,6 +121,7 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ qkv,  // [B,N,3,H,D]
     }
     __syncthreads();
   }
+  if (ablate >= 4) return;  // ablation: staging only
   const float ls = COSINE ? lscale[h] : scale;
 
   const __hip_bfloat16* q_src = qkv + (int64_t)b * N * 3 * H * D +
@@ -180,6 +182,7 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ qkv,  // [B,N,3,H,D]
       }
     }
 
+    if (ablate >= 3) continue;  // ablation: QK^T only
     // softmax rows (C/D: col = key = lane&15, row = (lane>>4)*4 + reg)
     const int col = lane & 15;
     float row_sum[4];
@@ -237,6 +240,7 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ qkv,  // [B,N,3,H,D]
       }
     }
 
+    if (ablate >= 2) continue;  // ablation: no PV / p_buf
     // O = P V, accumulated chunk by chunk through the small P buffer
     f32x4 o_acc[D / 16];
 #pragma unroll
@@ -278,6 +282,7 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ qkv,  // [B,N,3,H,D]
       }
     }
 
+    if (ablate == 1) continue;  // ablation: no O store
 #pragma unroll
     for (int dt = 0; dt < D / 16; ++dt) {
 #pragma unroll
@@ -688,6 +693,8 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor qkv, int64_t num_heads,
                   (int)sizeof(__hip_bfloat16);
   dim3 grid(B * H), block(nwaves * 64);
 
+  const char* abl_env = std::getenv("DLA_ATTN_ABLATE");
+  const int ablate = abl_env ? std::atoi(abl_env) : 0;
   auto launch = [&](auto dtag, auto btag, auto mtag, auto ptag) {
     constexpr int DD = decltype(dtag)::value;
     constexpr bool BB = decltype(btag)::value;
@@ -699,7 +706,7 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor qkv, int64_t num_heads,
                        mask_ptr, (__hip_bfloat16*)out.data_ptr(),
                        save_p ? (__hip_bfloat16*)p.data_ptr() : nullptr,
                        save_stats ? stats.data_ptr<float>() : nullptr,
-                       nullptr, B, N, H, n_win, (float)scale);
+                       nullptr, B, N, H, n_win, (float)scale, ablate);
   };
   auto d3 = [&](auto dtag) {
     using T = std::true_type;
