@@ -100,9 +100,9 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ qkv,  // [B,N,3,H,D]
   __hip_bfloat16* k_lds = (__hip_bfloat16*)lds_raw;       // [Npad][KPAD]
   __hip_bfloat16* vt_lds = k_lds + Npad * KPAD;           // [D][VROW]
   const int VROW = Npad + 8;
-  __hip_bfloat16* p_lds = vt_lds + D * VROW;              // [nwaves][PBUF]
+  __hip_bfloat16* p_lds = vt_lds + D * VROW;              // [nwaves][2*PBUF]
 
-  float* kinv = (float*)(p_lds + nwaves * PBUF);  // [Npad] (COSINE only)
+  float* kinv = (float*)(p_lds + nwaves * 2 * PBUF);  // [Npad] (COSINE only)
 
   const int64_t bh_stride = (int64_t)3 * H * D;
   stage_rows<D>(qkv + ((int64_t)b * N * 3 + 1) * H * D + (int64_t)h * D,
@@ -126,7 +126,7 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ qkv,  // [B,N,3,H,D]
 
   const __hip_bfloat16* q_src = qkv + (int64_t)b * N * 3 * H * D +
                                 (int64_t)h * D;
-  __hip_bfloat16* p_buf = p_lds + wave * PBUF;
+  __hip_bfloat16* p_buf = p_lds + wave * 2 * PBUF;  // double-buffered
   const int n_qblocks = (N + 15) / 16;
 
   for (int qb = wave; qb < n_qblocks; qb += nwaves) {
@@ -183,17 +183,23 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ qkv,  // [B,N,3,H,D]
     }
 
     if (ablate >= 3) continue;  // ablation: QK^T only
-    // softmax rows (C/D: col = key = lane&15, row = (lane>>4)*4 + reg)
+    // softmax rows (C/D: col = key = lane&15, row = (lane>>4)*4 + reg).
+    // kt-outer loops keep the 4 rows' max/exp/sum chains INDEPENDENT so
+    // they interleave (the row-outer form serialized a 16-deep fmax chain,
+    // 4 dependent shuffle reductions and a 16-deep exp+add chain per row:
+    // ablation measured the whole softmax phase at 95 us of a 336 us
+    // kernel before this restructure).
     const int col = lane & 15;
-    float row_sum[4];
+    float row_sum[4], mx[4];
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      float m = -INFINITY;
-      const int qrow = q0 + (lane >> 4) * 4 + r;
+    for (int r = 0; r < 4; ++r) mx[r] = -INFINITY;
 #pragma unroll
-      for (int kt = 0; kt < 16; ++kt) {
-        if (kt < n_ktiles) {
-          const int key = kt * 16 + col;
+    for (int kt = 0; kt < 16; ++kt) {
+      if (kt < n_ktiles) {
+        const int key = kt * 16 + col;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int qrow = q0 + (lane >> 4) * 4 + r;
           float sv = COSINE
               ? s_acc[kt][r] * ls * qn_sh[wave][(lane >> 4) * 4 + r] *
                     kinv[key]
@@ -205,23 +211,41 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ qkv,  // [B,N,3,H,D]
           }
           if (key >= N) sv = -INFINITY;
           s_acc[kt][r] = sv;
-          m = fmaxf(m, sv);
+          mx[r] = fmaxf(mx[r], sv);
         }
       }
-      m = wave16_max(m);
-      float sum = 0.f;
+    }
 #pragma unroll
-      for (int kt = 0; kt < 16; ++kt) {
-        if (kt < n_ktiles) {
-          const float p = __expf(s_acc[kt][r] - m);
+    for (int off = 8; off > 0; off >>= 1)
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        mx[r] = fmaxf(mx[r], __shfl_xor(mx[r], off, 64));
+#pragma unroll
+    for (int r = 0; r < 4; ++r) row_sum[r] = 0.f;
+#pragma unroll
+    for (int kt = 0; kt < 16; ++kt) {
+      if (kt < n_ktiles) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const float p = __expf(s_acc[kt][r] - mx[r]);
           s_acc[kt][r] = p;
-          sum += p;
+          row_sum[r] += p;
         }
       }
-      row_sum[r] = wave16_sum(sum);
-      if (stats != nullptr && col == 0 && qrow < N) {
-        stats[((int64_t)b * H + h) * N + qrow] = m;
-        stats[((int64_t)(B + b) * H + h) * N + qrow] = row_sum[r];
+    }
+#pragma unroll
+    for (int off = 8; off > 0; off >>= 1)
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        row_sum[r] += __shfl_xor(row_sum[r], off, 64);
+    if (stats != nullptr && col == 0) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int qrow = q0 + (lane >> 4) * 4 + r;
+        if (qrow < N) {
+          stats[((int64_t)b * H + h) * N + qrow] = mx[r];
+          stats[((int64_t)(B + b) * H + h) * N + qrow] = row_sum[r];
+        }
       }
     }
     if (SAVE_P) {
@@ -241,24 +265,35 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ qkv,  // [B,N,3,H,D]
     }
 
     if (ablate >= 2) continue;  // ablation: no PV / p_buf
-    // O = P V, accumulated chunk by chunk through the small P buffer
+    // O = P V through the per-wave P buffer, software-pipelined over two
+    // buffers: chunk c+1's LDS writes are issued BEFORE chunk c's fragment
+    // reads, so the write->read drain of a chunk hides under the previous
+    // chunk's MFMAs (ablation: the serialized form cost 67 us of 336).
     f32x4 o_acc[D / 16];
 #pragma unroll
     for (int dt = 0; dt < D / 16; ++dt) o_acc[dt] = {0.f, 0.f, 0.f, 0.f};
+    auto write_pchunk = [&](int c) {
+      const int t0 = c * 4;
+      const int nt = n_ktiles - t0 < 4 ? n_ktiles - t0 : 4;
+      __hip_bfloat16* pb = p_buf + (c & 1) * PBUF;
+#pragma unroll
+      for (int tt = 0; tt < 4; ++tt) {
+        if (tt < nt) {
+#pragma unroll
+          for (int r = 0; r < 4; ++r)
+            pb[((lane >> 4) * 4 + r) * KPAD + tt * 16 + col] =
+                __hip_bfloat16(s_acc[tt + (c << 2)][r]);
+        }
+      }
+    };
+    write_pchunk(0);
 #pragma unroll
     for (int c = 0; c < 4; ++c) {  // chunks of 4 key tiles (64 keys)
       const int t0 = c * 4;
       if (t0 < n_ktiles) {
         const int nt = n_ktiles - t0 < 4 ? n_ktiles - t0 : 4;
-#pragma unroll
-        for (int tt = 0; tt < 4; ++tt) {
-          if (tt < nt) {
-#pragma unroll
-            for (int r = 0; r < 4; ++r)
-              p_buf[((lane >> 4) * 4 + r) * KPAD + tt * 16 + col] =
-                  __hip_bfloat16(s_acc[t0 + tt][r]);
-          }
-        }
+        if ((c + 1) * 4 < n_ktiles) write_pchunk(c + 1);
+        const __hip_bfloat16* pb = p_buf + (c & 1) * PBUF;
         // MFMA over the chunk: pairs of tiles = 32-key K-steps
 #pragma unroll
         for (int kk = 0; kk < 2; ++kk) {
@@ -267,7 +302,7 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ qkv,  // [B,N,3,H,D]
             const bool valid = k0 < nt * 16;
             bf16x8 p_frag{};
             if (valid)
-              p_frag = *(const bf16x8*)(&p_buf[(lane & 15) * KPAD + k0]);
+              p_frag = *(const bf16x8*)(&pb[(lane & 15) * KPAD + k0]);
 #pragma unroll
             for (int dt = 0; dt < D / 16; ++dt) {
               const int d = dt * 16 + (lane & 15);
@@ -283,15 +318,38 @@ void attn_fwd_kernel(const __hip_bfloat16* __restrict__ qkv,  // [B,N,3,H,D]
     }
 
     if (ablate == 1) continue;  // ablation: no O store
+    // O store through an LDS transpose (p_buf reused as scratch): the
+    // fragment layout is column-per-lane, so direct stores were 16 scalar
+    // 2-byte globals per lane (ablation: 64 us of 336); the transpose
+    // turns them into coalesced 16-byte row stores.
+    {
+      float rinv[4];
 #pragma unroll
-    for (int dt = 0; dt < D / 16; ++dt) {
+      for (int r = 0; r < 4; ++r) rinv[r] = 1.f / row_sum[r];
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int qrow = q0 + (lane >> 4) * 4 + r;
-        if (qrow >= N) continue;
-        const int d = dt * 16 + (lane & 15);
-        out[((int64_t)b * N + qrow) * H * D + h * D + d] =
-            __hip_bfloat16(o_acc[dt][r] / row_sum[r]);
+      for (int dt = 0; dt < D / 16; ++dt)
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          p_buf[((lane >> 4) * 4 + r) * KPAD + dt * 16 + (lane & 15)] =
+              __hip_bfloat16(o_acc[dt][r] * rinv[r]);
+      // same-wave cross-lane LDS visibility: drain, and fence the reads
+      // below from being hoisted above the drain (guide rule 18)
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+      for (int i = 0; i < (16 * D / 8) / 64; ++i) {
+        const int t = lane + 64 * i;
+        const int row = t / (D / 8);
+        const int c8 = t % (D / 8);
+        const int qrow = q0 + row;
+        if (qrow < N) {
+          Vec<__hip_bfloat16, 8> v =
+              vload<__hip_bfloat16, 8>(&p_buf[row * KPAD + c8 * 8]);
+          vstore<__hip_bfloat16, 8>(
+              (__hip_bfloat16*)out + ((int64_t)b * N + qrow) * H * D +
+                  h * D + c8 * 8,
+              v);
+        }
       }
     }
   }
@@ -689,7 +747,7 @@ std::vector<torch::Tensor> attn_fwd(torch::Tensor qkv, int64_t num_heads,
 
   const int Npad = (N + 15) & ~15;
   const int nwaves = N > 96 ? 8 : 4;  // small-N: 4 waves cover all q blocks
-  const int lds = (Npad * dla::KPAD + D * (Npad + 8) + nwaves * dla::PBUF) *
+  const int lds = (Npad * dla::KPAD + D * (Npad + 8) + nwaves * 2 * dla::PBUF) *
                   (int)sizeof(__hip_bfloat16);
   dim3 grid(B * H), block(nwaves * 64);
 
@@ -820,7 +878,7 @@ std::vector<torch::Tensor> attn_fwd_cosine(torch::Tensor qkv,
   }
   const int Npad = (N + 15) & ~15;
   const int nwaves = N > 96 ? 8 : 4;
-  const int lds = (Npad * dla::KPAD + D * (Npad + 8) + nwaves * dla::PBUF) *
+  const int lds = (Npad * dla::KPAD + D * (Npad + 8) + nwaves * 2 * dla::PBUF) *
                       (int)sizeof(__hip_bfloat16) +
                   Npad * (int)sizeof(float);  // + kinv
   dim3 grid(B * H), block(nwaves * 64);
