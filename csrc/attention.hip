@@ -519,19 +519,37 @@ void attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ qkv,
       }
     }
 
+    // dK/dV stores through an LDS transpose (p_buf scratch): fragment
+    // layout is column-per-lane, direct stores were 32 scalar 2B globals
+    auto store_t = [&](bool is_v) {
 #pragma unroll
-    for (int dt = 0; dt < D / 16; ++dt) {
+      for (int dt = 0; dt < D / 16; ++dt)
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int key = key0 + (lane >> 4) * 4 + r;
-        if (key >= N) continue;
-        const int d = dt * 16 + (lane & 15);
-        const int64_t base = ((int64_t)(b * N + key) * 3) * H * D +
-                             (int64_t)h * D + d;
-        dqkv[base + (int64_t)H * D] = __hip_bfloat16(dk_acc[dt][r]);
-        dqkv[base + (int64_t)2 * H * D] = __hip_bfloat16(dv_acc[dt][r]);
+        for (int r = 0; r < 4; ++r)
+          p_buf[((lane >> 4) * 4 + r) * KPAD + dt * 16 + (lane & 15)] =
+              __hip_bfloat16(is_v ? dv_acc[dt][r] : dk_acc[dt][r]);
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_sched_barrier(0);
+      __hip_bfloat16* dst = dqkv + (int64_t)(is_v ? 2 : 1) * H * D +
+                            (int64_t)h * D;
+#pragma unroll
+      for (int i = 0; i < (16 * D / 8) / 64; ++i) {
+        const int t = lane + 64 * i;
+        const int row = t / (D / 8);
+        const int c8 = t % (D / 8);
+        const int key = key0 + row;
+        if (key < N) {
+          Vec<__hip_bfloat16, 8> v =
+              vload<__hip_bfloat16, 8>(&p_buf[row * KPAD + c8 * 8]);
+          vstore<__hip_bfloat16, 8>(
+              dst + ((int64_t)(b * N + key) * 3) * H * D + c8 * 8, v);
+        }
       }
-    }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_sched_barrier(0);
+    };
+    store_t(false);
+    store_t(true);
   }
 }
 
@@ -665,17 +683,32 @@ void attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ qkv,
       }
     }
 
+    // dQ store through an LDS transpose (p_buf scratch): coalesced 16B rows
 #pragma unroll
-    for (int dt = 0; dt < D / 16; ++dt) {
+    for (int dt = 0; dt < D / 16; ++dt)
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int qrow = q0 + (lane >> 4) * 4 + r;
-        if (qrow >= N) continue;
-        const int d = dt * 16 + (lane & 15);
-        dqkv[((int64_t)(b * N + qrow) * 3) * H * D + (int64_t)h * D + d] =
+      for (int r = 0; r < 4; ++r)
+        p_buf[((lane >> 4) * 4 + r) * KPAD + dt * 16 + (lane & 15)] =
             __hip_bfloat16(dq_acc[dt][r]);
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_sched_barrier(0);
+#pragma unroll
+    for (int i = 0; i < (16 * D / 8) / 64; ++i) {
+      const int t = lane + 64 * i;
+      const int row = t / (D / 8);
+      const int c8 = t % (D / 8);
+      const int qrow = q0 + row;
+      if (qrow < N) {
+        Vec<__hip_bfloat16, 8> v =
+            vload<__hip_bfloat16, 8>(&p_buf[row * KPAD + c8 * 8]);
+        vstore<__hip_bfloat16, 8>(
+            dqkv + ((int64_t)(b * N + qrow) * 3) * H * D + (int64_t)h * D +
+                c8 * 8,
+            v);
       }
     }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_sched_barrier(0);
   }
 }
 
