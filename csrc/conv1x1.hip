@@ -64,8 +64,11 @@ __global__ __launch_bounds__((BN == 256) ? 512 : 256) void conv1x1_nt_kernel(
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  const int64_t m0 = (int64_t)blockIdx.x * BM;
-  const int n0 = blockIdx.y * BN;
+  // grid (gy=n, gx=m) with n fastest: the N/BN blocks of one M-strip are
+  // dispatched together, so they share the strip's A tile through L2/L3
+  // instead of re-fetching it a full grid-pass later
+  const int64_t m0 = (int64_t)blockIdx.y * BM;
+  const int n0 = blockIdx.x * BN;
 
   extern __shared__ __attribute__((aligned(16))) char lds_raw[];
   bf16* const lds16 = (bf16*)lds_raw;
@@ -162,7 +165,7 @@ __global__ __launch_bounds__((BN == 256) ? 512 : 256) void conv1x1_nt_kernel(
   // its 4*MFR values per ni are rows of that channel. Padded rows (gm >= M)
   // must not contribute.
   if (STATS) {
-    // per-block partial sums -> slab row sums[blockIdx.x * 2N + ...], NO
+    // per-block partial sums -> slab row sums[m-block * 2N + ...], NO
     // global atomics (a 6k-deep atomic chain per channel measured 6x the
     // whole GEMM); the [2C] reduction happens in one torch sum downstream.
     float* sbuf = (float*)lds_raw;  // [WAVES_M][2*BN] (staging LDS is free)
@@ -194,7 +197,7 @@ __global__ __launch_bounds__((BN == 256) ? 512 : 256) void conv1x1_nt_kernel(
       }
     }
     __syncthreads();
-    float* slab = sums + (int64_t)blockIdx.x * (2 * N);
+    float* slab = sums + (int64_t)blockIdx.y * (2 * N);  // m-block row
     for (int t = threadIdx.x; t < 2 * BN; t += NWAVES * 64) {
       float tot = 0.f;
 #pragma unroll
@@ -437,7 +440,7 @@ std::vector<torch::Tensor> conv1x1_fwd(torch::Tensor a, torch::Tensor b,
         done[ST] = true;
       }
     }
-    hipLaunchKernelGGL((dla::conv1x1_nt_kernel<BN, ST>), dim3(gx, N / BN),
+    hipLaunchKernelGGL((dla::conv1x1_nt_kernel<BN, ST>), dim3(N / BN, gx),
                        dim3(NW * 64), lds, dla::stream(),
                        (const dla::bf16*)a.data_ptr(),
                        (const dla::bf16*)b.data_ptr(),
