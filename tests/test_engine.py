@@ -252,3 +252,20 @@ def test_scheduler_resume_matches_fresh():
         s2.step()
     assert o1.param_groups[0]["lr"] == pytest.approx(
         o2.param_groups[0]["lr"], rel=1e-9)
+
+
+def test_loggers_facade(tmp_path):
+    """yolov5-style Loggers fan-out (csv + tb/jsonl + wandb-if-present),
+    ref detection/yolov5/utils/loggers/__init__.py:17-55."""
+    from deeplearning_amd.engine.loggers import Loggers
+
+    lg = Loggers(tmp_path, use_wandb=True)  # wandb absent -> disabled
+    assert lg.wandb is None
+    for step in range(3):
+        lg.log_metrics({"train/loss": 1.0 / (step + 1), "lr": 0.1}, step)
+    lg.close()
+    csv_file = tmp_path / "results.csv"
+    assert csv_file.exists()
+    rows = csv_file.read_text().strip().splitlines()
+    assert rows[0].split(",") == ["step", "lr", "train/loss"]
+    assert len(rows) == 4
