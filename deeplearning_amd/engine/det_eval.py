@@ -106,8 +106,13 @@ class DetEvaluator:
         self.items = []  # (cls, matched [T,D], ignored [T,D], scores [D], n_gt)
 
     def update(self, predictions, targets):
-        """predictions/targets: lists of dicts with boxes/labels(/scores)."""
+        """predictions/targets: lists of dicts with boxes/labels(/scores);
+        preds may live on GPU while targets stayed host-side — everything is
+        matched on CPU."""
         for pred, gt in zip(predictions, targets):
+            pred = {k: v.detach().cpu() for k, v in pred.items()}
+            gt = {k: (v.cpu() if torch.is_tensor(v) else v)
+                  for k, v in gt.items()}
             classes = torch.cat([pred["labels"], gt["labels"]]).unique()
             crowd = gt.get("iscrowd",
                            torch.zeros_like(gt["labels"]))

@@ -230,13 +230,17 @@ def simota_assign(pred_boxes, pred_cls, pred_obj, gt_boxes, gt_labels,
     ious = box_iou(gt_boxes, xyxy)  # G, C
     iou_cost = -torch.log(ious + 1e-8)
 
-    cls_prob = (pred_cls[candidate].sigmoid() *
-                pred_obj[candidate].sigmoid()[:, None]).sqrt()
-    gt_onehot = F.one_hot(gt_labels, num_classes).float()  # G, nc
-    cls_cost = F.binary_cross_entropy(
-        cls_prob[None].expand(G, -1, -1),
-        gt_onehot[:, None].expand(-1, cls_prob.shape[0], -1),
-        reduction="none").sum(-1)
+    # fp32 + autocast off: BCE-on-probabilities is rejected inside autocast
+    # (the reference computes this cost block with amp disabled too,
+    # yolo_head.py get_assignments)
+    with torch.autocast(pred_cls.device.type, enabled=False):
+        cls_prob = (pred_cls[candidate].float().sigmoid() *
+                    pred_obj[candidate].float().sigmoid()[:, None]).sqrt()
+        gt_onehot = F.one_hot(gt_labels, num_classes).float()  # G, nc
+        cls_cost = F.binary_cross_entropy(
+            cls_prob[None].expand(G, -1, -1),
+            gt_onehot[:, None].expand(-1, cls_prob.shape[0], -1),
+            reduction="none").sum(-1)
     cost = cls_cost + 3.0 * iou_cost + 100000.0 * (~both.T)
 
     # dynamic-k: top-10 IoU sum per gt
