@@ -305,3 +305,21 @@ def test_bucketed_ddp_single_rank_rccl():
             torch.testing.assert_close(p, q, atol=1e-5, rtol=1e-5)
     finally:
         dist.destroy_process_group()
+
+
+def test_gemm_fwd_shape_fuzz():
+    """Randomized shape sweep over the kernel's domain (M arbitrary,
+    K,N % 64): catches tile-edge and path-selection regressions."""
+    import random
+
+    rng = random.Random(123)
+    for _ in range(10):
+        m = rng.randint(1, 2000)
+        k = 64 * rng.randint(1, 8)
+        n = 64 * rng.randint(1, 8)
+        a = _rand2d(m, k, m + k)
+        b = _rand2d(n, k, n + k)
+        y, _ = ext().conv1x1_fwd(a, b, None, None, None, None, False, False)
+        ref = a.float() @ b.float().t()
+        err = (y.float() - ref).abs().max() / ref.abs().max().clamp(min=1)
+        assert err < 3e-2, (m, k, n, float(err))
