@@ -28,6 +28,14 @@ if which == "attn":
         ext().attn_fwd(qkv, 12, 0.125)
     torch.cuda.synchronize()
     print(f"attn_fwd {(time.perf_counter() - t0) / iters * 1e3:.3f} ms")
+elif which == "fwd_fat":
+    m, k, n = 200704, 128, 512
+    a = torch.randn(m, k, device="cuda").to(torch.bfloat16)
+    w = torch.randn(n, k, device="cuda").to(torch.bfloat16)
+    torch.cuda.synchronize()
+    for _ in range(iters):
+        ext().conv1x1_fwd(a, w, None, None, None, None, False, False)
+    torch.cuda.synchronize()
 else:
     m, k, n = 802816, 64, 256
     a = torch.randn(m, k, device="cuda").to(torch.bfloat16)
