@@ -22,6 +22,8 @@
 // wgrad: dW[N][K] = sum_m dy[m][n] * x[m][k] — separate kernel contracting
 // over M with transposed LDS images (scatter ds_write staging, m-contiguous
 // fragment reads), fp32 atomicAdd accumulation over M-split blocks.
+#include <cstdlib>
+
 #include "common.h"
 #include "vec.h"
 
@@ -44,7 +46,7 @@ __device__ __forceinline__ void glds16(const bf16* gsrc, bf16* lds_dst) {
 // conv1x1 forward / dgrad kernel: C[M,N] = A[M,K] @ B[N,K]^T (bf16 in/out,
 // fp32 accumulate). K % 64 == 0, N % BN == 0.
 // ---------------------------------------------------------------------------
-template <int BN, bool STATS>
+template <int BN, bool STATS, int BK = 64>
 __global__ __launch_bounds__((BN == 256) ? 512 : 256) void conv1x1_nt_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
     bf16* __restrict__ C, const bf16* __restrict__ residual,  // [M,N] | null
@@ -53,7 +55,7 @@ __global__ __launch_bounds__((BN == 256) ? 512 : 256) void conv1x1_nt_kernel(
     const float* __restrict__ shift,                          // [N] | null
     float* __restrict__ sums,  // [gridDim.x, 2N] per-block partials | null
     int64_t M, int K, int N, bool relu) {
-  constexpr int BM = 128, BK = 64;
+  constexpr int BM = 128;
   constexpr int NWAVES = (BN == 256) ? 8 : 4;
   constexpr int WAVES_M = (BN == 64) ? 4 : 2;
   constexpr int WAVES_N = NWAVES / WAVES_M;  // 4 / 2 / 1
@@ -85,29 +87,37 @@ __global__ __launch_bounds__((BN == 256) ? 512 : 256) void conv1x1_nt_kernel(
   // stage one K-step of A[BM][BK] and B[BN][BK] into buffer `buf`.
   // LDS image is row-major with the read-side XOR swizzle baked into the
   // SOURCE address (rule 21: linear dest + inverse-swizzled source).
+  // A 1KB piece covers 1024/ROWB rows; swizzle granularity shrinks with the
+  // row size (BK=64: byte ^= (row&7)<<4 over 16B slots; BK=32: (row&3)<<4).
+  constexpr int RPP = 1024 / (BK * 2);       // rows per 1KB piece (8 or 16)
+  constexpr int LPR = 64 / RPP;              // lanes per row (8 or 4)
+  // BK=64 (128B rows): byte ^= (row&7)<<4 spreads a 16-lane column read
+  // over 8 slots. BK=32 (64B rows): rows alias every 4, so fold row>>2 in:
+  // ((row ^ row>>2)&3)<<4 gives 4 distinct slots per alias class.
+  auto swz = [&](int row) {
+    return BK == 64 ? (row & 7) << 4 : ((row ^ (row >> 2)) & 3) << 4;
+  };
   auto stage = [&](int buf, int kt) {
     const int k0b = kt * BK * 2;  // byte offset into a row of A/B
-    // A: 16 pieces of 1KB (8 rows x 128B)
     if (!direct_a) {
 #pragma unroll
-      for (int pp = 0; pp < 16 / NWAVES; ++pp) {
+      for (int pp = 0; pp < (BM / RPP) / NWAVES; ++pp) {
         const int p = wave + pp * NWAVES;
-        const int row = p * 8 + (lane >> 3);
-        const int b = (lane & 7) * 16;
+        const int row = p * RPP + lane / LPR;
+        const int b = (lane % LPR) * 16;
         const int64_t rg = m0 + row < M ? m0 + row : M - 1;
         const char* src = (const char*)A + rg * (int64_t)K * 2 + k0b +
-                          (b ^ ((row & 7) << 4));
+                          (b ^ swz(row));
         glds16((const bf16*)src, a_lds(buf) + p * 512);
       }
     }
-    // B: BN/8 pieces
 #pragma unroll
-    for (int pp = 0; pp < (BN / 8) / NWAVES; ++pp) {
+    for (int pp = 0; pp < (BN / RPP) / NWAVES; ++pp) {
       const int p = wave + pp * NWAVES;
-      const int row = p * 8 + (lane >> 3);
-      const int b = (lane & 7) * 16;
+      const int row = p * RPP + lane / LPR;
+      const int b = (lane % LPR) * 16;
       const char* src = (const char*)B + (int64_t)(n0 + row) * K * 2 + k0b +
-                        (b ^ ((row & 7) << 4));
+                        (b ^ swz(row));
       glds16((const bf16*)src, b_lds(buf) + p * 512);
     }
   };
@@ -129,7 +139,7 @@ __global__ __launch_bounds__((BN == 256) ? 512 : 256) void conv1x1_nt_kernel(
     const char* ab = (const char*)a_lds(cur);
     const char* bb = (const char*)b_lds(cur);
 #pragma unroll
-    for (int ks = 0; ks < 2; ++ks) {
+    for (int ks = 0; ks < BK / 32; ++ks) {
       const int kbyte = (ks * 32 + (lane >> 4) * 8) * 2;
       bf16x8 af[MFR], bfr[NFR];
 #pragma unroll
@@ -141,13 +151,13 @@ __global__ __launch_bounds__((BN == 256) ? 512 : 256) void conv1x1_nt_kernel(
                                     kbyte);
         } else {
           af[mi] =
-              *(const bf16x8*)(ab + row * ROWB + (kbyte ^ ((row & 7) << 4)));
+              *(const bf16x8*)(ab + row * ROWB + (kbyte ^ swz(row)));
         }
       }
 #pragma unroll
       for (int ni = 0; ni < NFR; ++ni) {
         const int row = wn_off + ni * 16 + (lane & 15);
-        bfr[ni] = *(const bf16x8*)(bb + row * ROWB + (kbyte ^ ((row & 7) << 4)));
+        bfr[ni] = *(const bf16x8*)(bb + row * ROWB + (kbyte ^ swz(row)));
       }
 #pragma unroll
       for (int mi = 0; mi < MFR; ++mi)
@@ -418,15 +428,16 @@ std::vector<torch::Tensor> conv1x1_fwd(torch::Tensor a, torch::Tensor b,
   float* sums_p = want_stats ? sums.data_ptr<float>() : nullptr;
 
   const int nk = K / 64;
-  auto launch = [&](auto bntag, auto stag) {
+  auto launch = [&](auto bntag, auto stag, auto bktag) {
     constexpr int BN = decltype(bntag)::value;
     constexpr bool ST = decltype(stag)::value;
+    constexpr int BKT = decltype(bktag)::value;
     constexpr int NW = (BN == 256) ? 8 : 4;
     constexpr int WM = (BN == 64) ? 32 : 64;
     // single staging buffer suffices when there is only one K-step; with one
     // K-step A skips LDS entirely (direct-global fragments)
-    const int nbuf = nk > 1 ? 2 : 1;
-    const int lds_stage = (128 * 64 + BN * 64) * 2 * nbuf;
+    const int nbuf = K / BKT > 1 ? 2 : 1;
+    const int lds_stage = (128 * BKT + BN * BKT) * 2 * nbuf;
     // chunked transpose scratch: half a wave-tile when MFR >= 4
     const int lds_ep = NW * (WM >= 64 ? WM / 2 : WM) * (64 + 8) * 2;
     const int lds_st = ST ? ((BN == 64) ? 4 : 2) * 2 * BN * 4 : 0;
@@ -435,21 +446,35 @@ std::vector<torch::Tensor> conv1x1_fwd(torch::Tensor a, torch::Tensor b,
       static bool done[2] = {false, false};
       if (!done[ST]) {
         (void)hipFuncSetAttribute(
-            (const void*)&dla::conv1x1_nt_kernel<BN, ST>,
+            (const void*)&dla::conv1x1_nt_kernel<BN, ST, BKT>,
             hipFuncAttributeMaxDynamicSharedMemorySize, 163840);
         done[ST] = true;
       }
     }
-    hipLaunchKernelGGL((dla::conv1x1_nt_kernel<BN, ST>), dim3(N / BN, gx),
+    hipLaunchKernelGGL((dla::conv1x1_nt_kernel<BN, ST, BKT>), dim3(N / BN, gx),
                        dim3(NW * 64), lds, dla::stream(),
                        (const dla::bf16*)a.data_ptr(),
                        (const dla::bf16*)b.data_ptr(),
                        (dla::bf16*)C.data_ptr(), res_p, bias_p, scale_p,
                        shift_p, sums_p, M, K, N, relu);
   };
+  // BK=32 experiment knob for the latency-bound fat-N deep-ish K shapes
+  // (halves the staging LDS -> 4 blocks/CU instead of 2)
+  static const bool bk32 = []{
+    const char* e = std::getenv("DLA_C1X1_BK32");
+    return e != nullptr && e[0] == '1';
+  }();
   auto pick = [&](auto bntag) {
-    if (want_stats) launch(bntag, std::true_type{});
-    else launch(bntag, std::false_type{});
+    constexpr int BNv = decltype(bntag)::value;
+    if (BNv == 128 && bk32 && nk >= 2) {
+      if (want_stats) launch(bntag, std::true_type{},
+                             std::integral_constant<int, 32>{});
+      else launch(bntag, std::false_type{}, std::integral_constant<int, 32>{});
+      return;
+    }
+    if (want_stats) launch(bntag, std::true_type{},
+                           std::integral_constant<int, 64>{});
+    else launch(bntag, std::false_type{}, std::integral_constant<int, 64>{});
   };
   // BN=256 (512 threads, single A pass) wins only at nk==1 where the LDS
   // fits 3 blocks/CU; at nk>=2 its 96KB double-buffer drops occupancy to
