@@ -458,12 +458,14 @@ std::vector<torch::Tensor> conv1x1_fwd(torch::Tensor a, torch::Tensor b,
                        (dla::bf16*)C.data_ptr(), res_p, bias_p, scale_p,
                        shift_p, sums_p, M, K, N, relu);
   };
-  // BK=32 experiment knob for the latency-bound fat-N deep-ish K shapes
-  // (halves the staging LDS -> 4 blocks/CU instead of 2)
-  static const bool bk32 = []{
+  // BK=32 halves the staging LDS (4 blocks/CU): measured +5% at K=128
+  // (nk==2, the exposed-drain case), neutral/negative deeper — default it
+  // only there (DLA_C1X1_BK32=1 forces it for all BN=128 shapes).
+  static const bool bk32_force = []{
     const char* e = std::getenv("DLA_C1X1_BK32");
     return e != nullptr && e[0] == '1';
   }();
+  const bool bk32 = bk32_force || nk == 2;
   auto pick = [&](auto bntag) {
     constexpr int BNv = decltype(bntag)::value;
     if (BNv == 128 && bk32 && nk >= 2) {
