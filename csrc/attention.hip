@@ -385,7 +385,7 @@ void attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ qkv,
   const int VROW = Npad + 8;
   __hip_bfloat16* qt_lds = do_lds + Npad * KPAD;          // [D][VROW]
   __hip_bfloat16* dot_lds = qt_lds + D * VROW;            // [D][VROW]
-  __hip_bfloat16* p_lds = dot_lds + D * VROW;             // [nwaves][PBUF]
+  __hip_bfloat16* p_lds = dot_lds + D * VROW;             // [nwaves][2*PBUF]
 
   const int64_t bh_stride = (int64_t)3 * H * D;
   stage_rows<D>(qkv + (int64_t)b * N * 3 * H * D + (int64_t)h * D,
@@ -401,7 +401,7 @@ void attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ qkv,
                                 (int64_t)h * D;
   const __hip_bfloat16* v_src = qkv + ((int64_t)b * N * 3 + 2) * H * D +
                                 (int64_t)h * D;
-  __hip_bfloat16* p_buf = p_lds + wave * PBUF;
+  __hip_bfloat16* p_buf = p_lds + wave * 2 * PBUF;  // P^T | dS^T split buffers
 
   for (int kt = wave; kt < (N + 15) / 16; kt += nwaves) {
     const int key0 = kt * 16;
@@ -466,6 +466,12 @@ void attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ qkv,
             for (int r = 0; r < 4; ++r)
               p_buf[((lane >> 4) * 4 + r) * KPAD + tt * 16 + (lane & 15)] =
                   __hip_bfloat16(st[tt][r]);
+            // dS^T staged to the SECOND buffer up-front: no WAR wait
+            // between the dV reads of P^T and the dS^T writes
+#pragma unroll
+            for (int r = 0; r < 4; ++r)
+              p_buf[PBUF + ((lane >> 4) * 4 + r) * KPAD + tt * 16 +
+                    (lane & 15)] = __hip_bfloat16(dpt[tt][r]);
           }
         }
 #pragma unroll
@@ -487,16 +493,7 @@ void attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ qkv,
             }
           }
         }
-        // stage dS^T, accumulate dK
-#pragma unroll
-        for (int tt = 0; tt < 4; ++tt) {
-          if (tt < nt) {
-#pragma unroll
-            for (int r = 0; r < 4; ++r)
-              p_buf[((lane >> 4) * 4 + r) * KPAD + tt * 16 + (lane & 15)] =
-                  __hip_bfloat16(dpt[tt][r]);
-          }
-        }
+        // accumulate dK from the second buffer
 #pragma unroll
         for (int kk = 0; kk < 2; ++kk) {
           if (kk * 32 < nt * 16) {
@@ -504,7 +501,8 @@ void attn_bwd_kv_kernel(const __hip_bfloat16* __restrict__ qkv,
             const bool valid = k0 < nt * 16;
             bf16x8 ds_frag{};
             if (valid)
-              ds_frag = *(const bf16x8*)(&p_buf[(lane & 15) * KPAD + k0]);
+              ds_frag =
+                  *(const bf16x8*)(&p_buf[PBUF + (lane & 15) * KPAD + k0]);
 #pragma unroll
             for (int dt = 0; dt < D / 16; ++dt) {
               const int d = dt * 16 + (lane & 15);
@@ -835,7 +833,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor qkv, torch::Tensor dout,
   const int nwaves = N > 96 ? 8 : 4;
   dim3 grid(B * H), block(nwaves * 64);
   const int lds_kv = (2 * Npad * dla::KPAD + 2 * D * (Npad + 8) +
-                      nwaves * dla::PBUF) * 2;
+                      nwaves * 2 * dla::PBUF) * 2;
   const int lds_q = (2 * Npad * dla::KPAD + D * (Npad + 8) +
                      nwaves * dla::PBUF) * 2;
   auto run = [&](auto dtag) {
