@@ -591,7 +591,7 @@ void attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ qkv,
                                 (int64_t)h * D;
   const __hip_bfloat16* do_src = dout + (int64_t)b * N * H * D +
                                  (int64_t)h * D;
-  __hip_bfloat16* p_buf = p_lds + wave * PBUF;
+  __hip_bfloat16* p_buf = p_lds + wave * 2 * PBUF;  // chunk-alternating
 
   for (int qb = wave; qb < (N + 15) / 16; qb += nwaves) {
     const int q0 = qb * 16;
@@ -655,8 +655,9 @@ void attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ qkv,
             }
 #pragma unroll
             for (int r = 0; r < 4; ++r)
-              p_buf[((lane >> 4) * 4 + r) * KPAD + tt * 16 + (lane & 15)] =
-                  __hip_bfloat16(st[tt][r]);
+              p_buf[(c & 1) * PBUF +
+                    ((lane >> 4) * 4 + r) * KPAD + tt * 16 + (lane & 15)] =
+                  __hip_bfloat16(st[tt][r]);  // alternate buffers: no WAR
           }
         }
 #pragma unroll
@@ -666,7 +667,8 @@ void attn_bwd_q_kernel(const __hip_bfloat16* __restrict__ qkv,
             const bool valid = k0 < nt * 16;
             bf16x8 ds_frag{};
             if (valid)
-              ds_frag = *(const bf16x8*)(&p_buf[(lane & 15) * KPAD + k0]);
+              ds_frag = *(const bf16x8*)(&p_buf[(c & 1) * PBUF +
+                                                (lane & 15) * KPAD + k0]);
 #pragma unroll
             for (int dt = 0; dt < D / 16; ++dt) {
               const int d = dt * 16 + (lane & 15);
@@ -835,7 +837,7 @@ std::vector<torch::Tensor> attn_bwd(torch::Tensor qkv, torch::Tensor dout,
   const int lds_kv = (2 * Npad * dla::KPAD + 2 * D * (Npad + 8) +
                       nwaves * 2 * dla::PBUF) * 2;
   const int lds_q = (2 * Npad * dla::KPAD + D * (Npad + 8) +
-                     nwaves * dla::PBUF) * 2;
+                     nwaves * 2 * dla::PBUF) * 2;
   auto run = [&](auto dtag) {
     constexpr int DD = decltype(dtag)::value;
     hipLaunchKernelGGL((dla::attn_bwd_kv_kernel<DD>), grid, block, lds_kv,
