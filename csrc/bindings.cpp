@@ -39,6 +39,13 @@ std::vector<torch::Tensor> conv1x1_fwd(torch::Tensor a, torch::Tensor b,
                                        c10::optional<torch::Tensor> residual,
                                        bool relu, bool want_stats);
 torch::Tensor conv1x1_wgrad(torch::Tensor dy, torch::Tensor x);
+std::vector<torch::Tensor> conv3x3_fwd(torch::Tensor a, torch::Tensor w9,
+                                       int64_t imgH, int64_t imgW,
+                                       c10::optional<torch::Tensor> bias,
+                                       c10::optional<torch::Tensor> scale,
+                                       c10::optional<torch::Tensor> shift,
+                                       c10::optional<torch::Tensor> residual,
+                                       bool relu, bool want_stats);
 // reorder.hip
 torch::Tensor transpose2d(torch::Tensor src);
 torch::Tensor stride2_gather(torch::Tensor x);
@@ -119,6 +126,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("shift") = py::none(), py::arg("residual") = py::none(),
         py::arg("relu") = false, py::arg("want_stats") = false);
   m.def("conv1x1_wgrad", &conv1x1_wgrad);
+  m.def("conv3x3_fwd", &conv3x3_fwd, py::arg("a"), py::arg("w9"),
+        py::arg("imgH"), py::arg("imgW"), py::arg("bias") = py::none(),
+        py::arg("scale") = py::none(), py::arg("shift") = py::none(),
+        py::arg("residual") = py::none(), py::arg("relu") = false,
+        py::arg("want_stats") = false);
   m.def("transpose2d", &transpose2d);
   m.def("stride2_gather", &stride2_gather);
   m.def("stride2_scatter", &stride2_scatter);

@@ -46,7 +46,13 @@ __device__ __forceinline__ void glds16(const bf16* gsrc, bf16* lds_dst) {
 // conv1x1 forward / dgrad kernel: C[M,N] = A[M,K] @ B[N,K]^T (bf16 in/out,
 // fp32 accumulate). K % 64 == 0, N % BN == 0.
 // ---------------------------------------------------------------------------
-template <int BN, bool STATS, int BK = 64>
+// TAPS=1: plain 1x1 GEMM. TAPS=9: 3x3 stride-1 pad-1 implicit GEMM —
+// K = 9*Cin tap-major, each BK k-step lies inside ONE (ky,kx) tap plane
+// (Cin % 64 == 0), the A row for output position m is the tap-shifted
+// input row with zero predication at the image boundary; everything else
+// (B staging from the wrapper-pre-permuted [Co][9Ci] weight, MFMA loop,
+// stats/bias/scale/relu/residual epilogue, store transpose) is shared.
+template <int BN, bool STATS, int BK = 64, int TAPS = 1>
 __global__ __launch_bounds__((BN == 256) ? 512 : 256) void conv1x1_nt_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
     bf16* __restrict__ C, const bf16* __restrict__ residual,  // [M,N] | null
@@ -54,7 +60,7 @@ __global__ __launch_bounds__((BN == 256) ? 512 : 256) void conv1x1_nt_kernel(
     const float* __restrict__ scale,                          // [N] | null
     const float* __restrict__ shift,                          // [N] | null
     float* __restrict__ sums,  // [gridDim.x, 2N] per-block partials | null
-    int64_t M, int K, int N, bool relu) {
+    int64_t M, int K, int N, bool relu, int imgH = 0, int imgW = 0) {
   constexpr int BM = 128;
   constexpr int NWAVES = (BN == 256) ? 8 : 4;
   constexpr int WAVES_M = (BN == 64) ? 4 : 2;
@@ -97,18 +103,50 @@ __global__ __launch_bounds__((BN == 256) ? 512 : 256) void conv1x1_nt_kernel(
   auto swz = [&](int row) {
     return BK == 64 ? (row & 7) << 4 : ((row ^ (row >> 2)) & 3) << 4;
   };
+  const int CinK = K / TAPS;  // physical A row length (elements)
   auto stage = [&](int buf, int kt) {
-    const int k0b = kt * BK * 2;  // byte offset into a row of A/B
+    const int k0b_all = kt * BK * 2;
+    // tap plane + byte offset within the tap's Cin slice
+    const int tap = TAPS == 1 ? 0 : (kt * BK) / CinK;
+    const int k0b = TAPS == 1 ? k0b_all : ((kt * BK) % CinK) * 2;
+    const int ky = tap / 3 - 1, kx = tap % 3 - 1;  // -1..1 (TAPS==9)
     if (!direct_a) {
 #pragma unroll
       for (int pp = 0; pp < (BM / RPP) / NWAVES; ++pp) {
         const int p = wave + pp * NWAVES;
         const int row = p * RPP + lane / LPR;
         const int b = (lane % LPR) * 16;
-        const int64_t rg = m0 + row < M ? m0 + row : M - 1;
-        const char* src = (const char*)A + rg * (int64_t)K * 2 + k0b +
-                          (b ^ swz(row));
-        glds16((const bf16*)src, a_lds(buf) + p * 512);
+        if (TAPS == 1) {
+          const int64_t rg = m0 + row < M ? m0 + row : M - 1;
+          const char* src = (const char*)A + rg * (int64_t)CinK * 2 + k0b +
+                            (b ^ swz(row));
+          glds16((const bf16*)src, a_lds(buf) + p * 512);
+        } else {
+          // per-row tap shift + boundary predication: a 128-row strip
+          // nearly always crosses a W boundary for kx != 0, so this path
+          // stages by predicated 16B loads + LDS vector stores
+          const int64_t m = m0 + row < M ? m0 + row : M - 1;
+          const int64_t img = m / ((int64_t)imgH * imgW);
+          const int r = (int)(m - img * imgH * imgW);
+          const int oh = r / imgW, ow = r % imgW;
+          const int ih = oh + ky, iw = ow + kx;
+          const bool valid = (unsigned)ih < (unsigned)imgH &&
+                             (unsigned)iw < (unsigned)imgW &&
+                             m0 + row < M;
+          Vec<bf16, 8> v;
+          if (valid) {
+            const int64_t in_row = (img * imgH + ih) * imgW + iw;
+            v = *(const Vec<bf16, 8>*)((const char*)A +
+                                       in_row * (int64_t)CinK * 2 + k0b +
+                                       (b ^ swz(row)));
+          } else {
+#pragma unroll
+            for (int j = 0; j < 8; ++j) v.v[j] = from_f32<bf16>(0.f);
+          }
+          vstore<bf16, 8>(a_lds(buf) + p * 512 + (lane % LPR) * 8 +
+                              (lane / LPR) * (BK),
+                          v);
+        }
       }
     }
 #pragma unroll
@@ -116,8 +154,8 @@ __global__ __launch_bounds__((BN == 256) ? 512 : 256) void conv1x1_nt_kernel(
       const int p = wave + pp * NWAVES;
       const int row = p * RPP + lane / LPR;
       const int b = (lane % LPR) * 16;
-      const char* src = (const char*)B + (int64_t)(n0 + row) * K * 2 + k0b +
-                        (b ^ swz(row));
+      const char* src = (const char*)B + (int64_t)(n0 + row) * K * 2 +
+                        k0b_all + (b ^ swz(row));
       glds16((const bf16*)src, b_lds(buf) + p * 512);
     }
   };
@@ -483,6 +521,64 @@ std::vector<torch::Tensor> conv1x1_fwd(torch::Tensor a, torch::Tensor b,
   // 1 block/CU and BN=128 (2 blocks/CU) measures faster.
   if (N % 256 == 0 && nk == 1) pick(std::integral_constant<int, 256>{});
   else if (N % 128 == 0) pick(std::integral_constant<int, 128>{});
+  else pick(std::integral_constant<int, 64>{});
+  HIP_CHECK_ERR();
+  return {C, sums};
+}
+
+// 3x3 stride-1 pad-1 conv as tap-major implicit GEMM (TAPS=9 variant of
+// the kernel above). a: x flattened NHWC [M=B*H*W, Cin]; w9: weight
+// pre-permuted to [Co, 9*Cin] (tap-major: permute(0,2,3,1) of the torch
+// [Co,Ci,3,3] layout). Returns {y [M,Co], stats partials}.
+std::vector<torch::Tensor> conv3x3_fwd(torch::Tensor a, torch::Tensor w9,
+                                       int64_t imgH, int64_t imgW,
+                                       c10::optional<torch::Tensor> bias,
+                                       c10::optional<torch::Tensor> scale,
+                                       c10::optional<torch::Tensor> shift,
+                                       c10::optional<torch::Tensor> residual,
+                                       bool relu, bool want_stats) {
+  DLA_CHECK_INPUT(a);
+  DLA_CHECK_INPUT(w9);
+  TORCH_CHECK(a.scalar_type() == torch::kBFloat16 &&
+                  w9.scalar_type() == torch::kBFloat16,
+              "conv3x3_fwd: bf16 only");
+  const int64_t M = a.size(0);
+  const int Cin = (int)a.size(1), N = (int)w9.size(0);
+  const int K = 9 * Cin;
+  TORCH_CHECK((int)w9.size(1) == K, "conv3x3_fwd: weight must be [Co, 9*Ci]");
+  TORCH_CHECK(Cin % 64 == 0 && N % 64 == 0, "conv3x3_fwd: C must be %64");
+  TORCH_CHECK(M % (imgH * imgW) == 0, "conv3x3_fwd: M != B*H*W");
+  auto C = torch::empty({M, N}, a.options());
+  auto opts_f = a.options().dtype(torch::kFloat);
+  const int gx = (int)((M + 127) / 128);
+  auto sums = want_stats ? torch::empty({gx, 2 * N}, opts_f)
+                         : torch::empty({0}, opts_f);
+  const dla::bf16* res_p =
+      residual.has_value() ? (const dla::bf16*)residual->data_ptr() : nullptr;
+  const float* bias_p = bias.has_value() ? bias->data_ptr<float>() : nullptr;
+  const float* scale_p = scale.has_value() ? scale->data_ptr<float>() : nullptr;
+  const float* shift_p = shift.has_value() ? shift->data_ptr<float>() : nullptr;
+  float* sums_p = want_stats ? sums.data_ptr<float>() : nullptr;
+  auto launch = [&](auto bntag, auto stag) {
+    constexpr int BN = decltype(bntag)::value;
+    constexpr bool ST = decltype(stag)::value;
+    const int lds_stage = (128 * 64 + BN * 64) * 2 * 2;
+    const int lds_ep = 4 * ((BN == 64) ? 32 : 64) / ((BN == 64) ? 1 : 2) *
+                       (64 + 8) * 2;
+    const int lds_st = ST ? ((BN == 64) ? 4 : 2) * 2 * BN * 4 : 0;
+    const int lds = std::max(std::max(lds_stage, lds_ep), lds_st);
+    hipLaunchKernelGGL((dla::conv1x1_nt_kernel<BN, ST, 64, 9>),
+                       dim3(N / BN, gx), dim3(256), lds, dla::stream(),
+                       (const dla::bf16*)a.data_ptr(),
+                       (const dla::bf16*)w9.data_ptr(),
+                       (dla::bf16*)C.data_ptr(), res_p, bias_p, scale_p,
+                       shift_p, sums_p, M, K, N, relu, (int)imgH, (int)imgW);
+  };
+  auto pick = [&](auto bntag) {
+    if (want_stats) launch(bntag, std::true_type{});
+    else launch(bntag, std::false_type{});
+  };
+  if (N % 128 == 0) pick(std::integral_constant<int, 128>{});
   else pick(std::integral_constant<int, 64>{});
   HIP_CHECK_ERR();
   return {C, sums};

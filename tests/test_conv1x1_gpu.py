@@ -323,3 +323,45 @@ def test_gemm_fwd_shape_fuzz():
         ref = a.float() @ b.float().t()
         err = (y.float() - ref).abs().max() / ref.abs().max().clamp(min=1)
         assert err < 3e-2, (m, k, n, float(err))
+
+
+@pytest.mark.parametrize("B,H,W,cin,cout", [
+    (2, 5, 5, 64, 64),       # tiny: every row touches a boundary
+    (2, 56, 56, 64, 64),     # ResNet basic-block class
+    (1, 28, 28, 128, 128),
+    (1, 14, 14, 256, 512),
+])
+def test_conv3x3_fwd_parity(B, H, W, cin, cout):
+    """3x3 s1 p1 implicit-GEMM (TAPS=9) vs F.conv2d fp32, including the
+    zero-predicated boundary taps."""
+    import torch.nn.functional as F
+
+    torch.manual_seed(B + H + cin)
+    x = torch.randn(B, H, W, cin, device="cuda").to(torch.bfloat16)
+    w = torch.randn(cout, cin, 3, 3, device="cuda").to(torch.bfloat16)
+    a = x.reshape(B * H * W, cin)
+    w9 = w.permute(0, 2, 3, 1).reshape(cout, 9 * cin).contiguous()
+    y, _ = ext().conv3x3_fwd(a, w9, H, W, None, None, None, None, False,
+                             False)
+    ref = F.conv2d(x.permute(0, 3, 1, 2).float(), w.float(), padding=1)
+    ref = ref.permute(0, 2, 3, 1).reshape(B * H * W, cout)
+    err = (y.float() - ref).abs().max() / ref.abs().max().clamp(min=1)
+    assert err < 3e-2, float(err)
+
+
+def test_conv3x3_stats_epilogue():
+    import torch.nn.functional as F
+
+    torch.manual_seed(0)
+    B, H, W, cin, cout = 2, 17, 13, 64, 128   # odd spatial, M edge
+    x = torch.randn(B, H, W, cin, device="cuda").to(torch.bfloat16)
+    w = torch.randn(cout, cin, 3, 3, device="cuda").to(torch.bfloat16)
+    a = x.reshape(-1, cin)
+    w9 = w.permute(0, 2, 3, 1).reshape(cout, 9 * cin).contiguous()
+    y, partials = ext().conv3x3_fwd(a, w9, H, W, None, None, None, None,
+                                    False, True)
+    sums = partials.sum(0)
+    ref = F.conv2d(x.permute(0, 3, 1, 2).float(), w.float(), padding=1)
+    ref = ref.permute(0, 2, 3, 1).reshape(-1, cout)
+    assert torch.allclose(sums[:cout], ref.sum(0), rtol=3e-2,
+                          atol=3e-2 * (B * H * W) ** 0.5)
