@@ -369,18 +369,24 @@ def test_conv3x3_stats_epilogue():
 
 def test_conv3x3_bn_train_parity():
     """Fused 3x3 conv+BN(+ReLU) train path (our fwd + fused stats, library
-    backward) vs fp32 eager."""
+    backward) vs fp32 eager. The route is opt-in (DLA_CONV3X3) so force it
+    here; can_fuse_conv3x3 must actually take it."""
     from deeplearning_amd.ops.batchnorm import BatchNorm2d
-    from deeplearning_amd.ops.conv1x1 import conv_bn
+    from deeplearning_amd.ops.conv1x1 import can_fuse_conv3x3, conv_bn
 
+    os.environ["DLA_CONV3X3"] = "1"
     torch.manual_seed(2)
     B, C, H, W, N = 4, 64, 14, 14, 128
     conv = torch.nn.Conv2d(C, N, 3, padding=1, bias=False).cuda()
     bn = BatchNorm2d(N, relu=True).cuda()
     x = torch.randn(B, C, H, W, device="cuda").to(torch.bfloat16) \
         .contiguous(memory_format=torch.channels_last).requires_grad_(True)
-    y = conv_bn(x, conv, bn)
-    y.float().square().mean().backward()
+    try:
+        assert can_fuse_conv3x3(x, conv)
+        y = conv_bn(x, conv, bn)
+        y.float().square().mean().backward()
+    finally:
+        os.environ["DLA_CONV3X3"] = "0"
 
     xr = x.detach().float().requires_grad_(True)
     convr = torch.nn.Conv2d(C, N, 3, padding=1, bias=False).cuda()
