@@ -109,3 +109,17 @@ def test_train_cli_smoke(proj, extra, tmp_path):
     r = subprocess.run(args, capture_output=True, text=True, timeout=900,
                        cwd=REPO)
     assert r.returncode == 0, (r.stdout[-800:], r.stderr[-1500:])
+
+
+@pytest.mark.slow
+@pytest.mark.parametrize("script,out", [
+    ("projects/detection/yolov5/export.py", "y5"),
+    ("projects/detection/YOLOX/export_onnx.py", "yx"),
+])
+def test_detection_export_scripts(script, out, tmp_path):
+    """Per-project export surface (ref yolov5/export.py, YOLOX
+    tools/export_onnx.py); TorchScript fallback when onnx is absent."""
+    dest = tmp_path / f"{out}.onnx"
+    r = _run(script, "--img-size", "320", "--out", str(dest))
+    assert r.returncode == 0, r.stderr[-1500:]
+    assert dest.exists() or (tmp_path / f"{out}.torchscript.pt").exists()
