@@ -46,6 +46,8 @@ std::vector<torch::Tensor> conv3x3_fwd(torch::Tensor a, torch::Tensor w9,
                                        c10::optional<torch::Tensor> shift,
                                        c10::optional<torch::Tensor> residual,
                                        bool relu, bool want_stats);
+torch::Tensor conv3x3_wgrad(torch::Tensor dy, torch::Tensor x, int64_t imgH,
+                            int64_t imgW);
 // reorder.hip
 torch::Tensor transpose2d(torch::Tensor src);
 torch::Tensor stride2_gather(torch::Tensor x);
@@ -131,6 +133,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("scale") = py::none(), py::arg("shift") = py::none(),
         py::arg("residual") = py::none(), py::arg("relu") = false,
         py::arg("want_stats") = false);
+  m.def("conv3x3_wgrad", &conv3x3_wgrad);
   m.def("transpose2d", &transpose2d);
   m.def("stride2_gather", &stride2_gather);
   m.def("stride2_scatter", &stride2_scatter);

@@ -322,12 +322,15 @@ __global__ __launch_bounds__((BN == 256) ? 512 : 256) void conv1x1_nt_kernel(
 // with an m-XOR so the channel-group writes spread over banks), fragments
 // then read m-contiguous b128. 64x64 output tile, 4 waves (32x32 each).
 // ---------------------------------------------------------------------------
-template <int BCO, int BCI>
+// TAPS=9: dW[co][tap*Ci+ci] = sum_m dy[m][co] * x[shift_tap(m)][ci]
+// (3x3 s1 p1 wgrad) — the x tile is staged from tap-shifted rows with
+// zero predication at the image boundary, everything else unchanged.
+template <int BCO, int BCI, int TAPS = 1>
 __global__ __launch_bounds__((BCO * BCI >= 8192) ? 512 : 256) void conv1x1_wgrad_kernel(
     const bf16* __restrict__ dy,  // [M,N]
-    const bf16* __restrict__ x,   // [M,K]
+    const bf16* __restrict__ x,   // [M,Cin] (K = TAPS*Cin)
     float* __restrict__ dW,       // [N,K] pre-zeroed
-    int64_t M, int K, int N, int64_t chunk) {
+    int64_t M, int K, int N, int64_t chunk, int imgH = 0, int imgW = 0) {
   constexpr int KM = 64;   // m per step
   constexpr int TR = 72;   // LDS row length (elements) for [ch][KM] image
   constexpr int NW = (BCO * BCI >= 8192) ? 8 : 4;
@@ -349,9 +352,10 @@ __global__ __launch_bounds__((BCO * BCI >= 8192) ? 512 : 256) void conv1x1_wgrad
   auto xt = [&](int buf) { return lds16 + 2 * BCO * TR + buf * (BCI * TR); };
 
   // transposed scatter-stage of a [KM m][BCH ch] global chunk into
-  // lds[ch][m ^ mswz(ch)] (mswz spreads the 8 channel-groups over banks)
+  // lds[ch][m ^ mswz(ch)] (mswz spreads the 8 channel-groups over banks).
+  // tap >= 0 (TAPS==9): rows are tap-shifted with boundary zeros.
   auto stage_t = [&](bf16* lds, const bf16* g, int stride_elems, int ch0g,
-                     int64_t mt, auto bch_tag) {
+                     int64_t mt, int tap, auto bch_tag) {
     constexpr int BCH = decltype(bch_tag)::value;
 #pragma unroll
     for (int it = 0; it < (KM * BCH / 8) / (NW * 64); ++it) {
@@ -359,8 +363,20 @@ __global__ __launch_bounds__((BCO * BCI >= 8192) ? 512 : 256) void conv1x1_wgrad
       const int m = idx / (BCH / 8);            // 0..63
       const int ch = (idx % (BCH / 8)) * 8;
       Vec<bf16, 8> v;
-      if (mt + m < m_end) {
-        v = vload<bf16, 8>(g + (mt + m) * (int64_t)stride_elems + ch0g + ch);
+      bool valid = mt + m < m_end;
+      int64_t src_row = mt + m;
+      if (TAPS != 1 && tap >= 0 && valid) {
+        const int ky = tap / 3 - 1, kx = tap % 3 - 1;
+        const int64_t img = src_row / ((int64_t)imgH * imgW);
+        const int r = (int)(src_row - img * imgH * imgW);
+        const int oh = r / imgW, ow = r % imgW;
+        const int ih = oh + ky, iw = ow + kx;
+        valid = (unsigned)ih < (unsigned)imgH &&
+                (unsigned)iw < (unsigned)imgW;
+        src_row = (img * imgH + ih) * imgW + iw;
+      }
+      if (valid) {
+        v = vload<bf16, 8>(g + src_row * (int64_t)stride_elems + ch0g + ch);
       } else {
 #pragma unroll
         for (int j = 0; j < 8; ++j) v.v[j] = from_f32<bf16>(0.f);
@@ -382,15 +398,19 @@ __global__ __launch_bounds__((BCO * BCI >= 8192) ? 512 : 256) void conv1x1_wgrad
   const int wco = (wave / WGI) * WCO;
   const int wci = (wave % WGI) * WCI;
 
-  stage_t(dyt(0), dy, N, n0, m_begin, std::integral_constant<int, BCO>{});
-  stage_t(xt(0), x, K, k0, m_begin, std::integral_constant<int, BCI>{});
+  const int CinK = K / TAPS;
+  const int tap = TAPS == 1 ? -1 : k0 / CinK;       // tap of this ci tile
+  const int ci0 = TAPS == 1 ? k0 : k0 % CinK;       // offset within x row
+  stage_t(dyt(0), dy, N, n0, m_begin, -1, std::integral_constant<int, BCO>{});
+  stage_t(xt(0), x, CinK, ci0, m_begin, tap,
+          std::integral_constant<int, BCI>{});
   __syncthreads();
   int cur = 0;
   for (int64_t mt = m_begin; mt < m_end; mt += KM) {
     if (mt + KM < m_end) {
-      stage_t(dyt(cur ^ 1), dy, N, n0, mt + KM,
+      stage_t(dyt(cur ^ 1), dy, N, n0, mt + KM, -1,
               std::integral_constant<int, BCO>{});
-      stage_t(xt(cur ^ 1), x, K, k0, mt + KM,
+      stage_t(xt(cur ^ 1), x, CinK, ci0, mt + KM, tap,
               std::integral_constant<int, BCI>{});
     }
 #pragma unroll
@@ -629,6 +649,62 @@ torch::Tensor conv1x1_wgrad(torch::Tensor dy, torch::Tensor x) {
   if (n128 && k128)
     launch(std::integral_constant<int, 128>{}, std::integral_constant<int, 128>{});
   else if (k128)
+    launch(std::integral_constant<int, 64>{}, std::integral_constant<int, 128>{});
+  else if (n128)
+    launch(std::integral_constant<int, 128>{}, std::integral_constant<int, 64>{});
+  else
+    launch(std::integral_constant<int, 64>{}, std::integral_constant<int, 64>{});
+  HIP_CHECK_ERR();
+  return dW;
+}
+
+// dW9 [Co, 9*Ci] fp32 = 3x3 s1 p1 wgrad (tap-major k, same order the
+// conv3x3_fwd weight uses)
+torch::Tensor conv3x3_wgrad(torch::Tensor dy, torch::Tensor x, int64_t imgH,
+                            int64_t imgW) {
+  DLA_CHECK_INPUT(dy);
+  DLA_CHECK_INPUT(x);
+  TORCH_CHECK(dy.scalar_type() == torch::kBFloat16 &&
+                  x.scalar_type() == torch::kBFloat16,
+              "conv3x3_wgrad: bf16 only");
+  const int64_t M = x.size(0);
+  const int Cin = (int)x.size(1), N = (int)dy.size(1);
+  const int K = 9 * Cin;
+  TORCH_CHECK(dy.size(0) == M, "conv3x3_wgrad: M mismatch");
+  TORCH_CHECK(Cin % 64 == 0 && N % 64 == 0, "conv3x3_wgrad: C must be %64");
+  auto dW = torch::zeros({N, K}, x.options().dtype(torch::kFloat));
+  auto launch = [&](auto cotag, auto citag) {
+    constexpr int BCO = decltype(cotag)::value;
+    constexpr int BCI = decltype(citag)::value;
+    const int tiles = (N / BCO) * (K / BCI);
+    int splits = (int)std::min<int64_t>(
+        std::max<int64_t>(1, 1024 / tiles),
+        (M + 2047) / 2048);
+    const int64_t chunk0 = (M + splits - 1) / splits;
+    const int64_t chunk = ((chunk0 + 63) / 64) * 64;
+    splits = (int)((M + chunk - 1) / chunk);
+    const int lds = 2 * (BCO + BCI) * 72 * 2;
+    if (lds > 65536) {
+      static bool done = false;
+      if (!done) {
+        (void)hipFuncSetAttribute(
+            (const void*)&dla::conv1x1_wgrad_kernel<BCO, BCI, 9>,
+            hipFuncAttributeMaxDynamicSharedMemorySize, 163840);
+        done = true;
+      }
+    }
+    hipLaunchKernelGGL((dla::conv1x1_wgrad_kernel<BCO, BCI, 9>),
+                       dim3(N / BCO, K / BCI, splits),
+                       dim3(BCO * BCI >= 8192 ? 512 : 256), lds,
+                       dla::stream(), (const dla::bf16*)dy.data_ptr(),
+                       (const dla::bf16*)x.data_ptr(), dW.data_ptr<float>(),
+                       M, K, N, chunk, (int)imgH, (int)imgW);
+  };
+  // BCI must divide Cin so a ci tile stays inside one tap plane
+  const bool n128 = N % 128 == 0, ci128 = Cin % 128 == 0;
+  if (n128 && ci128)
+    launch(std::integral_constant<int, 128>{}, std::integral_constant<int, 128>{});
+  else if (ci128)
     launch(std::integral_constant<int, 64>{}, std::integral_constant<int, 128>{});
   else if (n128)
     launch(std::integral_constant<int, 128>{}, std::integral_constant<int, 64>{});

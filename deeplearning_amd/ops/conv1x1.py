@@ -223,8 +223,9 @@ def conv_bn(x: torch.Tensor, conv: nn.Conv2d, bn) -> torch.Tensor:
 class _Conv3x3Fn(torch.autograd.Function):
     """3x3 s1 p1 conv: forward on the TAPS=9 implicit-GEMM MFMA kernel
     (beats MIOpen fwd on 3 of 4 ResNet shapes, up to 2.75x, and can fuse
-    the BN-stats epilogue); backward via aten.convolution_backward (the
-    hand-written 3x3 dgrad/wgrad are the next kernel item — ROADMAP)."""
+    the BN-stats epilogue); dgrad reuses the SAME kernel with the flipped,
+    channel-transposed weight; wgrad is the TAPS=9 variant of the
+    M-contraction wgrad kernel — conv2d 3x3 is hand-written end to end."""
 
     @staticmethod
     def forward(ctx, x4d, w, want_stats):
@@ -247,17 +248,30 @@ class _Conv3x3Fn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dy, *unused):
         xb, wb = ctx.saved_tensors
+        B, C, H, W = xb.shape
         if not dy.is_contiguous(memory_format=torch.channels_last):
             dy = dy.contiguous(memory_format=torch.channels_last)
         if dy.dtype != torch.bfloat16:
             dy = dy.to(torch.bfloat16)
-        dx, dw, _ = torch.ops.aten.convolution_backward(
-            dy, xb, wb, None, [1, 1], [1, 1], [1, 1], False, [0, 0], 1,
-            [ctx.needs_input_grad[0], ctx.needs_input_grad[1], False])
-        if dx is not None and dx.dtype != ctx.x_dtype:
-            dx = dx.to(ctx.x_dtype)
-        if dw is not None and dw.dtype != ctx.w_dtype:
-            dw = dw.to(ctx.w_dtype)
+        dy2d = _flatten_nhwc(dy)
+        dx = dw = None
+        if ctx.needs_input_grad[0]:
+            # dgrad IS the same TAPS=9 kernel: conv3x3 of dy with the
+            # spatially-flipped, channel-transposed weight
+            wd = wb.flip(2, 3).transpose(0, 1)       # [Ci, Co, 3, 3]
+            wd9 = wd.permute(0, 2, 3, 1).reshape(C, -1).contiguous()
+            dx2d, _ = ext().conv3x3_fwd(dy2d, wd9, H, W, None, None, None,
+                                        None, False, False)
+            dx = _unflatten_nhwc(dx2d, B, H, W)
+            if dx.dtype != ctx.x_dtype:
+                dx = dx.to(ctx.x_dtype)
+        if ctx.needs_input_grad[1]:
+            dw9 = ext().conv3x3_wgrad(dy2d, _flatten_nhwc(xb), H, W)
+            dw = dw9.view(wb.shape[0], 3, 3, C).permute(0, 3, 1, 2)
+            if dw.dtype != ctx.w_dtype:
+                dw = dw.to(ctx.w_dtype)
+            else:
+                dw = dw.contiguous()
         return dx, dw, None
 
 

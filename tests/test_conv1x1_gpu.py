@@ -403,3 +403,34 @@ def test_conv3x3_bn_train_parity():
     assert relerr(x.grad, xr.grad) < 8e-2
     assert relerr(conv.weight.grad, convr.weight.grad) < 8e-2
     assert relerr(bn.running_mean, bnr.running_mean) < 5e-2
+
+
+def test_conv3x3_dgrad_wgrad_parity():
+    """Hand-written 3x3 backward: dgrad (flipped-weight TAPS=9 fwd) and
+    wgrad (tap-shifted M-contraction) vs autograd on F.conv2d."""
+    import torch.nn.functional as F
+
+    from deeplearning_amd.ops.conv1x1 import _Conv3x3Fn
+
+    torch.manual_seed(4)
+    B, C, H, W, N = 3, 64, 15, 11, 128   # odd spatial: boundary taps hit
+    x = torch.randn(B, C, H, W, device="cuda").to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last).requires_grad_(True)
+    w = torch.randn(N, C, 3, 3, device="cuda", dtype=torch.float32) \
+        .mul(0.1).requires_grad_(True)
+    y = _Conv3x3Fn.apply(x, w, False)
+    y.float().square().mean().backward()
+
+    xr = x.detach().float().requires_grad_(True)
+    wr = w.detach().clone().requires_grad_(True)
+    yr = F.conv2d(xr, wr, padding=1)
+    yr.square().mean().backward()
+
+    def relerr(a, b):
+        return (a.float() - b.float()).abs().max() / \
+            b.float().abs().max().clamp(min=1e-4)
+
+    assert relerr(y, yr.detach()) < 4e-2
+    assert relerr(x.grad, xr.grad) < 8e-2, float(relerr(x.grad, xr.grad))
+    assert relerr(w.grad, wr.grad) < 8e-2, float(relerr(w.grad, wr.grad))
+    assert w.grad.dtype == torch.float32
