@@ -73,9 +73,10 @@ __device__ void stage_rows(const __hip_bfloat16* src, int64_t row_stride,
 // ---------------------------------------------------------------- forward --
 // COSINE (Swin-v2): S = logit_scale[h] * cos(q, k) instead of scale * q.k —
 // per-row q/k inverse norms are computed in-kernel; `scale` is unused and
-// `lscale` carries the per-head clamped-exp logit scales. Inference-staged
-// (backward through the normalize is not implemented; the autograd wrapper
-// only routes no-grad calls here).
+// `lscale` carries the per-head clamped-exp logit scales. Training uses
+// this forward with SAVE_P; the cosine chain rule (incl. logit_scale and
+// CPB-bias grads) runs from the saved P in the Python wrapper
+// (ops/attention._AttnCosineFn).
 template <int D, bool HAS_BIAS, bool HAS_MASK, bool SAVE_P, bool COSINE = false>
 __global__ __launch_bounds__(512)
 void attn_fwd_kernel(const __hip_bfloat16* __restrict__ qkv,  // [B,N,3,H,D]

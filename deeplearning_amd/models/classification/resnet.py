@@ -3,9 +3,11 @@ BatchNorm2d(relu=True) (HIP fused BN+ReLU) and add_relu (fused residual join).
 
 Reference parity: classification/resnet/models/networks.py (BasicBlock:38,
 Bottleneck:78, ResNet:127, factories :235-341) — re-designed, not translated:
-the conv->bn->relu chain runs conv (MIOpen implicit GEMM) + one fused HIP
-BN+ReLU kernel instead of three eager ops, and each residual join is one
-fused add+relu kernel.
+every 1x1 conv (bottleneck conv1/conv3, downsample) runs the hand-written
+implicit-GEMM MFMA kernel with the BatchNorm stats pass fused into the conv
+epilogue (csrc/conv1x1.hip); 3x3 convs run MIOpen by default with a fully
+hand-written TAPS=9 route available (DLA_CONV3X3=1, see ROADMAP); each
+residual join is one fused add+relu kernel.
 """
 from __future__ import annotations
 
@@ -98,7 +100,7 @@ class Bottleneck(nn.Module):
         identity = x
         # conv1/conv3 (1x1) ride the hand-written MFMA implicit-GEMM kernel
         # with the BN stats pass fused into the conv epilogue (conv1x1.hip);
-        # conv2 (3x3) stays on the library conv.
+        # conv2 (3x3) uses MIOpen by default (the TAPS=9 route is opt-in).
         out = conv_bn(x, self.conv1, self.bn1)
         out = conv_bn(out, self.conv2, self.bn2)  # 3x3 s1 -> TAPS=9 kernel
         out = conv_bn(out, self.conv3, self.bn3)

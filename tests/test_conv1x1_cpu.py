@@ -47,3 +47,15 @@ def test_resnet50_cpu_forward_unchanged():
         y = model(x)
     assert y.shape == (1, 10)
     assert torch.isfinite(y).all()
+
+
+def test_conv3x3_routing_default_off(monkeypatch):
+    """3x3 routing is opt-in (measured: integrated step regresses until the
+    backward kernels are tuned) — the gate must default off and respond to
+    DLA_CONV3X3."""
+    from deeplearning_amd.ops import conv1x1 as m
+
+    monkeypatch.delenv("DLA_CONV3X3", raising=False)
+    assert not m._conv3x3_enabled()
+    monkeypatch.setenv("DLA_CONV3X3", "1")
+    assert m._conv3x3_enabled()
