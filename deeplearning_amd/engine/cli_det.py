@@ -254,6 +254,7 @@ def detect_main(default_model: str, num_classes: int = 21):
     boxes = det["boxes"][keep].cpu()
     scores = det["scores"][keep].cpu()
     labels = det["labels"][keep].cpu()
+    print(f"{len(boxes)} detection(s) above score {args.score_thresh}")
     for b, s, l in zip(boxes.tolist(), scores.tolist(), labels.tolist()):
         print(f"class {l}  score {s:.3f}  box "
               f"[{b[0]:.1f}, {b[1]:.1f}, {b[2]:.1f}, {b[3]:.1f}]")
