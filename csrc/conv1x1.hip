@@ -389,67 +389,90 @@ __global__ __launch_bounds__((BCO * BCI >= 8192) ? 512 : 256) void conv1x1_wgrad
     }
   };
 
-  f32x4 acc[FCO][FCI];
+  // TAPS==9: one block owns a ky ROW of taps (3 of them) for its
+  // (co, ci) tile, so dy is staged ONCE per m-step instead of once per
+  // tap (the all-tap-tiles grid re-read dy ~9x: ~1.8 GB/call at the
+  // 56x56x64 shape). The kx taps loop inside over alternating x buffers.
+  constexpr int KXN = (TAPS == 1) ? 1 : 3;
+  f32x4 acc[KXN][FCO][FCI];
 #pragma unroll
-  for (int mi = 0; mi < FCO; ++mi)
+  for (int kx = 0; kx < KXN; ++kx)
 #pragma unroll
-    for (int ni = 0; ni < FCI; ++ni) acc[mi][ni] = f32x4{0.f, 0.f, 0.f, 0.f};
+    for (int mi = 0; mi < FCO; ++mi)
+#pragma unroll
+      for (int ni = 0; ni < FCI; ++ni)
+        acc[kx][mi][ni] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const int wco = (wave / WGI) * WCO;
   const int wci = (wave % WGI) * WCI;
 
   const int CinK = K / TAPS;
-  const int tap = TAPS == 1 ? -1 : k0 / CinK;       // tap of this ci tile
+  const int ky_idx = TAPS == 1 ? 0 : k0 / CinK;     // ky row of this block
   const int ci0 = TAPS == 1 ? k0 : k0 % CinK;       // offset within x row
+  auto tap_of = [&](int kx) { return TAPS == 1 ? -1 : ky_idx * 3 + kx; };
+
   stage_t(dyt(0), dy, N, n0, m_begin, -1, std::integral_constant<int, BCO>{});
-  stage_t(xt(0), x, CinK, ci0, m_begin, tap,
+  stage_t(xt(0), x, CinK, ci0, m_begin, tap_of(0),
           std::integral_constant<int, BCI>{});
   __syncthreads();
-  int cur = 0;
+  int xbuf = 0, dybuf = 0;
   for (int64_t mt = m_begin; mt < m_end; mt += KM) {
-    if (mt + KM < m_end) {
-      stage_t(dyt(cur ^ 1), dy, N, n0, mt + KM, -1,
-              std::integral_constant<int, BCO>{});
-      stage_t(xt(cur ^ 1), x, CinK, ci0, mt + KM, tap,
-              std::integral_constant<int, BCI>{});
-    }
 #pragma unroll
-    for (int ks = 0; ks < 2; ++ks) {
-      const int mm = ks * 32 + (lane >> 4) * 8;
-      bf16x8 af[FCO], bfr[FCI];
-#pragma unroll
-      for (int mi = 0; mi < FCO; ++mi) {
-        const int c = wco + mi * 16 + (lane & 15);
-        af[mi] = *(const bf16x8*)(dyt(cur) + c * TR +
-                                  (mm ^ (((c >> 3) & 7) << 3)));
+    for (int kx = 0; kx < KXN; ++kx) {
+      // prefetch the next x tile (next tap, or next m-step's first tap)
+      if (kx + 1 < KXN) {
+        stage_t(xt(xbuf ^ 1), x, CinK, ci0, mt, tap_of(kx + 1),
+                std::integral_constant<int, BCI>{});
+      } else if (mt + KM < m_end) {
+        stage_t(xt(xbuf ^ 1), x, CinK, ci0, mt + KM, tap_of(0),
+                std::integral_constant<int, BCI>{});
+        stage_t(dyt(dybuf ^ 1), dy, N, n0, mt + KM, -1,
+                std::integral_constant<int, BCO>{});
       }
 #pragma unroll
-      for (int ni = 0; ni < FCI; ++ni) {
-        const int c = wci + ni * 16 + (lane & 15);
-        bfr[ni] = *(const bf16x8*)(xt(cur) + c * TR +
-                                   (mm ^ (((c >> 3) & 7) << 3)));
+      for (int ks = 0; ks < 2; ++ks) {
+        const int mm = ks * 32 + (lane >> 4) * 8;
+        bf16x8 af[FCO], bfr[FCI];
+#pragma unroll
+        for (int mi = 0; mi < FCO; ++mi) {
+          const int c = wco + mi * 16 + (lane & 15);
+          af[mi] = *(const bf16x8*)(dyt(dybuf) + c * TR +
+                                    (mm ^ (((c >> 3) & 7) << 3)));
+        }
+#pragma unroll
+        for (int ni = 0; ni < FCI; ++ni) {
+          const int c = wci + ni * 16 + (lane & 15);
+          bfr[ni] = *(const bf16x8*)(xt(xbuf) + c * TR +
+                                     (mm ^ (((c >> 3) & 7) << 3)));
+        }
+#pragma unroll
+        for (int mi = 0; mi < FCO; ++mi)
+#pragma unroll
+          for (int ni = 0; ni < FCI; ++ni)
+            acc[kx][mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af[mi], bfr[ni], acc[kx][mi][ni], 0, 0, 0);
       }
-#pragma unroll
-      for (int mi = 0; mi < FCO; ++mi)
-#pragma unroll
-        for (int ni = 0; ni < FCI; ++ni)
-          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
+      __syncthreads();
+      xbuf ^= 1;
     }
-    __syncthreads();
-    cur ^= 1;
+    dybuf ^= 1;
   }
 
 #pragma unroll
-  for (int mi = 0; mi < FCO; ++mi)
+  for (int kx = 0; kx < KXN; ++kx)
 #pragma unroll
-    for (int ni = 0; ni < FCI; ++ni)
+    for (int mi = 0; mi < FCO; ++mi)
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int co = n0 + wco + mi * 16 + (lane >> 4) * 4 + r;
-        const int ci = k0 + wci + ni * 16 + (lane & 15);
-        atomicAdd(&dW[(int64_t)co * K + ci], acc[mi][ni][r]);
-      }
+      for (int ni = 0; ni < FCI; ++ni)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int co = n0 + wco + mi * 16 + (lane >> 4) * 4 + r;
+          const int ci_local = wci + ni * 16 + (lane & 15);
+          const int kg = TAPS == 1
+              ? k0 + ci_local
+              : (ky_idx * 3 + kx) * CinK + ci0 + ci_local;
+          atomicAdd(&dW[(int64_t)co * K + kg], acc[kx][mi][ni][r]);
+        }
 }
 
 }  // namespace dla
@@ -676,7 +699,7 @@ torch::Tensor conv3x3_wgrad(torch::Tensor dy, torch::Tensor x, int64_t imgH,
   auto launch = [&](auto cotag, auto citag) {
     constexpr int BCO = decltype(cotag)::value;
     constexpr int BCI = decltype(citag)::value;
-    const int tiles = (N / BCO) * (K / BCI);
+    const int tiles = (N / BCO) * (3 * Cin / BCI);  // ky-row-merged grid
     int splits = (int)std::min<int64_t>(
         std::max<int64_t>(1, 1024 / tiles),
         (M + 2047) / 2048);
@@ -694,7 +717,7 @@ torch::Tensor conv3x3_wgrad(torch::Tensor dy, torch::Tensor x, int64_t imgH,
       }
     }
     hipLaunchKernelGGL((dla::conv1x1_wgrad_kernel<BCO, BCI, 9>),
-                       dim3(N / BCO, K / BCI, splits),
+                       dim3(N / BCO, 3 * Cin / BCI, splits),
                        dim3(BCO * BCI >= 8192 ? 512 : 256), lds,
                        dla::stream(), (const dla::bf16*)dy.data_ptr(),
                        (const dla::bf16*)x.data_ptr(), dW.data_ptr<float>(),
