@@ -34,7 +34,7 @@ def main():
 
         ours = t_ms(lambda: E.conv3x3_wgrad(dyf, xf, H, H))
         y = torch.nn.functional.conv2d(x, w, padding=1)
-        g = torch.autograd.grad(y, w, dy, retain_graph=True)
+        torch.autograd.grad(y, w, dy, retain_graph=True)  # MIOpen find warmup
         miopen = t_ms(lambda: torch.autograd.grad(y, w, dy,
                                                   retain_graph=True))
         # parity vs fp32 eager
