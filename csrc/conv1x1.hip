@@ -52,7 +52,7 @@ __device__ __forceinline__ void glds16(const bf16* gsrc, bf16* lds_dst) {
 // input row with zero predication at the image boundary; everything else
 // (B staging from the wrapper-pre-permuted [Co][9Ci] weight, MFMA loop,
 // stats/bias/scale/relu/residual epilogue, store transpose) is shared.
-template <int BN, bool STATS, int BK = 64, int TAPS = 1>
+template <int BN, bool STATS, int BK = 64, int TAPS = 1, int RING = 0>
 __global__ __launch_bounds__((BN == 256) ? 512 : 256) void conv1x1_nt_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
     bf16* __restrict__ C, const bf16* __restrict__ residual,  // [M,N] | null
@@ -84,7 +84,9 @@ __global__ __launch_bounds__((BN == 256) ? 512 : 256) void conv1x1_nt_kernel(
   // direct-global A fragments measured SLOWER than LDS staging at nk==1
   // (latency-exposed loads, no glds prefetch): keep the staged path.
   const bool direct_a = false;
-  const int nbuf = nk > 1 ? 2 : 1;  // single staging buffer when one K-step
+  // RING>0: R-slot glds ring, stage 2 ahead (guide "pipelining across
+  // barriers": raw s_barrier + counted vmcnt leaves 2 stages in flight)
+  const int nbuf = RING > 0 ? RING : (nk > 1 ? 2 : 1);
   auto a_lds = [&](int buf) { return lds16 + buf * (BM * BK); };
   auto b_lds = [&](int buf) {
     return (direct_a ? lds16 : lds16 + nbuf * BM * BK) + buf * (BN * BK);
@@ -169,13 +171,7 @@ __global__ __launch_bounds__((BN == 256) ? 512 : 256) void conv1x1_nt_kernel(
 #pragma unroll
     for (int ni = 0; ni < NFR; ++ni) acc[mi][ni] = f32x4{0.f, 0.f, 0.f, 0.f};
 
-  stage(0, 0);
-  __syncthreads();  // drains the glds (vmcnt0 inside the barrier)
-  int cur = 0;
-  for (int kt = 0; kt < nk; ++kt) {
-    if (kt + 1 < nk) stage(cur ^ 1, kt + 1);
-    const char* ab = (const char*)a_lds(cur);
-    const char* bb = (const char*)b_lds(cur);
+  auto compute_step = [&](const char* ab, const char* bb) {
 #pragma unroll
     for (int ks = 0; ks < BK / 32; ++ks) {
       const int kbyte = (ks * 32 + (lane >> 4) * 8) * 2;
@@ -204,8 +200,43 @@ __global__ __launch_bounds__((BN == 256) ? 512 : 256) void conv1x1_nt_kernel(
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
     }
-    __syncthreads();
-    cur ^= 1;
+  };
+
+  if constexpr (RING == 0) {
+    stage(0, 0);
+    __syncthreads();  // drains the glds (vmcnt0 inside the barrier)
+    int cur = 0;
+    for (int kt = 0; kt < nk; ++kt) {
+      if (kt + 1 < nk) stage(cur ^ 1, kt + 1);
+      compute_step((const char*)a_lds(cur), (const char*)b_lds(cur));
+      __syncthreads();
+      cur ^= 1;
+    }
+  } else {
+    // 4-slot ring, stage 2 ahead. Slot s is overwritten by stage(s+RING)
+    // issued 2 iterations after its readers' barrier (R >= D+2 bounds the
+    // wave skew: issue at iteration j implies all computes <= j-2 done).
+    // Counted vmcnt before a RAW barrier per the guide: __syncthreads()
+    // would drain every in-flight glds to vmcnt(0).
+    static_assert(RING == 4, "ring depth fixed at 4 (stage 2 ahead)");
+    constexpr int G = ((BM / RPP) / NWAVES) + ((BN / RPP) / NWAVES);
+    stage(0, 0);
+    if (nk > 1) stage(1, 1);
+    for (int kt = 0; kt < nk; ++kt) {
+      if (kt + 2 < nk) {
+        stage((kt + 2) % RING, kt + 2);
+        asm volatile("s_waitcnt vmcnt(%0)" ::"n"(2 * G) : "memory");
+      } else if (kt + 1 < nk) {
+        asm volatile("s_waitcnt vmcnt(%0)" ::"n"(G) : "memory");
+      } else {
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      compute_step((const char*)a_lds(kt % RING),
+                   (const char*)b_lds(kt % RING));
+    }
+    __syncthreads();  // all glds drained: bare barrier before LDS reuse
   }
 
   // ---- stats from registers (fp32-exact, before the bf16 rounding) -------
@@ -546,9 +577,30 @@ std::vector<torch::Tensor> conv1x1_fwd(torch::Tensor a, torch::Tensor b,
     const char* e = std::getenv("DLA_C1X1_BK32");
     return e != nullptr && e[0] == '1';
   }();
+  // 4-slot glds ring (BK=32, 64KB LDS, 2 blocks/CU) — opt-in pending A/B
+  // on the wait-bound fat-N shapes (ROADMAP item 2; guide "3-buf span +83%")
+  static const bool ring_on = []{
+    const char* e = std::getenv("DLA_C1X1_RING");
+    return e != nullptr && e[0] == '1';
+  }();
   const bool bk32 = bk32_force || nk == 2;
+  auto launch_ring = [&](auto stag) {
+    constexpr bool ST = decltype(stag)::value;
+    constexpr int lds = 4 * (128 * 32 + 128 * 32) * 2;  // 64 KB
+    hipLaunchKernelGGL((dla::conv1x1_nt_kernel<128, ST, 32, 1, 4>),
+                       dim3(N / 128, gx), dim3(256), lds, dla::stream(),
+                       (const dla::bf16*)a.data_ptr(),
+                       (const dla::bf16*)b.data_ptr(),
+                       (dla::bf16*)C.data_ptr(), res_p, bias_p, scale_p,
+                       shift_p, sums_p, M, K, N, relu);
+  };
   auto pick = [&](auto bntag) {
     constexpr int BNv = decltype(bntag)::value;
+    if (BNv == 128 && ring_on && K % 32 == 0 && K / 32 >= 4) {
+      if (want_stats) launch_ring(std::true_type{});
+      else launch_ring(std::false_type{});
+      return;
+    }
     if (BNv == 128 && bk32 && nk >= 2) {
       if (want_stats) launch(bntag, std::true_type{},
                              std::integral_constant<int, 32>{});
