@@ -27,7 +27,7 @@ def match_image_native(det_boxes, det_scores, gt_boxes, gt_crowd, iou_thrs,
     from ..ops._ext import ext
     matched, ignored, scores, n_gt = ext().cocoeval_match_image(
         det_boxes.float(), det_scores.float(), gt_boxes.float(),
-        gt_crowd.to(torch.bool), torch.tensor(iou_thrs), max_dets)
+        gt_crowd.to(torch.bool), torch.as_tensor(iou_thrs), max_dets)
     return matched, ignored, scores, int(n_gt)
 
 

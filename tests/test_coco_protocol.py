@@ -177,3 +177,63 @@ def test_detevaluator_matches_pycocotools_protocol(seed):
     assert abs(got["mAP"] - ref_map) < 1e-4, (got["mAP"], ref_map)
     assert abs(got["mAP50"] - ref_50) < 1e-4, (got["mAP50"], ref_50)
     assert abs(got["mAP75"] - ref_75) < 1e-4, (got["mAP75"], ref_75)
+
+
+# ---------------------------------------------------------------------------
+# Randomized three-way matcher equivalence: the torch matcher
+# (engine.det_eval.match_image), the C++ matcher (csrc/cocoeval.cpp) and the
+# straight-line numpy protocol reference above must agree bit-for-bit on
+# adversarial inputs: integer-quantized boxes (exact IoU ties, duplicates,
+# zero-overlap), any crowd pattern, empty det/gt sets, maxDets truncation.
+# ---------------------------------------------------------------------------
+from hypothesis import given, settings
+from hypothesis import strategies as st
+
+from deeplearning_amd.engine.det_eval import match_image, match_image_native
+
+
+def _qbox(draw):
+    x0 = draw(st.integers(0, 12))
+    y0 = draw(st.integers(0, 12))
+    w = draw(st.integers(1, 8))
+    h = draw(st.integers(1, 8))
+    return [float(x0), float(y0), float(x0 + w), float(y0 + h)]
+
+
+@settings(max_examples=120, deadline=None)
+@given(data=st.data())
+def test_matcher_threeway_equivalence_fuzz(data):
+    D = data.draw(st.integers(0, 8), label="D")
+    G = data.draw(st.integers(0, 6), label="G")
+    dt = torch.tensor([_qbox(data.draw) for _ in range(D)],
+                      dtype=torch.float32).reshape(D, 4)
+    gt = torch.tensor([_qbox(data.draw) for _ in range(G)],
+                      dtype=torch.float32).reshape(G, 4)
+    crowd = torch.tensor([data.draw(st.booleans()) for _ in range(G)],
+                         dtype=torch.bool)
+    max_dets = data.draw(st.sampled_from([2, 5, 100]), label="max_dets")
+    # distinct scores: tied scores would expose unstable-sort order, which
+    # the protocol leaves unspecified (pycocotools relies on mergesort)
+    perm = torch.randperm(D, generator=torch.Generator().manual_seed(
+        data.draw(st.integers(0, 1000), label="seed")))
+    scores = (0.1 + 0.8 * perm.float() / max(D, 1)).clamp(max=0.99)
+
+    thrs = torch.tensor(COCO_IOU_THRS)
+    m_py, ig_py, sc_py, ngt_py = match_image(dt, scores, gt, crowd, thrs,
+                                             max_dets)
+    m_cc, ig_cc, sc_cc, ngt_cc = match_image_native(dt, scores, gt, crowd,
+                                                    thrs, max_dets)
+    assert torch.equal(m_py, m_cc)
+    assert torch.equal(ig_py, ig_cc)
+    assert torch.allclose(sc_py, sc_cc)
+    assert ngt_py == ngt_cc
+
+    m_np, ig_np, sc_np, ngt_np = _evaluate_img(
+        dt.numpy(), scores.numpy(), gt.numpy(),
+        crowd.numpy().astype(np.int64), np.array(COCO_IOU_THRS), max_dets)
+    # pycocotools sets dtm for crowd matches too (with dtIg=True); our
+    # matchers report matched=False there. Both yield identical PR curves:
+    # an ignored det is neither TP nor FP, so only (m & ~ig) and ig matter.
+    assert np.array_equal((m_py & ~ig_py).numpy(), m_np & ~ig_np)
+    assert np.array_equal(ig_py.numpy(), ig_np)
+    assert ngt_py == ngt_np
