@@ -1,0 +1,80 @@
+"""Property tests for data/samplers.py invariants (ref fasterRcnn
+utils/group_by_aspect_ratio.py, YOLOX yolox/data/samplers.py, swin
+dataLoader/samplers.py)."""
+import itertools
+
+from hypothesis import given, settings
+from hypothesis import strategies as st
+
+from deeplearning_amd.data.samplers import (GroupedBatchSampler,
+                                            InfiniteSampler,
+                                            SubsetRandomSampler,
+                                            create_aspect_ratio_groups)
+
+
+@settings(max_examples=60, deadline=None)
+@given(
+    n=st.integers(1, 40),
+    batch_size=st.integers(1, 6),
+    n_groups=st.integers(1, 4),
+    seed=st.integers(0, 100),
+)
+def test_grouped_batch_sampler_invariants(n, batch_size, n_groups, seed):
+    import random
+
+    rng = random.Random(seed)
+    group_ids = [rng.randrange(n_groups) for _ in range(n)]
+    order = list(range(n))
+    rng.shuffle(order)
+    s = GroupedBatchSampler(order, group_ids, batch_size)
+    batches = list(s)
+    # 1. declared length matches reality (schedulers size from len(loader))
+    assert len(batches) == len(s)
+    # 2. every batch is full and group-uniform
+    for b in batches:
+        assert len(b) == batch_size
+        assert len({group_ids[i] for i in b}) == 1
+    # 3. every index the sampler produced appears at least once
+    seen = {i for b in batches for i in b}
+    assert seen == set(order)
+
+
+def test_infinite_sampler_rank_sharding_partitions_stream():
+    world = 4
+    size = 13
+    epoch = [list(itertools.islice(iter(InfiniteSampler(
+        size, shuffle=True, seed=7, rank=r, world_size=world)), 0, 12))
+        for r in range(world)]
+    # interleaved shards reassemble the single-rank stream
+    solo = list(itertools.islice(iter(InfiniteSampler(
+        size, shuffle=True, seed=7, rank=0, world_size=1)), 0, 48))
+    merged = [epoch[i % world][i // world] for i in range(48)]
+    assert merged == solo
+    # every full window of `size` interleaved draws is a permutation
+    assert sorted(solo[:size]) == list(range(size))
+
+
+def test_subset_random_sampler_epoch_determinism():
+    idx = [3, 5, 8, 13, 21]
+    s = SubsetRandomSampler(idx)
+    s.set_epoch(1)
+    a = list(s)
+    b = list(s)
+    assert a == b and sorted(a) == sorted(idx)
+    s.set_epoch(2)
+    c = list(s)
+    assert sorted(c) == sorted(idx)
+    assert a != c or len(idx) <= 1
+
+
+@given(st.lists(st.floats(0.2, 5.0), min_size=1, max_size=50),
+       st.integers(0, 4))
+@settings(max_examples=40, deadline=None)
+def test_aspect_ratio_groups_monotone(ratios, k):
+    gids = create_aspect_ratio_groups(ratios, k=k)
+    assert len(gids) == len(ratios)
+    assert all(0 <= g <= 2 * k + 1 for g in gids)
+    # grouping must be monotone in the ratio
+    pairs = sorted(zip(ratios, gids))
+    for (r1, g1), (r2, g2) in zip(pairs, pairs[1:]):
+        assert g1 <= g2
