@@ -6,8 +6,11 @@ usual; per-param post-accumulate hooks mark bucket members ready; a full bucket
 is flattened into a preallocated flat buffer and all-reduced asynchronously
 (RCCL schedules on its own stream -> overlaps the remaining backward). Buckets
 are filled in reverse parameter order (approximate backward order). Bucket size
-defaults to 64 MiB — sized so each ring step saturates a single xGMI link while
-still giving several buckets of overlap (tuned on hardware; see profiles/).
+defaults to 64 MiB — sized by xGMI arithmetic, not yet swept on an 8-GPU node
+(single-GPU rounds): a 64 MiB ring step is far above RCCL's latency floor and
+the whole-model all-reduce (~1-4 ms for resnet50/vit_b16 at ring-8 link speed)
+hides under a 23-46 ms backward with >=2 buckets of overlap. `bench.py
+--bucket-mb` sweeps 16/32/64/128 when a multi-GPU node is available.
 
 Differences from torch DDP kept deliberately: no graph rebuilding, no
 find_unused_parameters machinery (CV models here are static), SUM+divide (works
