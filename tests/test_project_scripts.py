@@ -123,3 +123,19 @@ def test_detection_export_scripts(script, out, tmp_path):
     r = _run(script, "--img-size", "320", "--out", str(dest))
     assert r.returncode == 0, r.stderr[-1500:]
     assert dest.exists() or (tmp_path / f"{out}.torchscript.pt").exists()
+
+
+def test_all_project_scripts_compile():
+    """Syntax safety net over the whole projects/ tree (the per-script run
+    smokes above cover a subset; this catches a broken edit in any of the
+    ~110 thin CLI wrappers)."""
+    import pathlib
+    import py_compile
+
+    bad = []
+    for f in sorted(pathlib.Path("projects").rglob("*.py")):
+        try:
+            py_compile.compile(str(f), doraise=True)
+        except py_compile.PyCompileError as e:
+            bad.append(f"{f}: {e.msg}")
+    assert not bad, "\n".join(bad)
