@@ -139,3 +139,37 @@ def test_all_project_scripts_compile():
         except py_compile.PyCompileError as e:
             bad.append(f"{f}: {e.msg}")
     assert not bad, "\n".join(bad)
+
+
+def test_serve_endpoint_inprocess(tmp_path):
+    """Drive the FastAPI serving app in-process: /healthz and a /predict
+    round trip on a random PNG through a small model."""
+    import io
+    import sys
+
+    from fastapi.testclient import TestClient
+    from PIL import Image
+
+    sys.path.insert(0, "projects/others/deploy")
+    try:
+        from serve import create_app
+    finally:
+        sys.path.pop(0)
+
+    app = create_app(model_name="resnet18", num_classes=10, device="cpu",
+                     topk=3, image_size=64)
+    client = TestClient(app)
+    r = client.get("/healthz")
+    assert r.status_code == 200 and r.json()["status"] == "ok"
+
+    buf = io.BytesIO()
+    Image.new("RGB", (40, 52), (90, 120, 200)).save(buf, format="PNG")
+    r = client.post("/predict", content=buf.getvalue(),
+                    headers={"content-type": "image/png"})
+    assert r.status_code == 200
+    top = r.json()["topk"]
+    assert len(top) == 3
+    assert all(0 <= t["class"] < 10 and 0.0 <= t["score"] <= 1.0
+               for t in top)
+    scores = [t["score"] for t in top]
+    assert scores == sorted(scores, reverse=True)
