@@ -18,7 +18,8 @@ from fastapi import FastAPI, Request
 
 
 def create_app(model_name="resnet50", weights="", num_classes=1000,
-               device=None, topk=5, image_size=224):
+               device=None, topk=5, image_size=224, task="cls",
+               score_thresh=0.3):
     from deeplearning_amd.core.checkpoint import load_pretrained
     from deeplearning_amd.models import build_model
 
@@ -54,17 +55,28 @@ def create_app(model_name="resnet50", weights="", num_classes=1000,
 
     # raw body (python-multipart is not in the image, so no UploadFile):
     #   curl -X POST --data-binary @img.png http://host:8000/predict
+    def _forward(x):
+        with torch.no_grad():
+            if device.type == "cuda":
+                with torch.autocast("cuda", dtype=torch.bfloat16):
+                    return model([x[0]] if task == "det" else x)
+            return model([x[0]] if task == "det" else x)
+
     @app.post("/predict")
     async def predict(request: Request):
         data = await request.body()
         x = _preprocess(data)
-        with torch.no_grad():
-            if device.type == "cuda":
-                with torch.autocast("cuda", dtype=torch.bfloat16):
-                    logits = model(x)
-            else:
-                logits = model(x)
-        probs = logits.float().softmax(-1)[0]
+        out = _forward(x)
+        if task == "det":
+            det = out[0]
+            keep = det["scores"] >= score_thresh
+            return {"detections": [
+                {"box": [round(v, 2) for v in b],
+                 "class": int(l), "score": round(float(s), 4)}
+                for b, s, l in zip(det["boxes"][keep].tolist(),
+                                   det["scores"][keep].tolist(),
+                                   det["labels"][keep].tolist())]}
+        probs = out.float().softmax(-1)[0]
         k = min(topk, probs.numel())
         score, idx = probs.topk(k)
         return {"topk": [{"class": int(i), "score": float(s)}
@@ -83,11 +95,16 @@ def main():
     p.add_argument("--device", default=None)
     p.add_argument("--topk", type=int, default=5)
     p.add_argument("--img-size", type=int, default=224)
+    p.add_argument("--task", default="cls", choices=["cls", "det"],
+                   help="det: list-in/dict-out detectors (fasterrcnn_*, "
+                        "retinanet_*, fcos_*); yolo models have their own "
+                        "detect.py postprocess")
+    p.add_argument("--score-thresh", type=float, default=0.3)
     p.add_argument("--host", default="127.0.0.1")
     p.add_argument("--port", type=int, default=8000)
     args = p.parse_args()
     app = create_app(args.model, args.weights, args.num_classes, args.device,
-                     args.topk, args.img_size)
+                     args.topk, args.img_size, args.task, args.score_thresh)
     uvicorn.run(app, host=args.host, port=args.port)
 
 

@@ -173,3 +173,32 @@ def test_serve_endpoint_inprocess(tmp_path):
                for t in top)
     scores = [t["score"] for t in top]
     assert scores == sorted(scores, reverse=True)
+
+
+def test_serve_detection_task():
+    """Detection task on the serving endpoint returns a well-formed (possibly
+    empty) detections list from a random image."""
+    import io
+    import sys
+
+    from fastapi.testclient import TestClient
+    from PIL import Image
+
+    sys.path.insert(0, "projects/others/deploy")
+    try:
+        import serve
+    finally:
+        sys.path.pop(0)
+
+    app = serve.create_app(model_name="retinanet_resnet50_fpn",
+                           num_classes=5, device="cpu", image_size=256,
+                           task="det", score_thresh=0.0)
+    client = TestClient(app)
+    buf = io.BytesIO()
+    Image.new("RGB", (300, 200), (40, 90, 160)).save(buf, format="PNG")
+    r = client.post("/predict", content=buf.getvalue(),
+                    headers={"content-type": "image/png"})
+    assert r.status_code == 200
+    dets = r.json()["detections"]
+    for d in dets:
+        assert len(d["box"]) == 4 and isinstance(d["class"], int)
