@@ -37,6 +37,16 @@ def create_app(model_name="resnet50", weights="", num_classes=1000,
     app.state.model = model
     app.state.device = device
 
+    # prometheus metrics (prometheus_client ships in the image); a private
+    # registry so repeated create_app() calls (tests) don't collide
+    from prometheus_client import (CONTENT_TYPE_LATEST, CollectorRegistry,
+                                   Counter, Histogram, generate_latest)
+    registry = CollectorRegistry()
+    reqs = Counter("dla_serve_requests_total", "predict requests",
+                   ["status"], registry=registry)
+    lat = Histogram("dla_serve_latency_seconds", "predict latency",
+                    registry=registry)
+
     def _preprocess(data: bytes) -> torch.Tensor:
         from PIL import Image
 
@@ -62,11 +72,25 @@ def create_app(model_name="resnet50", weights="", num_classes=1000,
                     return model([x[0]] if task == "det" else x)
             return model([x[0]] if task == "det" else x)
 
+    @app.get("/metrics")
+    def metrics():
+        from fastapi import Response
+        return Response(generate_latest(registry),
+                        media_type=CONTENT_TYPE_LATEST)
+
     @app.post("/predict")
     async def predict(request: Request):
+        import time
+        t0 = time.perf_counter()
         data = await request.body()
-        x = _preprocess(data)
-        out = _forward(x)
+        try:
+            x = _preprocess(data)
+            out = _forward(x)
+        except Exception:
+            reqs.labels(status="error").inc()
+            raise
+        reqs.labels(status="ok").inc()
+        lat.observe(time.perf_counter() - t0)
         if task == "det":
             det = out[0]
             keep = det["scores"] >= score_thresh

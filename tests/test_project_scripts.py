@@ -202,3 +202,7 @@ def test_serve_detection_task():
     dets = r.json()["detections"]
     for d in dets:
         assert len(d["box"]) == 4 and isinstance(d["class"], int)
+    m = client.get("/metrics")
+    assert m.status_code == 200
+    assert 'dla_serve_requests_total{status="ok"} 1.0' in m.text
+    assert "dla_serve_latency_seconds" in m.text
