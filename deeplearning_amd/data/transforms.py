@@ -101,9 +101,15 @@ class Normalize:
         return (t - self.mean) / self.std
 
 
-def classification_train_transform(img_size=224):
-    return Compose([RandomResizedCrop(img_size), RandomHorizontalFlip(),
-                    ToTensor(), Normalize()])
+def classification_train_transform(img_size=224, rand_augment=False,
+                                   ra_magnitude=9.0, ra_mstd=0.5):
+    """rand_augment=True inserts RandAugment between the crop and flip
+    (swin recipe rand-m9-mstd0.5-inc1; data/autoaugment.py)."""
+    ops = [RandomResizedCrop(img_size), RandomHorizontalFlip()]
+    if rand_augment:
+        from .autoaugment import RandAugment
+        ops.append(RandAugment(magnitude=ra_magnitude, mstd=ra_mstd))
+    return Compose(ops + [ToTensor(), Normalize()])
 
 
 def classification_eval_transform(img_size=224, crop_pct=0.875):

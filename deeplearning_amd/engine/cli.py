@@ -72,6 +72,10 @@ def classification_argparser(default_model: str, **defaults):
     p.add_argument("--output", default="runs")
     p.add_argument("--name", default=defaults.get("name", "exp"))
     p.add_argument("--syncbn", action="store_true")
+    p.add_argument("--rand-augment", action="store_true",
+                   help="RandAugment rand-m9-mstd0.5 on the train pipeline "
+                        "(swin recipe; data/autoaugment.py)")
+    p.add_argument("--ra-magnitude", type=float, default=9.0)
     p.add_argument("--mixup", action="store_true",
                    help="Mixup/CutMix soft-target training "
                         "(swin dataLoader/build.py:90-96)")
@@ -94,7 +98,10 @@ def build_classification_loaders(args):
     if args.data_path:
         tp, tl, vp, vl, classes = read_split_data(args.data_path)
         train_ds = ClassificationDataset(
-            tp, tl, classification_train_transform(args.img_size))
+            tp, tl, classification_train_transform(
+                args.img_size,
+                rand_augment=getattr(args, "rand_augment", False),
+                ra_magnitude=getattr(args, "ra_magnitude", 9.0)))
         val_ds = ClassificationDataset(
             vp, vl, classification_eval_transform(args.img_size))
     else:

@@ -233,3 +233,37 @@ def test_zip_image_dataset(tmp_path):
     x, y = ds[2]
     assert x.shape == (3, 8, 8) and y == 2
     assert len(ds) == 3
+
+
+def test_rand_augment_policy():
+    """RandAugment preserves size/mode, is deterministic under a seeded
+    random module, and every op in the policy runs standalone."""
+    import random
+
+    from PIL import Image
+
+    from deeplearning_amd.data.autoaugment import OPS, RandAugment
+    from deeplearning_amd.data.transforms import \
+        classification_train_transform
+
+    img = Image.new("RGB", (48, 40))
+    px = img.load()
+    for i in range(48):
+        for j in range(40):
+            px[i, j] = (5 * i % 256, 6 * j % 256, (i + j) % 256)
+
+    for name, fn in OPS:
+        out = fn(img, 9.0)
+        assert out.size == img.size and out.mode == "RGB", name
+
+    ra = RandAugment(num_ops=2, magnitude=9, mstd=0.5)
+    random.seed(7)
+    a = ra(img)
+    random.seed(7)
+    b = ra(img)
+    assert list(a.getdata()) == list(b.getdata())
+    assert a.size == img.size
+
+    t = classification_train_transform(32, rand_augment=True)
+    x = t(img)
+    assert x.shape == (3, 32, 32)
