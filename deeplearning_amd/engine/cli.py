@@ -170,6 +170,9 @@ def classification_train_main(args) -> dict:
                             img_size=args.img_size)
     except TypeError:
         model = build_model(args.model, num_classes=args.num_classes)
+    from ..core.complexity import complexity_str
+    logger.info(f"model {args.model}: "
+                f"{complexity_str(model, torch.randn(1, 3, args.img_size, args.img_size))}")
     model = model.to(device)
     if args.weights:
         load_pretrained(model, args.weights, logger=logger)

@@ -90,3 +90,31 @@ def test_tensorboard_writer_noop_off_rank(tmp_path):
     w.flush()
     w.close()
     assert w.writer is None
+
+
+def test_complexity_counts_match_published():
+    """Hook-based MAC counter vs published model complexities (swin main.py
+    parity: n_parameters + flops logged at startup)."""
+    from deeplearning_amd.core.complexity import (complexity_str,
+                                                  count_params,
+                                                  estimate_macs)
+    from deeplearning_amd.models import build_model
+
+    x = torch.randn(1, 3, 224, 224)
+    m = build_model("resnet18", num_classes=1000)
+    assert abs(count_params(m) / 1e6 - 11.69) < 0.02
+    assert abs(estimate_macs(m, x) / 1e9 - 1.814) < 0.01
+    m50 = build_model("resnet50", num_classes=1000)
+    assert abs(estimate_macs(m50, x) / 1e9 - 4.089) < 0.01
+    s = complexity_str(m, x)
+    assert "params 11.7M" in s and "MACs 1.81G" in s
+
+
+def test_complexity_batch_scales_linearly():
+    from deeplearning_amd.core.complexity import estimate_macs
+    from deeplearning_amd.models import build_model
+
+    m = build_model("mnist_fcn", num_classes=10)
+    m1 = estimate_macs(m, torch.randn(1, 1, 28, 28))
+    m4 = estimate_macs(m, torch.randn(4, 1, 28, 28))
+    assert m4 == 4 * m1 > 0
