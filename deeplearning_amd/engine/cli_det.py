@@ -131,6 +131,8 @@ def det_train_main(args, model_kwargs=None) -> dict:
 
     model = build_model(args.model, num_classes=args.num_classes,
                         **(model_kwargs or {})).to(device)
+    from ..core.complexity import count_params
+    logger.info(f"model {args.model}: params {count_params(model) / 1e6:.1f}M")
     if get_world_size() > 1:
         from ..parallel import wrap_data_parallel
         model = wrap_data_parallel(model)

@@ -109,6 +109,8 @@ def seg_train_main(args) -> dict:
                            dist_rank=get_rank())
 
     model = build_model(args.model, num_classes=args.num_classes).to(device)
+    from ..core.complexity import count_params
+    logger.info(f"model {args.model}: params {count_params(model) / 1e6:.1f}M")
     if get_world_size() > 1:
         from ..parallel import wrap_data_parallel
         model = wrap_data_parallel(model)

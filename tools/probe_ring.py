@@ -3,10 +3,12 @@ fat-N conv1x1 fwd shapes. Run twice (env on / off) to A/B — the gate is
 latched at first kernel call."""
 import os
 import sys
+from pathlib import Path
 
 import torch
 
-from deeplearning_amd.ops.conv1x1 import ext
+sys.path.insert(0, str(Path(__file__).resolve().parents[1]))
+from deeplearning_amd.ops.conv1x1 import ext  # noqa: E402
 
 
 def t_ms(fn, iters=50):

@@ -1,8 +1,12 @@
 """Time conv3x3_wgrad (ky-row-merged taps) vs MIOpen wrw at resnet50 3x3
 shapes, batch 64. One line per shape -> gpurun_out/wgrad3.txt friendly."""
+import sys
+from pathlib import Path
+
 import torch
 
-from deeplearning_amd.ops.conv1x1 import ext
+sys.path.insert(0, str(Path(__file__).resolve().parents[1]))
+from deeplearning_amd.ops.conv1x1 import ext  # noqa: E402
 
 
 def t_ms(fn, iters=30):
