@@ -206,3 +206,70 @@ def test_serve_detection_task():
     assert m.status_code == 200
     assert 'dla_serve_requests_total{status="ok"} 1.0' in m.text
     assert "dla_serve_latency_seconds" in m.text
+
+
+def test_serve_microbatcher_batches_concurrent_requests():
+    """MicroBatcher must run concurrent infer() calls as ONE forward and
+    return each caller its own slice; errors propagate to all waiters."""
+    import asyncio
+    import sys
+
+    import torch
+
+    sys.path.insert(0, "projects/others/deploy")
+    try:
+        from serve import MicroBatcher
+    finally:
+        sys.path.pop(0)
+
+    calls = []
+
+    def fwd(x):
+        calls.append(x.shape[0])
+        return x * 2
+
+    async def drive():
+        b = MicroBatcher(fwd, max_batch=4, max_wait_ms=50)
+        outs = await asyncio.gather(*[
+            b.infer(torch.full((1, 3), float(i))) for i in range(4)])
+        return outs
+
+    outs = asyncio.run(drive())
+    assert calls == [4]  # one batched forward for 4 concurrent requests
+    for i, o in enumerate(outs):
+        assert o.shape == (1, 3) and float(o[0, 0]) == 2.0 * i
+
+    async def drive_err():
+        def bad(x):
+            raise RuntimeError("boom")
+        b = MicroBatcher(bad, max_batch=2, max_wait_ms=5)
+        import pytest as _pt
+        with _pt.raises(RuntimeError):
+            await b.infer(torch.zeros(1, 3))
+
+    asyncio.run(drive_err())
+
+
+def test_serve_endpoint_with_microbatching():
+    """End-to-end /predict through the batcher (single request path)."""
+    import io
+    import sys
+
+    from fastapi.testclient import TestClient
+    from PIL import Image
+
+    sys.path.insert(0, "projects/others/deploy")
+    try:
+        import serve
+    finally:
+        sys.path.pop(0)
+
+    app = serve.create_app(model_name="resnet18", num_classes=10,
+                           device="cpu", topk=2, image_size=64,
+                           max_batch=4, batch_wait_ms=1.0)
+    client = TestClient(app)
+    buf = io.BytesIO()
+    Image.new("RGB", (40, 52), (10, 200, 90)).save(buf, format="PNG")
+    r = client.post("/predict", content=buf.getvalue(),
+                    headers={"content-type": "image/png"})
+    assert r.status_code == 200 and len(r.json()["topk"]) == 2
