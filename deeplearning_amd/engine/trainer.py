@@ -14,7 +14,6 @@ import torch
 import torch.nn as nn
 
 from ..core import (MetricLogger, SmoothedValue, create_logger, is_main_process)
-from ..core.checkpoint import unwrap_model
 from .metrics import accuracy
 
 

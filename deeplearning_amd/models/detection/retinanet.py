@@ -8,7 +8,6 @@ framework's HIP kernel (ops.sigmoid_focal_loss), NMS is the HIP batched-NMS.
 from __future__ import annotations
 
 import math
-from collections import OrderedDict
 
 import torch
 import torch.nn.functional as F

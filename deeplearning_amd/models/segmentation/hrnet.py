@@ -12,7 +12,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from ...ops import BatchNorm2d, add_relu
+from ...ops import BatchNorm2d
 from ..classification.resnet import BasicBlock, Bottleneck
 from ..registry import register_model
 

@@ -11,7 +11,7 @@ import torch
 
 from deeplearning_amd.core.checkpoint import load_pretrained
 from deeplearning_amd.core.env import select_device
-from deeplearning_amd.engine.cli_det import SyntheticDetection, \
+from deeplearning_amd.engine.cli_det import \
     build_det_dataset, det_argparser
 from deeplearning_amd.engine.det_eval import DetEvaluator
 from deeplearning_amd.models import build_model

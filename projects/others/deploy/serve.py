@@ -60,7 +60,7 @@ class MicroBatcher:
                     break
             try:
                 out = self.forward_fn(torch.cat([i[0] for i in items]))
-                for (slot, f), o in zip(items, out.split(1)):
+                for (_, f), o in zip(items, out.split(1)):
                     if not f.done():
                         f.set_result(o)
             except Exception as e:  # propagate to every waiter

@@ -8,7 +8,6 @@ from pathlib import Path
 sys.path.insert(0, str(Path(__file__).resolve().parents[3]))
 
 import argparse
-import math
 
 import torch
 from torch.utils.data import DataLoader
