@@ -22,6 +22,8 @@ class DataPrefetcher:
     def _to_device(self, x):
         if torch.is_tensor(x):
             return x.to(self.device, non_blocking=True)
+        if isinstance(x, tuple) and hasattr(x, "_make"):  # namedtuple
+            return type(x)._make(self._to_device(v) for v in x)
         if isinstance(x, (list, tuple)):
             return type(x)(self._to_device(v) for v in x)
         if isinstance(x, dict):
