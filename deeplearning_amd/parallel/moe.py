@@ -19,6 +19,27 @@ import torch.nn.functional as F
 from ..core.dist import get_world_size, is_dist
 
 
+class _AllToAll(torch.autograd.Function):
+    """Autograd-aware all_to_all_single (equal splits): raw dist collectives
+    are invisible to autograd, so the EP dispatch/return would otherwise cut
+    the graph — expert weights and the router would get NO gradient in EP
+    mode. Backward of an equal-split all-to-all is the same all-to-all on
+    the gradient (send/recv roles swap symmetrically)."""
+
+    @staticmethod
+    def forward(ctx, x, group):
+        ctx.group = group
+        out = torch.empty_like(x)
+        dist.all_to_all_single(out, x.contiguous(), group=group)
+        return out
+
+    @staticmethod
+    def backward(ctx, grad):
+        g = torch.empty_like(grad)
+        dist.all_to_all_single(g, grad.contiguous(), group=ctx.group)
+        return g, None
+
+
 class CosineRouter(nn.Module):
     """Cosine-similarity router (Swin-MoE style) with learnable temperature."""
 
@@ -130,17 +151,14 @@ class MoEMlp(nn.Module):
         buf[flat_expert, slot] = tokens[flat_tok]
         # send experts to their owner rank: rank r owns experts [r*L,(r+1)*L)
         send = buf.reshape(W, L * cap, C)
-        recv = torch.empty_like(send)
-        dist.all_to_all_single(recv, send, group=self.ep_group)
+        recv = _AllToAll.apply(send, self.ep_group)
         # recv: [W senders][L local experts * cap] -> per local expert batch
         recv = recv.reshape(W, L, cap, C).transpose(0, 1) \
             .reshape(L, W * cap, C)
         h = F.gelu(torch.bmm(recv, self.w1) + self.b1[:, None])
         y = torch.bmm(h, self.w2) + self.b2[:, None]      # L, W*cap, C
         y = y.reshape(L, W, cap, C).transpose(0, 1).reshape(W, L * cap, C)
-        back = torch.empty_like(y)
-        dist.all_to_all_single(back, y, group=self.ep_group)
-        back = back.reshape(E, cap, C)
+        back = _AllToAll.apply(y, self.ep_group).reshape(E, cap, C)
         contrib = (back[flat_expert, slot] *
                    flat_gate[:, None]).to(out.dtype)
         out.index_add_(0, flat_tok, contrib)

@@ -83,6 +83,13 @@ def _ep_worker(rank, world, port, q):
     out_ep, _ = ep(x)
     out_ref, _ = ref(x)
     err = (out_ep - out_ref).abs().max().item()
+    # EP backward must reach the local experts and the router (the raw
+    # collective would silently cut the graph here)
+    out_ep.sum().backward()
+    assert ep.router.weight.grad is not None
+    grads = [ep.w1.grad, ep.w2.grad]
+    assert any(g is not None and g.abs().sum() > 0 for g in grads), \
+        "EP expert weights got no gradient"
     if rank == 0:
         q.put(err)
     else:
