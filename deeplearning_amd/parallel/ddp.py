@@ -56,6 +56,10 @@ class BucketedDataParallel(nn.Module):
         if self.enabled and broadcast_params:
             with torch.no_grad():
                 for t in list(module.parameters()) + list(module.buffers()):
+                    # expert-parallel params are rank-LOCAL (each rank owns
+                    # its expert shard): neither broadcast nor all-reduce
+                    if getattr(t, "expert", False):
+                        continue
                     dist.broadcast(t, src=0, group=self.pg)
 
         self.buckets: list[_Bucket] = []
@@ -64,7 +68,11 @@ class BucketedDataParallel(nn.Module):
             self._build_buckets(bucket_cap_mb, grad_dtype)
 
     def _build_buckets(self, cap_mb: float, grad_dtype):
-        params = [p for p in self.module.parameters() if p.requires_grad]
+        # skip expert-marked params (parallel/moe.py): each rank owns its own
+        # expert shard, so averaging their grads across ranks would corrupt
+        # EP training (reference _ddp_params_and_buffers_to_ignore pattern)
+        params = [p for p in self.module.parameters()
+                  if p.requires_grad and not getattr(p, "expert", False)]
         cap = int(cap_mb * 1024 * 1024)
         cur, cur_bytes = [], 0
         buckets_params = []
