@@ -14,6 +14,7 @@ import torch
 import torch.nn as nn
 
 from ..core import (MetricLogger, SmoothedValue, create_logger, is_main_process)
+from ..core.checkpoint import unwrap_model
 from .metrics import accuracy
 
 
@@ -27,13 +28,16 @@ def train_one_epoch(model, criterion, data_loader, optimizer, device, epoch,
                     lr_scheduler=None, accum_steps: int = 1, clip_grad: float = 0.0,
                     amp: bool = True, amp_dtype: torch.dtype = torch.bfloat16,
                     ema=None, print_freq: int = 50, logger=None,
-                    mixup_fn=None) -> dict:
+                    mixup_fn=None, aux_loss_weight: float = 0.01) -> dict:
     model.train()
     metric = MetricLogger(logger=logger)
     metric.add_meter("lr", SmoothedValue(window_size=1, fmt="{value:.6f}"))
     optimizer.zero_grad(set_to_none=True)
     finalize = getattr(model, "finalize", None)
     no_sync = getattr(model, "no_sync", None)
+    # MoE models expose a load-balance aux loss (swin_moe); the reference
+    # adds it to the criterion (swin-moe main: cfg TRAIN.MOE.AUX_LOSS_WEIGHT)
+    aux_fn = getattr(unwrap_model(model), "aux_loss", None)
 
     for it, (samples, targets) in enumerate(
             metric.log_every(data_loader, print_freq, f"Epoch [{epoch}]")):
@@ -45,7 +49,10 @@ def train_one_epoch(model, criterion, data_loader, optimizer, device, epoch,
         is_accum = (it + 1) % accum_steps != 0
         with _autocast(amp, amp_dtype):
             outputs = model(samples)
-            loss = criterion(outputs, targets) / accum_steps
+            loss = criterion(outputs, targets)
+            if aux_fn is not None:
+                loss = loss + aux_loss_weight * aux_fn()
+            loss = loss / accum_steps
         if is_accum and no_sync is not None:
             with no_sync():
                 loss.backward()

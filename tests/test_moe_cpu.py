@@ -175,3 +175,41 @@ def test_ddp_excludes_expert_params():
         assert ok_bcast, f"rank {rank}: expert param was broadcast"
         assert not_in_bucket, f"rank {rank}: expert param bucketed"
         assert shared_synced, f"rank {rank}: shared grads not synced"
+
+
+def test_trainer_adds_moe_aux_loss():
+    """train_one_epoch must fold model.aux_loss() into the objective
+    (reference swin-moe adds cfg TRAIN.MOE.AUX_LOSS_WEIGHT * l_aux)."""
+    import torch.nn as nn
+    from torch.utils.data import DataLoader, TensorDataset
+
+    from deeplearning_amd.engine.trainer import train_one_epoch
+
+    class Tiny(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.fc = nn.Linear(4, 3)
+            self.aux_calls = 0
+
+        def forward(self, x):
+            return self.fc(x)
+
+        def aux_loss(self):
+            self.aux_calls += 1
+            return self.fc.weight.pow(2).sum()  # differentiable, nonzero
+
+    torch.manual_seed(0)
+    ds = TensorDataset(torch.randn(8, 4), torch.randint(0, 3, (8,)))
+    m = Tiny()
+    m2 = Tiny()
+    m2.load_state_dict(m.state_dict())
+    opt = torch.optim.SGD(m.parameters(), lr=0.1)
+    opt2 = torch.optim.SGD(m2.parameters(), lr=0.1)
+    loader = DataLoader(ds, batch_size=4)
+    train_one_epoch(m, nn.CrossEntropyLoss(), loader, opt, "cpu", 0,
+                    amp=False, aux_loss_weight=1.0)
+    assert m.aux_calls == 2  # called every iteration
+    train_one_epoch(m2, nn.CrossEntropyLoss(), loader, opt2, "cpu", 0,
+                    amp=False, aux_loss_weight=0.0)
+    # a nonzero aux weight must change the training trajectory
+    assert not torch.allclose(m.fc.weight, m2.fc.weight)
