@@ -34,15 +34,17 @@ class Loggers:
     def log_metrics(self, metrics: dict, step: int):
         if self.csv_path is not None:
             keys = ["step"] + sorted(metrics)
-            new_file = not self.csv_path.exists() or self._csv_keys != keys
-            mode = "w" if new_file and self._csv_keys != keys and \
-                not self.csv_path.exists() else "a"
-            with open(self.csv_path, mode, newline="") as f:
+            exists = self.csv_path.exists()
+            # header on a fresh file or when the metric set changes mid-run;
+            # resuming into an existing file appends without re-writing it
+            write_header = not exists or (self._csv_keys is not None and
+                                          self._csv_keys != keys)
+            with open(self.csv_path, "a" if exists else "w", newline="") as f:
                 w = csv.writer(f)
-                if new_file:
+                if write_header:
                     w.writerow(keys)
-                    self._csv_keys = keys
                 w.writerow([step] + [metrics[k] for k in sorted(metrics)])
+            self._csv_keys = keys
         if self.tb is not None:
             for k, v in metrics.items():
                 self.tb.add_scalar(k, v, step)
