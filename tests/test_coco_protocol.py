@@ -237,3 +237,109 @@ def test_matcher_threeway_equivalence_fuzz(data):
     assert np.array_equal((m_py & ~ig_py).numpy(), m_np & ~ig_np)
     assert np.array_equal(ig_py.numpy(), ig_np)
     assert ngt_py == ngt_np
+
+
+# ---------------------------------------------------------------------------
+# Area-range APs (small/medium/large) + AR@k tiers vs a literal translation
+# of pycocotools evaluateImg/accumulate with areaRng / maxDets semantics.
+# ---------------------------------------------------------------------------
+def _areas(b):
+    return (b[:, 2] - b[:, 0]) * (b[:, 3] - b[:, 1]) if len(b) else \
+        np.zeros(0)
+
+
+def _evaluate_img_arng(dt, dts, gt, crowd, thrs, max_dets, lo, hi):
+    """COCOeval.evaluateImg, bbox, one areaRng — literal loop translation."""
+    ga = _areas(gt)
+    gt_ig0 = crowd.astype(bool) | (ga < lo) | (ga > hi)
+    gind = np.argsort(gt_ig0, kind="mergesort")           # ignore-last
+    gt_s, crowd_s, ig_s = gt[gind], crowd.astype(bool)[gind], gt_ig0[gind]
+    order = np.argsort(-dts, kind="mergesort")[:max_dets]
+    dt_s, ds_s = dt[order], dts[order]
+    T, D, G = len(thrs), len(dt_s), len(gt_s)
+    ious = _iou(dt_s, gt_s, crowd_s)
+    dtm = -np.ones((T, D), dtype=int)
+    gtm = -np.ones((T, G), dtype=int)
+    dt_ig = np.zeros((T, D), dtype=bool)
+    for t, thr in enumerate(thrs):
+        for d in range(D):
+            iou0 = min(thr, 1 - 1e-10)
+            m = -1
+            for g in range(G):
+                if gtm[t, g] >= 0 and not crowd_s[g]:
+                    continue
+                if m > -1 and not ig_s[m] and ig_s[g]:
+                    break
+                if ious[d, g] < iou0:
+                    continue
+                iou0 = ious[d, g]
+                m = g
+            if m == -1:
+                continue
+            dt_ig[t, d] = bool(ig_s[m])
+            dtm[t, d] = m
+            gtm[t, m] = d
+    da = _areas(dt_s)
+    out = (da < lo) | (da > hi)
+    dt_ig = dt_ig | ((dtm < 0) & out[None, :])
+    return dtm >= 0, dt_ig, ds_s, int((~gt_ig0).sum())
+
+
+def _ref_range_map(preds, gts, thrs, max_dets, lo, hi):
+    per_class = {}
+    for pred, gt in zip(preds, gts):
+        classes = torch.cat([pred["labels"], gt["labels"]]).unique()
+        for c in classes.tolist():
+            dm = pred["labels"] == c
+            gm = gt["labels"] == c
+            r = _evaluate_img_arng(
+                pred["boxes"][dm].numpy(), pred["scores"][dm].numpy(),
+                gt["boxes"][gm].numpy(),
+                gt["iscrowd"][gm].numpy(), thrs, max_dets, lo, hi)
+            per_class.setdefault(c, []).append(r)
+    aps = [np.mean(a) for a in
+           (_accumulate(items, thrs) for items in per_class.values())
+           if a is not None]
+    return float(np.mean(aps)) if aps else 0.0
+
+
+def _ref_ar(preds, gts, thrs, max_dets, k):
+    per_class = {}
+    for pred, gt in zip(preds, gts):
+        classes = torch.cat([pred["labels"], gt["labels"]]).unique()
+        for c in classes.tolist():
+            dm = pred["labels"] == c
+            gm = gt["labels"] == c
+            r = _evaluate_img_arng(
+                pred["boxes"][dm].numpy(), pred["scores"][dm].numpy(),
+                gt["boxes"][gm].numpy(),
+                gt["iscrowd"][gm].numpy(), thrs, max_dets, 0.0, 1e10)
+            per_class.setdefault(c, []).append(r)
+    recs = []
+    for items in per_class.values():
+        n_gt = sum(it[3] for it in items)
+        if n_gt == 0:
+            continue
+        for t in range(len(thrs)):
+            tp = sum(int((it[0][t, :k] & ~it[1][t, :k]).sum())
+                     for it in items)
+            recs.append(tp / n_gt)
+    return float(np.mean(recs)) if recs else 0.0
+
+
+@pytest.mark.parametrize("seed", [0, 3])
+def test_area_range_and_ar_match_pycocotools_protocol(seed):
+    from deeplearning_amd.engine.det_eval import AREA_RNG, DetEvaluator
+
+    preds, gts = _synthetic_set(seed=seed)
+    ev = DetEvaluator()
+    ev.update(preds, gts)
+    stats = ev.summarize()
+    thrs = np.array(COCO_IOU_THRS)
+    for name, (lo, hi) in AREA_RNG.items():
+        ref = _ref_range_map(preds, gts, thrs, 100, lo, hi)
+        assert abs(stats[f"mAP_{name}"] - ref) < 1e-6, \
+            (name, stats[f"mAP_{name}"], ref)
+    for k in (1, 10, 100):
+        ref = _ref_ar(preds, gts, thrs, 100, k)
+        assert abs(stats[f"AR{k}"] - ref) < 1e-6, (k, stats[f"AR{k}"], ref)
