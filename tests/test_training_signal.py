@@ -41,8 +41,8 @@ def test_retinanet_loss_decreases():
         losses = m(imgs, targets)
         loss = sum(losses.values())
         if i == 0:
-            first = float(loss)
-        last = float(loss)
+            first = float(loss.detach())
+        last = float(loss.detach())
         opt.zero_grad()
         loss.backward()
         torch.nn.utils.clip_grad_norm_(m.parameters(), 10.0)
@@ -64,8 +64,8 @@ def test_yolox_loss_decreases():
         losses = m(x, targets)
         loss = sum(losses.values())
         if i == 0:
-            first = float(loss)
-        last = float(loss)
+            first = float(loss.detach())
+        last = float(loss.detach())
         opt.zero_grad()
         loss.backward()
         opt.step()
