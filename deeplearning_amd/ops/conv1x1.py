@@ -62,7 +62,7 @@ class _Conv1x1Fn(torch.autograd.Function):
         dy2d = dy2d.contiguous()
         if dy2d.dtype != torch.bfloat16:
             dy2d = dy2d.to(torch.bfloat16)
-        dx = dw = None
+        dx = dw = db = None
         if ctx.needs_input_grad[0]:
             # LDS-tiled transpose kernel (torch's strided copy is ~15x slower)
             wt = ext().transpose2d(wb)  # [K,N] bf16
@@ -74,7 +74,11 @@ class _Conv1x1Fn(torch.autograd.Function):
             dw = ext().conv1x1_wgrad(dy2d, xb)  # fp32 [N,K]
             if dw.dtype != ctx.w_dtype:
                 dw = dw.to(ctx.w_dtype)
-        return dx, dw, None, None, None, None, None, None
+        if ctx.needs_input_grad[2]:
+            # bias rides the fused epilogue in forward; its grad is the
+            # per-channel column sum of dy
+            db = dy2d.float().sum(0)
+        return dx, dw, db, None, None, None, None, None, None
 
 
 class _BNFromStatsFn(torch.autograd.Function):

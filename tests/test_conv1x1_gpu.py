@@ -434,3 +434,25 @@ def test_conv3x3_dgrad_wgrad_parity():
     assert relerr(x.grad, xr.grad) < 8e-2, float(relerr(x.grad, xr.grad))
     assert relerr(w.grad, wr.grad) < 8e-2, float(relerr(w.grad, wr.grad))
     assert w.grad.dtype == torch.float32
+
+
+@pytest.mark.gpu
+def test_conv1x1_bias_grad():
+    """A biased 1x1 conv through the fused path must train its bias: the
+    epilogue applies it in forward, so backward owes db = dy.sum over rows."""
+    torch.manual_seed(5)
+    conv = torch.nn.Conv2d(64, 128, 1, bias=True).cuda().bfloat16()
+    x = torch.randn(2, 64, 8, 8, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True).to(memory_format=torch.channels_last)
+    from deeplearning_amd.ops.conv1x1 import can_fuse_conv1x1, conv1x1
+    assert can_fuse_conv1x1(x, conv)
+    y = conv1x1(x, conv)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    x32 = x.detach().float().requires_grad_()
+    b32 = conv.bias.detach().float().requires_grad_()
+    y32 = torch.nn.functional.conv2d(x32, conv.weight.detach().float(), b32)
+    y32.backward(dy.float())
+    assert conv.bias.grad is not None
+    assert torch.allclose(conv.bias.grad.float(), b32.grad, rtol=0.05,
+                          atol=0.5)
