@@ -343,3 +343,43 @@ def test_area_range_and_ar_match_pycocotools_protocol(seed):
     for k in (1, 10, 100):
         ref = _ref_ar(preds, gts, thrs, 100, k)
         assert abs(stats[f"AR{k}"] - ref) < 1e-6, (k, stats[f"AR{k}"], ref)
+
+
+@settings(max_examples=80, deadline=None)
+@given(data=st.data())
+def test_ranged_matcher_fuzz_vs_literal_loop(data):
+    """match_image_ranged + det-range ignore vs the literal pycocotools
+    evaluateImg translation, on tie-heavy quantized inputs with random
+    crowd flags and a random area range."""
+    from deeplearning_amd.engine.det_eval import (_box_areas,
+                                                  match_image_ranged)
+
+    D = data.draw(st.integers(0, 7), label="D")
+    G = data.draw(st.integers(0, 5), label="G")
+    dt = torch.tensor([_qbox(data.draw) for _ in range(D)],
+                      dtype=torch.float64).reshape(D, 4)
+    gt = torch.tensor([_qbox(data.draw) for _ in range(G)],
+                      dtype=torch.float64).reshape(G, 4)
+    crowd = torch.tensor([data.draw(st.booleans()) for _ in range(G)],
+                         dtype=torch.bool)
+    lo = float(data.draw(st.sampled_from([0, 4, 16, 36])))
+    hi = float(data.draw(st.sampled_from([16, 36, 64, 1e10])))
+    perm = torch.randperm(D, generator=torch.Generator().manual_seed(
+        data.draw(st.integers(0, 999), label="seed")))
+    scores = (0.1 + 0.8 * perm.double() / max(D, 1)).clamp(max=0.99)
+
+    order = scores.argsort(descending=True, stable=True)
+    dt_s, sc_s = dt[order], scores[order]
+    ga = _box_areas(gt)
+    extra = (ga < lo) | (ga > hi) if G else crowd.clone()
+    m, ig, sc, n = match_image_ranged(dt_s, sc_s, gt, crowd, extra,
+                                      COCO_IOU_THRS, 100, presorted=True)
+    da = _box_areas(dt_s)
+    ig = ig | (~m & ((da < lo) | (da > hi))[None, :])
+
+    r_m, r_ig, r_sc, r_n = _evaluate_img_arng(
+        dt.numpy(), scores.numpy(), gt.numpy(),
+        crowd.numpy().astype(np.int64), np.array(COCO_IOU_THRS), 100, lo, hi)
+    assert n == r_n
+    assert np.array_equal((m & ~ig).numpy(), r_m & ~r_ig)
+    assert np.array_equal(ig.numpy(), r_ig)
