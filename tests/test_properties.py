@@ -22,7 +22,7 @@ def boxes_strategy(n_min=1, n_max=12):
     return st.composite(lambda draw: build(draw))()
 
 
-@settings(max_examples=40, deadline=None)
+@settings(max_examples=40, deadline=None, derandomize=True)
 @given(boxes_strategy())
 def test_iou_properties(boxes):
     iou = box_iou(boxes, boxes)
@@ -32,7 +32,7 @@ def test_iou_properties(boxes):
     assert float(iou.min()) >= -1e-6 and float(iou.max()) <= 1 + 1e-6
 
 
-@settings(max_examples=40, deadline=None)
+@settings(max_examples=40, deadline=None, derandomize=True)
 @given(boxes_strategy())
 def test_box_coder_roundtrip_property(boxes):
     coder = BoxCoder()
@@ -42,7 +42,7 @@ def test_box_coder_roundtrip_property(boxes):
     assert torch.allclose(decoded, boxes, atol=1e-2)
 
 
-@settings(max_examples=30, deadline=None)
+@settings(max_examples=30, deadline=None, derandomize=True)
 @given(boxes_strategy(n_min=2, n_max=10),
        st.floats(min_value=0.1, max_value=0.9))
 def test_nms_properties(boxes, thr):
@@ -61,7 +61,7 @@ def test_nms_properties(boxes, thr):
     assert 0 in keep.tolist()
 
 
-@settings(max_examples=30, deadline=None)
+@settings(max_examples=30, deadline=None, derandomize=True)
 @given(st.integers(0, 64), st.integers(2, 50))
 def test_voc_ap_bounds(extra_pos, n_det):
     from deeplearning_amd.engine.det_eval import voc_ap

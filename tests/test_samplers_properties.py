@@ -12,7 +12,7 @@ from deeplearning_amd.data.samplers import (GroupedBatchSampler,
                                             create_aspect_ratio_groups)
 
 
-@settings(max_examples=60, deadline=None)
+@settings(max_examples=60, deadline=None, derandomize=True)
 @given(
     n=st.integers(1, 40),
     batch_size=st.integers(1, 6),
@@ -69,7 +69,7 @@ def test_subset_random_sampler_epoch_determinism():
 
 @given(st.lists(st.floats(0.2, 5.0), min_size=1, max_size=50),
        st.integers(0, 4))
-@settings(max_examples=40, deadline=None)
+@settings(max_examples=40, deadline=None, derandomize=True)
 def test_aspect_ratio_groups_monotone(ratios, k):
     gids = create_aspect_ratio_groups(ratios, k=k)
     assert len(gids) == len(ratios)
