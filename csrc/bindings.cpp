@@ -95,12 +95,10 @@ torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor B);
 // my_add.cpp (custom-op export tutorial)
 torch::Tensor my_add(torch::Tensor a, torch::Tensor b);
 // cocoeval.cpp (CPU)
-std::vector<torch::Tensor> cocoeval_match_image(torch::Tensor det_boxes,
-                                                torch::Tensor det_scores,
-                                                torch::Tensor gt_boxes,
-                                                torch::Tensor gt_crowd,
-                                                torch::Tensor iou_thrs,
-                                                int64_t max_dets);
+std::vector<torch::Tensor> cocoeval_match_image(
+    torch::Tensor det_boxes, torch::Tensor det_scores, torch::Tensor gt_boxes,
+    torch::Tensor gt_crowd, torch::Tensor iou_thrs, int64_t max_dets,
+    c10::optional<torch::Tensor> gt_extra_ignore);
 // window.hip
 torch::Tensor window_partition_fwd(torch::Tensor x, int64_t ws, int64_t shift);
 torch::Tensor window_partition_bwd(torch::Tensor grad, int64_t B, int64_t H,
@@ -155,7 +153,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("bias") = py::none(), py::arg("mask") = py::none(),
         py::arg("save_p") = false);
   m.def("mfma_probe", &mfma_probe);
-  m.def("cocoeval_match_image", &cocoeval_match_image);
+  m.def("cocoeval_match_image", &cocoeval_match_image, py::arg("det_boxes"),
+        py::arg("det_scores"), py::arg("gt_boxes"), py::arg("gt_crowd"),
+        py::arg("iou_thrs"), py::arg("max_dets"),
+        py::arg("gt_extra_ignore") = py::none());
   m.def("my_add", &my_add);
   m.def("window_partition_fwd", &window_partition_fwd);
   m.def("window_partition_bwd", &window_partition_bwd);

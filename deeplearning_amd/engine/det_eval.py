@@ -31,6 +31,17 @@ def match_image_native(det_boxes, det_scores, gt_boxes, gt_crowd, iou_thrs,
     return matched, ignored, scores, int(n_gt)
 
 
+def match_image_ranged_native(det_boxes, det_scores, gt_boxes, gt_crowd,
+                              gt_extra_ignore, iou_thrs, max_dets=100):
+    """C++ fast path of match_image_ranged (area-range evaluation)."""
+    from ..ops._ext import ext
+    matched, ignored, scores, n_gt = ext().cocoeval_match_image(
+        det_boxes.float(), det_scores.float(), gt_boxes.float(),
+        gt_crowd.to(torch.bool), torch.as_tensor(iou_thrs), max_dets,
+        gt_extra_ignore.to(torch.bool))
+    return matched, ignored, scores, int(n_gt)
+
+
 def match_image(det_boxes, det_scores, gt_boxes, gt_crowd, iou_thrs,
                 max_dets=100):
     """Greedy COCO matching for one image+class.
@@ -242,9 +253,14 @@ class DetEvaluator:
             db_s, ds_s = db[order], ds[order]
             ga = _box_areas(gb)
             extra = (ga < lo) | (ga > hi) if gb.numel() else cr.clone()
-            m, ig, sc, n = match_image_ranged(db_s, ds_s, gb, cr, extra,
-                                              self.iou_thrs, self.max_dets,
-                                              presorted=True)
+            if has_ext():
+                m, ig, sc, n = match_image_ranged_native(
+                    db_s, ds_s, gb, cr, extra, self.iou_thrs, self.max_dets)
+            else:
+                m, ig, sc, n = match_image_ranged(db_s, ds_s, gb, cr, extra,
+                                                  self.iou_thrs,
+                                                  self.max_dets,
+                                                  presorted=True)
             da = _box_areas(db_s)
             det_out = (da < lo) | (da > hi)
             ig = ig | (~m & det_out[None, :])

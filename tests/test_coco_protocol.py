@@ -383,3 +383,12 @@ def test_ranged_matcher_fuzz_vs_literal_loop(data):
     assert n == r_n
     assert np.array_equal((m & ~ig).numpy(), r_m & ~r_ig)
     assert np.array_equal(ig.numpy(), r_ig)
+
+    from deeplearning_amd.engine.det_eval import match_image_ranged_native
+    from deeplearning_amd.ops._ext import has_ext
+    if has_ext():
+        c_m, c_ig, c_sc, c_n = match_image_ranged_native(
+            dt_s, sc_s, gt, crowd, extra, COCO_IOU_THRS, 100)
+        c_ig = c_ig | (~c_m & ((da < lo) | (da > hi))[None, :])
+        assert c_n == n
+        assert torch.equal(c_m, m) and torch.equal(c_ig, ig)
