@@ -213,7 +213,13 @@ def det_train_main(args, model_kwargs=None) -> dict:
             evaluator.update(dets, targets)
     evaluator.synchronize_between_processes()
     stats = evaluator.summarize()
-    logger.info(f"final mAP {stats['mAP']:.4f} mAP50 {stats['mAP50']:.4f}")
+    logger.info(
+        f"final mAP {stats['mAP']:.4f} mAP50 {stats['mAP50']:.4f} "
+        f"mAP75 {stats['mAP75']:.4f} | s/m/l "
+        f"{stats.get('mAP_small', 0):.4f}/{stats.get('mAP_medium', 0):.4f}/"
+        f"{stats.get('mAP_large', 0):.4f} | AR1/10/100 "
+        f"{stats.get('AR1', 0):.4f}/{stats.get('AR10', 0):.4f}/"
+        f"{stats.get('AR100', 0):.4f}")
     cleanup()
     return {"mAP": stats["mAP"], "run_dir": str(run_dir)}
 
