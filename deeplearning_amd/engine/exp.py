@@ -96,6 +96,16 @@ class YoloxExp(BaseExp):
         return DataLoader(ds, batch_size=batch_size, shuffle=True,
                           collate_fn=SyntheticDetection.collate_fn)
 
+    def close_mosaic_if_due(self, loader, epoch) -> bool:
+        """Disable mosaic for the last `no_aug_epochs` epochs (ref
+        yolox/core/trainer.py before_epoch: close_mosaic + L1 switch).
+        Returns True when augmentation is off for this epoch."""
+        off = epoch >= self.max_epoch - self.no_aug_epochs
+        ds = getattr(loader, "dataset", None)
+        if off and hasattr(ds, "enabled"):
+            ds.enabled = False
+        return off
+
 
 def get_exp(exp_file: str | None = None, exp_name: str | None = None):
     """Load an Exp from a python file (ref yolox/exp/build.py get_exp)."""

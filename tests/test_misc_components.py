@@ -163,3 +163,19 @@ def test_generic_export_covers_pose_model(tmp_path):
         capture_output=True, text=True, timeout=300)
     assert r.returncode == 0, r.stderr[-1500:]
     assert "exported" in r.stdout
+
+
+def test_exp_closes_mosaic_for_no_aug_epochs():
+    """YOLOX protocol: the last no_aug_epochs train WITHOUT mosaic
+    (ref yolox/core/trainer.py before_epoch)."""
+    from deeplearning_amd.engine.exp import get_exp
+
+    exp = get_exp(exp_name="yolox_s")
+    exp.max_epoch = 20
+    exp.no_aug_epochs = 5
+    loader = exp.get_data_loader(batch_size=2, synthetic_size=4)
+    assert loader.dataset.enabled  # mosaic on initially
+    assert not exp.close_mosaic_if_due(loader, epoch=10)
+    assert loader.dataset.enabled
+    assert exp.close_mosaic_if_due(loader, epoch=15)
+    assert not loader.dataset.enabled
