@@ -187,7 +187,8 @@ class YOLOX(nn.Module):
         if self.training:
             assert targets is not None
             return yolox_loss(outputs, targets, self.nc,
-                              YOLOXHead.strides)
+                              YOLOXHead.strides,
+                              use_l1=getattr(self, "use_l1", False))
         return decode_outputs(outputs, YOLOXHead.strides)
 
 
@@ -267,8 +268,10 @@ def simota_assign(pred_boxes, pred_cls, pred_obj, gt_boxes, gt_labels,
     return fg_mask, matched_gt, matched_ious
 
 
-def yolox_loss(outputs, targets, num_classes, strides):
-    """targets: list of dicts with 'boxes' (xyxy pixels) and 'labels'."""
+def yolox_loss(outputs, targets, num_classes, strides, use_l1=False):
+    """targets: list of dicts with 'boxes' (xyxy pixels) and 'labels'.
+    use_l1 adds the raw-output L1 regression term the reference enables for
+    the no-aug tail epochs (yolo_head.py get_losses/get_l1_target)."""
     device = outputs[0].device
     flat = [o.flatten(2).permute(0, 2, 1) for o in outputs]
     out = torch.cat(flat, dim=1)  # B, P, 5+nc
@@ -284,6 +287,7 @@ def yolox_loss(outputs, targets, num_classes, strides):
     loss_iou = out.new_zeros(())
     loss_obj = out.new_zeros(())
     loss_cls = out.new_zeros(())
+    loss_l1 = out.new_zeros(())
     for b in range(B):
         t = targets[b]
         gt, labels = t["boxes"], t["labels"]
@@ -313,9 +317,22 @@ def yolox_loss(outputs, targets, num_classes, strides):
             iou.detach()[:, None]
         loss_cls = loss_cls + F.binary_cross_entropy_with_logits(
             pred_cls[b][fg_mask], cls_target, reduction="sum")
+        if use_l1:
+            # raw-output-space target (ref get_l1_target): t_xy = gt_c/stride
+            # - grid, t_wh = log(gt_wh/stride)
+            st = stride_t.squeeze(-1)[fg_mask, None]
+            g_c = (gb[:, :2] + gb[:, 2:]) / 2
+            g_wh = (gb[:, 2:] - gb[:, :2]).clamp(min=1e-8)
+            l1_t = torch.cat([g_c / st - grids[fg_mask],
+                              torch.log(g_wh / st)], 1)
+            loss_l1 = loss_l1 + F.l1_loss(out[b][fg_mask][:, :4], l1_t,
+                                          reduction="sum")
     n = max(num_fg_total, 1)
-    return {"iou_loss": 5.0 * loss_iou / n, "obj_loss": loss_obj / n,
-            "cls_loss": loss_cls / n}
+    losses = {"iou_loss": 5.0 * loss_iou / n, "obj_loss": loss_obj / n,
+              "cls_loss": loss_cls / n}
+    if use_l1:
+        losses["l1_loss"] = loss_l1 / n
+    return losses
 
 
 def yolox_postprocess(decoded, num_classes, conf_thre=0.25, nms_thre=0.45):

@@ -320,3 +320,26 @@ def test_transform_postprocess_rescales_back():
     out = tr.postprocess(det, il.image_sizes, [(50, 80)])
     assert torch.allclose(out[0]["boxes"],
                           torch.tensor([[0.0, 0.0, 80.0, 50.0]]), atol=0.5)
+
+
+def test_yolox_l1_switch():
+    """The no-aug-phase L1 term (ref yolo_head.py use_l1): off by default,
+    adds a finite differentiable l1_loss when the model flag is set."""
+    import torch
+
+    from deeplearning_amd.models import build_model
+
+    torch.manual_seed(0)
+    m = build_model("yolox_s", num_classes=5)
+    m.train()
+    x = torch.randn(1, 3, 256, 256)
+    t = [{"boxes": torch.tensor([[30.0, 40.0, 120.0, 160.0]]),
+          "labels": torch.tensor([2])}]
+    losses = m(x, t)
+    assert "l1_loss" not in losses
+    m.use_l1 = True
+    losses = m(x, t)
+    assert "l1_loss" in losses and torch.isfinite(losses["l1_loss"])
+    sum(losses.values()).backward()
+    assert any(p.grad is not None and p.grad.abs().sum() > 0
+               for p in m.head.parameters())
