@@ -39,6 +39,8 @@ def parse_args():
     p.add_argument("--device", default="cuda")
     p.add_argument("--output", default="runs")
     p.add_argument("--synthetic-size", type=int, default=16)
+    p.add_argument("--no-aug-epochs", type=int, default=0,
+                   help="last N epochs: L1 term on (ref no_aug_epochs)")
     p.add_argument("--multiscale", action="store_true",
                    help="random input size every 10 iters, broadcast-synced "
                         "across ranks (ref yolox_base.py:167-187)")
@@ -73,6 +75,11 @@ def main(args):
     cur_size = args.img_size
     it_count = 0
     for epoch in range(args.epochs):
+        if args.no_aug_epochs > 0 and \
+                epoch >= args.epochs - args.no_aug_epochs:
+            # no-aug tail: reference also enables the raw-output L1 term
+            from deeplearning_amd.core.checkpoint import unwrap_model
+            unwrap_model(model).use_l1 = True
         model.train()
         t0 = time.time()
         tot = 0.0
