@@ -62,7 +62,7 @@ class RetinaNet(nn.Module):
     def __init__(self, num_classes=80, min_size=800, max_size=1333,
                  score_thresh=0.05, nms_thresh=0.5, detections_per_img=300,
                  topk_candidates=1000, fg_iou_thresh=0.5, bg_iou_thresh=0.4,
-                 trainable_backbone_layers=3):
+                 trainable_backbone_layers=3, reg_loss="l1"):
         super().__init__()
         self.backbone = resnet_fpn_backbone(
             returned_layers=(2, 3, 4), extra_blocks="p6p7",
@@ -82,6 +82,8 @@ class RetinaNet(nn.Module):
         self.detections_per_img = detections_per_img
         self.topk_candidates = topk_candidates
         self.num_classes = num_classes
+        assert reg_loss in ("l1", "giou"), reg_loss
+        self.reg_loss = reg_loss  # ref network_files/retinanet.py:153
 
     def compute_loss(self, targets, cls_logits, bbox_regression, anchors):
         cls_losses, reg_losses = [], []
@@ -104,11 +106,19 @@ class RetinaNet(nn.Module):
                 cls_logits[i][valid], gt_cls[valid], alpha=0.25, gamma=2.0,
                 reduction="sum") / num_fg)
             if fg.any():
-                gt_deltas = self.box_coder.encode(
-                    t["boxes"][matched[fg]], anchors_i[fg])
-                reg_losses.append(F.l1_loss(
-                    bbox_regression[i][fg], gt_deltas,
-                    reduction="sum") / num_fg)
+                if self.reg_loss == "giou":
+                    from ...ops.boxes import bbox_iou_aligned
+                    pred = self.box_coder.decode(bbox_regression[i][fg],
+                                                 anchors_i[fg])
+                    g = bbox_iou_aligned(pred, t["boxes"][matched[fg]],
+                                         xywh=False, GIoU=True).squeeze(-1)
+                    reg_losses.append((1.0 - g).sum() / num_fg)
+                else:
+                    gt_deltas = self.box_coder.encode(
+                        t["boxes"][matched[fg]], anchors_i[fg])
+                    reg_losses.append(F.l1_loss(
+                        bbox_regression[i][fg], gt_deltas,
+                        reduction="sum") / num_fg)
             else:
                 reg_losses.append(bbox_regression[i].sum() * 0)
         return {"classification": torch.stack(cls_losses).mean(),

@@ -343,3 +343,23 @@ def test_yolox_l1_switch():
     sum(losses.values()).backward()
     assert any(p.grad is not None and p.grad.abs().sum() > 0
                for p in m.head.parameters())
+
+
+def test_retinanet_giou_reg_loss():
+    """reg_loss='giou' (ref RetinaNet losses.py:153 GIoU option) trains and
+    differs from the default L1 objective."""
+    import torch
+
+    from deeplearning_amd.models import build_model
+
+    torch.manual_seed(0)
+    m = build_model("retinanet_resnet50_fpn", num_classes=5, min_size=128,
+                    max_size=128, reg_loss="giou")
+    m.train()
+    x = [torch.randn(3, 128, 128)]
+    t = [{"boxes": torch.tensor([[10.0, 12.0, 60.0, 70.0]]),
+          "labels": torch.tensor([1])}]
+    losses = m(x, t)
+    assert torch.isfinite(losses["bbox_regression"])
+    sum(losses.values()).backward()
+    assert m.head.bbox_pred.weight.grad is not None
