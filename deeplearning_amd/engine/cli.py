@@ -182,7 +182,10 @@ def classification_train_main(args) -> dict:
     if get_world_size() > 1:
         model = wrap_data_parallel(model)
 
-    params = [p for p in model.parameters() if p.requires_grad]
+    # norm scales / biases / model-declared keys train without weight decay
+    # (ref swin utils/optimizer.py set_weight_decay)
+    from .optim_groups import param_groups_weight_decay
+    params = param_groups_weight_decay(model, args.weight_decay)
     if args.optimizer == "adamw":
         optimizer = torch.optim.AdamW(params, lr=args.lr,
                                       weight_decay=args.weight_decay)

@@ -330,6 +330,15 @@ class SwinTransformer(nn.Module):
             if m.bias is not None:
                 nn.init.zeros_(m.bias)
 
+    @torch.jit.ignore
+    def no_weight_decay(self):
+        # ref swin_transformer.py no_weight_decay{_keywords}
+        return {"absolute_pos_embed"}
+
+    @torch.jit.ignore
+    def no_weight_decay_keywords(self):
+        return {"relative_position_bias_table", "logit_scale", "cpb_mlp"}
+
     def forward_features(self, x):
         from .vit import patch_embed_gemm
         x = patch_embed_gemm(x, self.patch_embed.weight,

@@ -151,6 +151,10 @@ class VisionTransformer(nn.Module):
             if m.bias is not None:
                 nn.init.zeros_(m.bias)
 
+    @torch.jit.ignore
+    def no_weight_decay(self):
+        return {"pos_embed", "cls_token", "dist_token"}
+
     def forward_features(self, x):
         x = self.patch_embed(x)
         cls = self.cls_token.expand(x.shape[0], -1, -1)

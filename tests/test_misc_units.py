@@ -118,3 +118,30 @@ def test_complexity_batch_scales_linearly():
     m1 = estimate_macs(m, torch.randn(1, 1, 28, 28))
     m4 = estimate_macs(m, torch.randn(4, 1, 28, 28))
     assert m4 == 4 * m1 > 0
+
+
+def test_weight_decay_groups():
+    """set_weight_decay parity (swin utils/optimizer.py): biases, 1-D norm
+    scales and model-declared keys land in the wd=0 group."""
+    from deeplearning_amd.engine.optim_groups import param_groups_weight_decay
+    from deeplearning_amd.models import build_model
+
+    m = build_model("swin_t", num_classes=10)
+    groups = param_groups_weight_decay(m, 0.05)
+    assert len(groups) == 2
+    assert groups[0]["weight_decay"] == 0.05
+    assert groups[1]["weight_decay"] == 0.0
+    nd_ids = {id(p) for p in groups[1]["params"]}
+    for name, p in m.named_parameters():
+        if "relative_position_bias_table" in name or name.endswith(".bias") \
+                or p.ndim <= 1:
+            assert id(p) in nd_ids, name
+    total = sum(len(g["params"]) for g in groups)
+    assert total == sum(1 for p in m.parameters() if p.requires_grad)
+
+    v = build_model("vit_b16", num_classes=10)
+    gv = param_groups_weight_decay(v, 0.05)
+    ndv = {id(p) for p in gv[1]["params"]}
+    named = dict(v.named_parameters())
+    assert id(named["pos_embed"]) in ndv
+    assert id(named["cls_token"]) in ndv
