@@ -253,12 +253,15 @@ def classification_train_main(args) -> dict:
                             loss = loss + 0.3 * cross_entropy(aux, y_hard)
             scaler.scale(loss / args.accumulate_steps).backward()
             if (it + 1) % args.accumulate_steps == 0:
-                if args.clip_grad > 0:
-                    scaler.unscale_(optimizer)
-                    torch.nn.utils.clip_grad_norm_(params, args.clip_grad)
+                # sync FIRST: finalize() writes the all-reduced averaged
+                # grads into p.grad — clipping before it would be overwritten
                 finalize = getattr(model, "finalize", None)
                 if finalize is not None:
                     finalize()
+                if args.clip_grad > 0:
+                    scaler.unscale_(optimizer)
+                    torch.nn.utils.clip_grad_norm_(model.parameters(),
+                                                   args.clip_grad)
                 scaler.step(optimizer)
                 scaler.update()
                 optimizer.zero_grad(set_to_none=True)
