@@ -108,3 +108,28 @@ def test_cross_entropy_cpu():
     target = torch.randint(0, 10, (16,))
     torch.testing.assert_close(ops.cross_entropy(logits, target),
                                F.cross_entropy(logits, target))
+
+
+def test_eager_attention_mask_broadcast_vs_loop():
+    """The eager fallback's [nW,N,N] window-mask broadcast must equal a
+    per-window loop (guards the CPU path of fused_attention)."""
+    import torch
+
+    from deeplearning_amd.ops.attention import _eager_attention
+
+    torch.manual_seed(0)
+    nW, B_per, N, H, d = 4, 3, 16, 2, 8
+    B = nW * B_per
+    qkv = torch.randn(B, N, 3 * H * d)
+    bias = torch.randn(H, N, N)
+    mask = torch.randn(nW, N, N)
+    out = _eager_attention(qkv, H, 0.3, bias=bias, mask=mask)
+
+    q, k, v = qkv.reshape(B, N, 3, H, d).permute(2, 0, 3, 1, 4).unbind(0)
+    ref = torch.empty(B, H, N, d)
+    for b in range(B):
+        w = b % nW  # windows are the FAST index in [B//nW, nW] batching
+        a = (q[b] @ k[b].transpose(-2, -1)) * 0.3 + bias + mask[w]
+        ref[b] = a.softmax(-1) @ v[b]
+    ref = ref.transpose(1, 2).reshape(B, N, H * d)
+    assert torch.allclose(out, ref, atol=1e-5)
