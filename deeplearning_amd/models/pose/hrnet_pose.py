@@ -87,15 +87,29 @@ def heatmap_nms(heat: torch.Tensor, kernel=3) -> torch.Tensor:
     return heat * (hmax == heat).to(heat.dtype)
 
 
-def decode_heatmaps(heat: torch.Tensor, stride=4):
+def decode_heatmaps(heat: torch.Tensor, stride=4, refine=True):
     """[N,K,H,W] -> coords [N,K,2] (input pixels) + scores [N,K]
-    (ref utils/train_and_eval.py:136-217)."""
-    heat = heatmap_nms(torch.sigmoid(heat))
-    N, K, H, W = heat.shape
-    flat = heat.view(N, K, -1)
+    (ref utils/train_and_eval.py:136-217). refine: quarter-pixel shift
+    toward the higher neighbor (simple-baselines post-processing)."""
+    prob = torch.sigmoid(heat)
+    peaks = heatmap_nms(prob)
+    N, K, H, W = peaks.shape
+    flat = peaks.view(N, K, -1)
     scores, idx = flat.max(-1)
     ys = (idx // W).float()
     xs = (idx % W).float()
+    if refine and H > 2 and W > 2:
+        pf = prob.view(N, K, -1)
+        xi = idx % W
+        yi = idx // W
+        inner_x = (xi > 0) & (xi < W - 1)
+        inner_y = (yi > 0) & (yi < H - 1)
+        g = lambda off, lo, hi: pf.gather(
+            -1, (idx + off).clamp(min=lo, max=hi).unsqueeze(-1)).squeeze(-1)
+        right, left = g(1, 0, H * W - 1), g(-1, 0, H * W - 1)
+        down, up = g(W, 0, H * W - 1), g(-W, 0, H * W - 1)
+        xs = xs + 0.25 * torch.sign(right - left) * inner_x.float()
+        ys = ys + 0.25 * torch.sign(down - up) * inner_y.float()
     coords = torch.stack([xs, ys], dim=-1) * stride
     return coords, scores
 

@@ -114,3 +114,22 @@ def test_lovasz_losses():
     loss = lovasz_hinge(logits, torch.randint(0, 2, (50,)))
     loss.backward()
     assert logits.grad is not None
+
+
+def test_pose_decode_subpixel_refinement():
+    """Quarter-offset refinement shifts toward the higher neighbor
+    (simple-baselines post-processing in the reference decode)."""
+    import torch
+
+    from deeplearning_amd.models.pose.hrnet_pose import decode_heatmaps
+
+    h = torch.full((1, 1, 9, 9), -8.0)
+    h[0, 0, 4, 4] = 4.0    # peak at (x=4, y=4)
+    h[0, 0, 4, 5] = 2.0    # right neighbor hotter than left
+    h[0, 0, 3, 4] = 2.0    # upper neighbor hotter than lower
+    coords, scores = decode_heatmaps(h, stride=4, refine=True)
+    x, y = coords[0, 0].tolist()
+    assert x == (4 + 0.25) * 4
+    assert y == (4 - 0.25) * 4
+    coords_raw, _ = decode_heatmaps(h, stride=4, refine=False)
+    assert coords_raw[0, 0].tolist() == [16.0, 16.0]
