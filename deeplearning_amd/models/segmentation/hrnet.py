@@ -167,7 +167,14 @@ class OhemCrossEntropy(nn.Module):
         min_value = pred[min(self.min_kept, pred.numel() - 1)]
         threshold = max(min_value, self.thresh)
         pixel_losses = pixel_losses[mask][ind]
-        return pixel_losses[pred < threshold].mean()
+        keep = pred < threshold
+        if not bool(keep.any()):
+            # all predictions equally confident (e.g. near-uniform logits):
+            # strict < selects nothing and mean() would be NaN — keep the
+            # min_kept hardest instead
+            k = min(self.min_kept, pixel_losses.numel())
+            return pixel_losses[:k].mean()
+        return pixel_losses[keep].mean()
 
 
 @register_model

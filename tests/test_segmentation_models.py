@@ -66,3 +66,19 @@ def test_dice_and_confusion_matrix():
     acc, _, iou = cm.compute()
     assert acc.item() == pytest.approx(1.0)
     assert iou[iou == iou].mean().item() == pytest.approx(1.0)  # ignore NaN rows
+
+
+def test_ohem_uniform_logits_no_nan():
+    """OHEM edge: identical confidences make the strict threshold select
+    nothing (reference code NaNs there) — must fall back to the hardest-k."""
+    import torch
+
+    from deeplearning_amd.models.segmentation.hrnet import OhemCrossEntropy
+
+    crit = OhemCrossEntropy(thres=0.7, min_kept=10)
+    logits = torch.zeros(1, 3, 8, 8, requires_grad=True)  # uniform softmax
+    target = torch.randint(0, 3, (1, 8, 8))
+    loss = crit(logits, target)
+    assert torch.isfinite(loss)
+    loss.backward()
+    assert torch.isfinite(logits.grad).all()
