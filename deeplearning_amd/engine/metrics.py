@@ -82,35 +82,12 @@ def dice_loss(logits: torch.Tensor, target_onehot: torch.Tensor) -> torch.Tensor
 @torch.no_grad()
 def count_flops(model, input_shape=(1, 3, 224, 224), device="cpu",
                 inputs=None):
-    """Forward-pass FLOPs (multiply-adds x2) via module hooks for
-    Conv2d/Linear/bmm-free models (reference SwinTransformer.flops /
-    ViT fvcore flops.py parity). Returns (flops, params)."""
-    import torch.nn as nn
+    """Forward-pass FLOPs (multiply-adds x2 — reference
+    SwinTransformer.flops / fvcore convention) and parameter count.
+    Thin front over core/complexity.estimate_macs (the single counting
+    implementation; also covers Conv1d/3d/ConvTranspose)."""
+    from ..core.complexity import count_params, estimate_macs
 
-    flops = [0]
-
-    def conv_hook(m, inp, out):
-        kh, kw = m.kernel_size
-        cout = out.shape[1]
-        flops[0] += 2 * out.numel() * (inp[0].shape[1] // m.groups) * kh * kw
-
-    def linear_hook(m, inp, out):
-        flops[0] += 2 * out.numel() * m.in_features
-
-    handles = []
-    for mod in model.modules():
-        if isinstance(mod, nn.Conv2d):
-            handles.append(mod.register_forward_hook(conv_hook))
-        elif isinstance(mod, nn.Linear):
-            handles.append(mod.register_forward_hook(linear_hook))
-    was_training = model.training
-    model.eval()
     x = inputs if inputs is not None else torch.randn(*input_shape,
                                                       device=device)
-    model(x)
-    for h in handles:
-        h.remove()
-    if was_training:
-        model.train()
-    params = sum(p.numel() for p in model.parameters())
-    return flops[0], params
+    return 2 * estimate_macs(model, x), count_params(model)
