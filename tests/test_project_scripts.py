@@ -126,14 +126,16 @@ def test_detection_export_scripts(script, out, tmp_path):
 
 
 def test_all_project_scripts_compile():
-    """Syntax safety net over the whole projects/ tree (the per-script run
+    """Syntax safety net over projects/ AND tools/ (the per-script run
     smokes above cover a subset; this catches a broken edit in any of the
-    ~110 thin CLI wrappers)."""
+    thin CLI wrappers or GPU-only tools)."""
+    import itertools
     import pathlib
     import py_compile
 
     bad = []
-    for f in sorted(pathlib.Path("projects").rglob("*.py")):
+    for f in sorted(itertools.chain(pathlib.Path("projects").rglob("*.py"),
+                                    pathlib.Path("tools").glob("*.py"))):
         try:
             py_compile.compile(str(f), doraise=True)
         except py_compile.PyCompileError as e:
